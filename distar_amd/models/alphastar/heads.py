@@ -1,0 +1,394 @@
+"""Autoregressive policy heads.
+
+Functional parity with the reference's
+`distar/agent/default/model/head/{action_type_head,action_arg_head}.py`;
+module/key layout matches the reference checkpoints.
+
+MI355X-first restructure (the big one): the reference's SelectedUnitsHead
+*training* path runs a Python loop of up to 64 sequential steps, each doing
+fc+LSTM+masking on the whole batch (`action_arg_head.py:168-216`).  Under
+teacher forcing, both the autoregressive-embedding evolution and the
+logits-mask evolution are pure functions of the *labels*, so we compute them
+in closed form:
+
+  - cumulative label one-hots / first-occurrence / end-flag gating via
+    `cumsum` over the selection axis (no loop),
+  - all S lstm inputs with one batched fc pair,
+  - the (S, B, 32) query chain with the tiny LN-LSTM (the only truly
+    sequential part, 32-wide),
+  - the (B, S, N+1) logits with a single batched GEMM against the keys.
+
+This removes ~60 sequential launches x 10 kernels from the SL/RL train step.
+The sampling path (data-dependent) keeps the reference's step loop.
+"""
+import math
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+from torch import Tensor
+
+from ..nn.blocks import (fc_block, conv2d_block, build_activation, ResBlock,
+                         ResFCBlock, GatedResBlock, GLU, sequence_mask)
+from ..nn.lnlstm import script_lnlstm
+from ...lib.consts import MAX_ENTITY_NUM, MAX_SELECTED_UNITS_NUM
+from ...lib.stat import ACTION_RACE_MASK
+
+
+class ActionTypeHead(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.whole_cfg = cfg
+        self.cfg = cfg.model.policy.head.action_type_head
+        self.act = build_activation(self.cfg.activation)
+        self.project = fc_block(self.cfg.input_dim, self.cfg.res_dim, activation=self.act,
+                                norm_type=None)
+        self.res = nn.Sequential(*[ResFCBlock(self.cfg.res_dim, self.act, self.cfg.norm_type)
+                                   for _ in range(self.cfg.res_num)])
+        self.action_fc = GLU(self.cfg.res_dim, self.cfg.action_num, self.cfg.context_dim)
+        self.action_map_fc1 = fc_block(self.cfg.action_num, self.cfg.action_map_dim,
+                                       activation=self.act, norm_type=None)
+        self.action_map_fc2 = fc_block(self.cfg.action_map_dim, self.cfg.action_map_dim,
+                                       activation=None, norm_type=None)
+        self.glu1 = GLU(self.cfg.action_map_dim, self.cfg.gate_dim, self.cfg.context_dim)
+        self.glu2 = GLU(self.cfg.input_dim, self.cfg.gate_dim, self.cfg.context_dim)
+        self.action_num = self.cfg.action_num
+        self.use_mask = cfg.get('common', {}).get('type', 'train') == 'play'
+        self.race = 'zerg'
+
+    def forward(self, lstm_output, scalar_context, action_type: Optional[Tensor] = None
+                ) -> Tuple[Tensor, Tensor, Tensor]:
+        x = self.project(lstm_output)
+        x = self.res(x)
+        x = self.action_fc(x, scalar_context)
+        x = x / self.whole_cfg.model.temperature
+        if self.use_mask:
+            mask = ACTION_RACE_MASK[self.race].to(x.device)
+            x = x.masked_fill(~mask.unsqueeze(0), -1e9)
+        if action_type is None:
+            p = F.softmax(x, dim=1)
+            action_type = torch.multinomial(p, 1)[:, 0]
+        action_one_hot = F.one_hot(action_type.long(), self.action_num).float()
+        embedding1 = self.action_map_fc1(action_one_hot)
+        embedding1 = self.action_map_fc2(embedding1)
+        embedding1 = self.glu1(embedding1, scalar_context)
+        embedding2 = self.glu2(lstm_output, scalar_context)
+        return x, action_type, embedding1 + embedding2
+
+
+class DelayHead(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.whole_cfg = cfg
+        self.cfg = cfg.model.policy.head.delay_head
+        self.act = build_activation(self.cfg.activation)
+        self.fc1 = fc_block(self.cfg.input_dim, self.cfg.decode_dim, activation=self.act, norm_type=None)
+        self.fc2 = fc_block(self.cfg.decode_dim, self.cfg.decode_dim, activation=self.act, norm_type=None)
+        self.fc3 = fc_block(self.cfg.decode_dim, self.cfg.delay_dim, activation=None, norm_type=None)
+        self.embed_fc1 = fc_block(self.cfg.delay_dim, self.cfg.delay_map_dim, activation=self.act, norm_type=None)
+        self.embed_fc2 = fc_block(self.cfg.delay_map_dim, self.cfg.input_dim, activation=None, norm_type=None)
+        self.delay_dim = self.cfg.delay_dim
+
+    def forward(self, embedding, delay: Optional[Tensor] = None):
+        x = self.fc3(self.fc2(self.fc1(embedding)))
+        if delay is None:
+            p = F.softmax(x, dim=1)
+            delay = torch.multinomial(p, 1)[:, 0]
+        delay_one_hot = F.one_hot(delay.long(), self.delay_dim).float()
+        embedding_delay = self.embed_fc2(self.embed_fc1(delay_one_hot))
+        return x, delay, embedding + embedding_delay
+
+
+class QueuedHead(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.whole_cfg = cfg
+        self.cfg = cfg.model.policy.head.queued_head
+        self.act = build_activation(self.cfg.activation)
+        self.fc1 = fc_block(self.cfg.input_dim, self.cfg.decode_dim, activation=self.act, norm_type=None)
+        self.fc2 = fc_block(self.cfg.decode_dim, self.cfg.decode_dim, activation=self.act, norm_type=None)
+        self.fc3 = fc_block(self.cfg.decode_dim, self.cfg.queued_dim, activation=None, norm_type=None)
+        self.embed_fc1 = fc_block(self.cfg.queued_dim, self.cfg.queued_map_dim, activation=self.act, norm_type=None)
+        self.embed_fc2 = fc_block(self.cfg.queued_map_dim, self.cfg.input_dim, activation=None, norm_type=None)
+        self.queued_dim = self.cfg.queued_dim
+
+    def forward(self, embedding, queued: Optional[Tensor] = None):
+        x = self.fc3(self.fc2(self.fc1(embedding)))
+        x = x / self.whole_cfg.model.temperature
+        if queued is None:
+            p = F.softmax(x, dim=1)
+            queued = torch.multinomial(p, 1)[:, 0]
+        queued_one_hot = F.one_hot(queued.long(), self.queued_dim).float()
+        embedding_queued = self.embed_fc2(self.embed_fc1(queued_one_hot))
+        return x, queued, embedding + embedding_queued
+
+
+class SelectedUnitsHead(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.whole_cfg = cfg
+        self.cfg = cfg.model.policy.head.selected_units_head
+        self.act = build_activation(self.cfg.activation)
+        self.key_fc = fc_block(self.cfg.entity_embedding_dim, self.cfg.key_dim,
+                               activation=None, norm_type=None)
+        self.query_fc1 = fc_block(self.cfg.input_dim, self.cfg.func_dim, activation=self.act)
+        self.query_fc2 = fc_block(self.cfg.func_dim, self.cfg.key_dim, activation=None)
+        self.embed_fc1 = fc_block(self.cfg.key_dim, self.cfg.func_dim, activation=self.act, norm_type=None)
+        self.embed_fc2 = fc_block(self.cfg.func_dim, self.cfg.input_dim, activation=None, norm_type=None)
+        self.max_select_num = MAX_SELECTED_UNITS_NUM
+        self.max_entity_num = MAX_ENTITY_NUM
+        self.key_dim = self.cfg.key_dim
+        self.num_layers = self.cfg.num_layers
+        self.lstm = script_lnlstm(self.cfg.key_dim, self.cfg.hidden_dim, self.cfg.num_layers)
+        self.end_embedding = nn.Parameter(torch.empty(1, self.key_dim))
+        stdv = 1. / math.sqrt(self.end_embedding.size(1))
+        self.end_embedding.data.uniform_(-stdv, stdv)
+        self.extra_units = self.whole_cfg.get('agent', {}).get('extra_units', False)
+        assert self.whole_cfg.model.entity_reduce_type == 'selected_units_num', \
+            'this build implements the default reduce type (reference default)'
+
+    def _get_key_mask(self, entity_embedding, entity_num):
+        """Keys (with the learned end-embedding spliced in at index
+        entity_num) + availability mask (reference `action_arg_head.py:118-149`)."""
+        bs, n = entity_embedding.shape[:2]
+        key = self.key_fc(entity_embedding)                       # B, N, k
+        key = torch.cat([key, key.new_zeros(bs, 1, self.key_dim)], dim=1)  # B, N+1, k
+        flag = torch.ones(bs, n + 1, 1, dtype=torch.bool, device=key.device)
+        flag[torch.arange(bs, device=key.device), entity_num] = 0
+        key = key * flag + self.end_embedding.squeeze(0) * (~flag)
+        key_embeddings = key     # reduce_type == 'selected_units_num'
+        new_entity_num = entity_num + 1      # end slot is a valid position
+        mask = sequence_mask(new_entity_num, max_len=n + 1)
+        return key, mask, key_embeddings
+
+    def _su_embedding_from_labels(self, key_embeddings, labels, entity_num,
+                                  selected_units_num, seq_len):
+        """Closed-form replacement of the reference's per-step ae update.
+
+        Returns ae_delta (B, S, input_dim): the embed_fc2(embed_fc1(mean of
+        already-selected keys)) term for each step s (s=0 term is defined as
+        exactly zero — the reference uses the raw base embedding at step 0).
+        """
+        bs = labels.shape[0]
+        L = labels[:, :seq_len]                                    # B, S
+        is_end = L == entity_num.unsqueeze(1)                      # B, S
+        # included once the end token has appeared (inclusive) -> excluded
+        ended = torch.cummax(is_end.int(), dim=1)[0].bool()        # B, S
+        include = ~ended
+        # first occurrence of each label (set semantics of the one-hot)
+        onehot = torch.zeros(bs, seq_len, key_embeddings.shape[1],
+                             dtype=torch.int16, device=L.device)
+        onehot.scatter_(2, L.unsqueeze(-1), 1)
+        cum = onehot.cumsum(dim=1)                                 # B, S, N+1 (int16)
+        seen_before = (cum.gather(2, L.unsqueeze(-1)).squeeze(-1) - 1) > 0
+        contrib_gate = (include & ~seen_before).float()            # B, S
+        keys_at_labels = key_embeddings.gather(
+            1, L.unsqueeze(-1).expand(-1, -1, self.key_dim))       # B, S, k
+        contrib = keys_at_labels * contrib_gate.unsqueeze(-1)
+        cum_sum = contrib.cumsum(dim=1)                            # B, S, k
+        cum_cnt = contrib_gate.cumsum(dim=1)                       # B, S
+        # ae at step s uses labels < s: shift right by one
+        sum_s = torch.cat([cum_sum.new_zeros(bs, 1, self.key_dim), cum_sum[:, :-1]], dim=1)
+        cnt_s = torch.cat([cum_cnt.new_zeros(bs, 1), cum_cnt[:, :-1]], dim=1)
+        mean_s = sum_s.clone()
+        # reference: divide only rows with selected_units_num != 0; guard
+        # count==0 (keeps the raw sum, avoiding the reference's 0/0 edge)
+        div_rows = (selected_units_num != 0).unsqueeze(1) & (cnt_s > 0)
+        mean_s = torch.where(div_rows.unsqueeze(-1), sum_s / cnt_s.clamp(min=1).unsqueeze(-1), sum_s)
+        ae_delta = self.embed_fc2(self.embed_fc1(mean_s))          # B, S, input_dim
+        ae_delta[:, 0] = 0.                                        # step 0: raw base ae
+        # prev-selected mask for the logits (any label j < s disables index L[b,j])
+        prev_cnt = torch.cat([cum.new_zeros(bs, 1, cum.shape[2]), cum[:, :-1]], dim=1)
+        prev_selected = prev_cnt > 0                               # B, S, N+1
+        # final ae after the full unroll (consumed by the target-unit head)
+        fin_div = (selected_units_num != 0) & (cum_cnt[:, -1] > 0)
+        fin_mean = torch.where(fin_div.unsqueeze(-1),
+                               cum_sum[:, -1] / cum_cnt[:, -1].clamp(min=1).unsqueeze(-1),
+                               cum_sum[:, -1])
+        final_delta = self.embed_fc2(self.embed_fc1(fin_mean))
+        return ae_delta, prev_selected, final_delta
+
+    def _query_train(self, key, entity_num, autoregressive_embedding, logits_mask,
+                     key_embeddings, selected_units_num, selected_units):
+        bs = autoregressive_embedding.shape[0]
+        seq_len = max(int(selected_units_num.max()), 1)
+        ae_delta, prev_selected, final_delta = self._su_embedding_from_labels(
+            key_embeddings, selected_units, entity_num, selected_units_num, seq_len)
+        ae_all = autoregressive_embedding.unsqueeze(1) + ae_delta          # B, S, D
+        lstm_input = self.query_fc2(self.query_fc1(ae_all))                # B, S, k
+        lstm_input = lstm_input.transpose(0, 1).contiguous()               # S, B, k
+        state = [(lstm_input.new_zeros(bs, self.cfg.hidden_dim),
+                  lstm_input.new_zeros(bs, self.cfg.hidden_dim))
+                 for _ in range(self.num_layers)]
+        queries, _ = self.lstm(lstm_input, state)                          # S, B, k
+        logits = torch.einsum('sbk,bnk->bsn', queries, key)                # B, S, N+1
+        # mask: availability, minus previously-selected, end slot from step 1
+        arange = torch.arange(bs, device=key.device)
+        mask = logits_mask.unsqueeze(1) & ~prev_selected                   # B, S, N+1
+        mask[arange, 0, entity_num] = False
+        logits = logits.masked_fill(~mask, -1e9)
+        final_ae = autoregressive_embedding + final_delta
+        return logits, None, final_ae, selected_units_num, None
+
+    def _query_sample(self, key, entity_num, autoregressive_embedding, logits_mask,
+                      key_embeddings, su_mask):
+        """Data-dependent sampling loop (reference `action_arg_head.py:262-313`)."""
+        ae = autoregressive_embedding
+        bs = ae.shape[0]
+        device = ae.device
+        end_flag = torch.zeros(bs, dtype=torch.bool, device=device)
+        results_list, logits_list = [], []
+        state = [(ae.new_zeros(bs, self.cfg.hidden_dim), ae.new_zeros(bs, self.cfg.hidden_dim))
+                 for _ in range(self.num_layers)]
+        arange = torch.arange(bs, device=device)
+        logits_mask = logits_mask.clone()
+        logits_mask[arange, entity_num] = False
+        selected_units_num = torch.full((bs,), self.max_select_num, dtype=torch.long, device=device)
+        end_flag[~su_mask] = True
+        selected_units_num[~su_mask] = 0
+        result: Optional[Tensor] = None
+        sel_sum = ae.new_zeros(bs, self.key_dim)
+        sel_cnt = ae.new_zeros(bs)
+        step_logits = None
+        for i in range(self.max_select_num):
+            if i == 1:
+                logits_mask[arange, entity_num] = True     # end flag selectable now
+            if result is not None:
+                logits_mask[arange, result.detach()] = False
+            lstm_input = self.query_fc2(self.query_fc1(ae)).unsqueeze(0)
+            lstm_output, state = self.lstm(lstm_input, state)
+            queries = lstm_output.permute(1, 0, 2)                      # B, 1, k
+            step_logits = (queries * key).sum(dim=2)                    # B, N+1
+            step_logits = step_logits.masked_fill(~logits_mask, -1e9)
+            units = torch.multinomial(F.softmax(step_logits / self.whole_cfg.model.temperature,
+                                                dim=-1), 1)[:, 0]
+            result = units
+            newly_ended = (result == entity_num) & ~end_flag
+            selected_units_num[newly_ended] = i + 1
+            end_flag[result == entity_num] = True
+            results_list.append(result)
+            logits_list.append(step_logits)
+            picked = (~end_flag).float()
+            sel_sum = sel_sum + key_embeddings[arange, result] * picked.unsqueeze(1)
+            sel_cnt = sel_cnt + picked
+            mean = torch.where((sel_cnt > 0).unsqueeze(1), sel_sum / sel_cnt.clamp(min=1).unsqueeze(1),
+                               sel_sum)
+            ae = autoregressive_embedding + self.embed_fc2(self.embed_fc1(mean))
+            if bool(end_flag.all()):
+                break
+        extra_units = torch.zeros(bs, MAX_ENTITY_NUM + 1, device=device)
+        if self.extra_units and step_logits is not None:
+            end_flag_logit = step_logits[arange, entity_num]
+            extra_units = ((step_logits > end_flag_logit.unsqueeze(1)) & ~end_flag.unsqueeze(1)).float()
+        results = torch.stack(results_list, dim=0).transpose(1, 0).contiguous()
+        logits = torch.stack(logits_list, dim=0).transpose(1, 0).contiguous()
+        return logits, results, ae, selected_units_num, extra_units
+
+    def forward(self, embedding, entity_embedding, entity_num,
+                selected_units_num: Optional[Tensor] = None,
+                selected_units: Optional[Tensor] = None,
+                su_mask: Optional[Tensor] = None):
+        key, mask, key_embeddings = self._get_key_mask(entity_embedding, entity_num)
+        if selected_units is not None and selected_units_num is not None:
+            return self._query_train(key, entity_num, embedding, mask, key_embeddings,
+                                     selected_units_num, selected_units)
+        return self._query_sample(key, entity_num, embedding, mask, key_embeddings, su_mask)
+
+
+class TargetUnitHead(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.whole_cfg = cfg
+        self.cfg = cfg.model.policy.head.target_unit_head
+        self.act = build_activation(self.cfg.activation)
+        self.key_fc = fc_block(self.cfg.entity_embedding_dim, self.cfg.key_dim,
+                               activation=None, norm_type=None)
+        self.query_fc1 = fc_block(self.cfg.input_dim, self.cfg.key_dim, activation=self.act,
+                                  norm_type=None)
+        self.query_fc2 = fc_block(self.cfg.key_dim, self.cfg.key_dim, activation=None,
+                                  norm_type=None)
+        self.key_dim = self.cfg.key_dim
+        self.max_entity_num = MAX_ENTITY_NUM
+
+    def forward(self, embedding, entity_embedding, entity_num,
+                target_unit: Optional[Tensor] = None):
+        key = self.key_fc(entity_embedding)
+        mask = sequence_mask(entity_num, max_len=entity_embedding.shape[1])
+        query = self.query_fc2(self.query_fc1(embedding))
+        logits = (query.unsqueeze(1) * key).sum(dim=2)
+        logits = logits.masked_fill(~mask, -1e9)
+        logits = logits / self.whole_cfg.model.temperature
+        if target_unit is None:
+            p = F.softmax(logits, dim=1)
+            target_unit = torch.multinomial(p, 1)[:, 0]
+        return logits, target_unit
+
+
+class LocationHead(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        self.whole_cfg = cfg
+        self.cfg = cfg.model.policy.head.location_head
+        self.act = build_activation(self.cfg.activation)
+        self.reshape_size = self.cfg.reshape_size
+        self.reshape_channel = self.cfg.reshape_channel
+        self.conv1 = conv2d_block(self.cfg.map_skip_dim + self.cfg.reshape_channel,
+                                  self.cfg.res_dim, 1, 1, 0,
+                                  activation=build_activation(self.cfg.activation),
+                                  norm_type=None)
+        self.res_dim = self.cfg.res_dim
+        self.use_gate = self.cfg.get('gate', False)
+        self.use_unet = self.cfg.get('unet', False)
+        spatial_y = self.whole_cfg.model.spatial_y
+        spatial_x = self.whole_cfg.model.spatial_x
+        self.project_embed = fc_block(
+            self.cfg.input_dim, (spatial_y // 8) * (spatial_x // 8) * 4,
+            activation=build_activation(self.cfg.activation))
+        self.res = nn.ModuleList()
+        for _ in range(self.cfg.res_num):
+            if self.use_gate:
+                self.res.append(GatedResBlock(self.res_dim, self.res_dim, 3, 1, 1,
+                                              activation=build_activation(self.cfg.activation),
+                                              norm_type=None))
+            else:
+                self.res.append(ResBlock(self.res_dim, build_activation(self.cfg.activation),
+                                         norm_type=None))
+        self.upsample = nn.ModuleList()
+        dims = [self.res_dim] + list(self.cfg.upsample_dims)
+        assert self.cfg.upsample_type in ('deconv', 'nearest', 'bilinear')
+        from ..nn.blocks import deconv2d_block
+        for i in range(len(self.cfg.upsample_dims)):
+            activation = None if i == len(self.cfg.upsample_dims) - 1 \
+                else build_activation(self.cfg.activation)
+            if self.cfg.upsample_type == 'deconv':
+                self.upsample.append(deconv2d_block(dims[i], dims[i + 1], 4, 2, 1,
+                                                    activation=activation, norm_type=None))
+            else:
+                self.upsample.append(conv2d_block(dims[i], dims[i + 1], 3, 1, 1,
+                                                  activation=activation, norm_type=None))
+
+    def forward(self, embedding, map_skip: List[Tensor], location: Optional[Tensor] = None):
+        spatial_y = self.whole_cfg.model.spatial_y
+        spatial_x = self.whole_cfg.model.spatial_x
+        projected = self.project_embed(embedding)
+        reshaped = projected.reshape(projected.shape[0], self.reshape_channel,
+                                     spatial_y // 8, spatial_x // 8)
+        cat_feature = torch.cat([reshaped, map_skip[-1]], dim=1)
+        x = self.conv1(self.act(cat_feature))
+        for i in range(len(self.res)):
+            x = x + map_skip[len(map_skip) - i - 1]
+            x = self.res[i](x, x) if self.use_gate else self.res[i](x)
+        for i, layer in enumerate(self.upsample):
+            if self.cfg.upsample_type == 'nearest':
+                x = F.interpolate(x, scale_factor=2., mode='nearest')
+            elif self.cfg.upsample_type == 'bilinear':
+                x = F.interpolate(x, scale_factor=2., mode='bilinear')
+            if self.use_unet:
+                x = x + map_skip[len(map_skip) - len(self.res) - i - 1]
+            x = layer(x)
+        logits_flat = x.reshape(x.shape[0], -1) / self.whole_cfg.model.temperature
+        if location is None:
+            p = F.softmax(logits_flat, dim=1)
+            location = torch.multinomial(p, 1)[:, 0]
+        return logits_flat, location
