@@ -1,0 +1,1 @@
+from .fake_dataloader import FakeSLDataloader, FakeRLDataloader

@@ -173,6 +173,179 @@ def fake_rl_learner_data(batch_size=4, unroll_len=16, entity_num_range=(64, 256)
     return rl_collate(trajs)
 
 
+def _batched_obs_tensors(n, entity_width, seed):
+    """Vectorized synthetic obs: batched tensors built directly (no per-step
+    Python loop) — same shapes/dtypes/vocabularies as fake_step_data."""
+    from .consts import SPATIAL_INFO, SCALAR_INFO, ENTITY_INFO, EFFECT_LEN
+    g = torch.Generator().manual_seed(seed)
+
+    def ri(high, size, dtype):
+        return torch.randint(0, max(int(high), 1), size=size, dtype=dtype, generator=g)
+
+    spatial_info = {}
+    for k, dtype in SPATIAL_INFO:
+        if 'effect' in k:
+            spatial_info[k] = ri(SPATIAL_SIZE[0] * SPATIAL_SIZE[1], (n, EFFECT_LEN), dtype)
+        else:
+            high = {'height_map': 256, 'visibility_map': 4, 'player_relative': 5}.get(k, 2)
+            spatial_info[k] = ri(high, (n, *SPATIAL_SIZE), dtype)
+    scalar_info = {}
+    for k, dtype, size in SCALAR_INFO:
+        if k == 'time':
+            scalar_info[k] = torch.rand((n,), generator=g) * 1000
+        elif k == 'agent_statistics':
+            scalar_info[k] = torch.rand((n, *size), generator=g) * 10
+        elif k == 'beginning_order':
+            scalar_info[k] = ri(174, (n, *size), dtype)
+        else:
+            high = {'home_race': 5, 'away_race': 5, 'last_queued': 2,
+                    'last_delay': MAX_DELAY + 1, 'last_action_type': NUM_ACTIONS,
+                    'bo_location': SPATIAL_SIZE[0] * SPATIAL_SIZE[1]}.get(k, 2)
+            scalar_info[k] = ri(high, (n, *size), dtype)
+    entity_info = {}
+    ENT_HIGH = {'unit_type': 260, 'alliance': 5, 'cargo_space_taken': 9,
+                'display_type': 5, 'x': SPATIAL_SIZE[1], 'y': SPATIAL_SIZE[0],
+                'cloak': 5, 'cargo_space_max': 9, 'assigned_harvesters': 24,
+                'weapon_cooldown': 32, 'order_length': 9, 'order_id_0': NUM_ACTIONS,
+                'order_id_1': 49, 'buff_id_0': 50, 'buff_id_1': 50,
+                'addon_unit_type': 9, 'order_id_2': 49, 'order_id_3': 49,
+                'attack_upgrade_level': 4, 'armor_upgrade_level': 4,
+                'shield_upgrade_level': 4}
+    for k, dtype in ENTITY_INFO:
+        if dtype in (torch.float16, torch.float32):
+            entity_info[k] = torch.rand((n, entity_width), generator=g).to(dtype)
+        else:
+            entity_info[k] = ri(ENT_HIGH.get(k, 2), (n, entity_width), dtype)
+    return spatial_info, scalar_info, entity_info, g
+
+
+def fake_sl_batch_fast(batch_size=32, traj_len=64, entity_min=100, seed=0):
+    """Vectorized SL batch at the reference SL layout: fixed 512-entity slab
+    width (the reference's shared-memory slab always runs the transformer on
+    512 entities with a mask), (B*T) lane-major rows."""
+    n = batch_size * traj_len
+    spatial_info, scalar_info, entity_info, g = _batched_obs_tensors(
+        n, MAX_ENTITY_NUM, seed)
+    entity_num = torch.randint(entity_min, MAX_ENTITY_NUM, (n,), generator=g)
+    su_num = torch.minimum(
+        torch.randint(1, MAX_SELECTED_UNITS_NUM, (n,), generator=g), entity_num)
+    # distinct selections per row, end token at position su_num-1
+    sel = torch.rand(n, MAX_ENTITY_NUM, generator=g).argsort(dim=1)
+    sel = sel % entity_num.unsqueeze(1)
+    # argsort of random is a permutation of [0,512); mod en breaks distinctness,
+    # so instead take a permutation restricted per-row via argsort of masked rand
+    r = torch.rand(n, MAX_ENTITY_NUM, generator=g)
+    r[torch.arange(MAX_ENTITY_NUM).unsqueeze(0) >= entity_num.unsqueeze(1)] = 2.0
+    sel = r.argsort(dim=1)[:, :MAX_SELECTED_UNITS_NUM]
+    sel.scatter_(1, (su_num - 1).unsqueeze(1), entity_num.unsqueeze(1))
+    action_info = {
+        'action_type': torch.randint(0, NUM_ACTIONS, (n,), generator=g),
+        'delay': torch.randint(0, MAX_DELAY + 1, (n,), generator=g),
+        'queued': torch.randint(0, 2, (n,), generator=g),
+        'selected_units': sel.long(),
+        'target_unit': (torch.randint(0, 1 << 30, (n,), generator=g) % entity_num).long(),
+        'target_location': torch.randint(0, SPATIAL_SIZE[0] * SPATIAL_SIZE[1], (n,),
+                                         generator=g),
+    }
+    action_mask = {k: torch.ones(n, dtype=torch.bool) for k in ACTION_INFO}
+    return {
+        'spatial_info': spatial_info, 'scalar_info': scalar_info,
+        'entity_info': entity_info, 'entity_num': entity_num,
+        'action_info': action_info, 'action_mask': action_mask,
+        'selected_units_num': su_num,
+        'traj_lens': [traj_len] * batch_size,
+        'new_episodes': [False] * batch_size,
+    }
+
+
+def fake_rl_learner_data_fast(batch_size=4, unroll_len=16, entity_num=256, seed=0):
+    """Vectorized RL learner batch (uniform entity width = batch-max padding
+    outcome), time-major obs over (T+1)*B rows like the RL collate."""
+    T, B, EN = unroll_len, batch_size, entity_num
+    n = (T + 1) * B
+    spatial_info, scalar_info, entity_info, g = _batched_obs_tensors(n, EN, seed)
+    entity_nums = torch.randint(EN // 2, EN, (n,), generator=g)
+    su_num = torch.minimum(
+        torch.randint(1, MAX_SELECTED_UNITS_NUM, (T, B), generator=g),
+        entity_nums[:T * B].view(T, B))
+    r = torch.rand(T * B, EN, generator=g)
+    en_flat = entity_nums[:T * B]
+    r[torch.arange(EN).unsqueeze(0) >= en_flat.unsqueeze(1)] = 2.0
+    sel = r.argsort(dim=1)[:, :MAX_SELECTED_UNITS_NUM]
+    sel.scatter_(1, (su_num.view(-1) - 1).unsqueeze(1), en_flat.unsqueeze(1))
+    sel = sel.view(T, B, MAX_SELECTED_UNITS_NUM).long()
+    action_info = {
+        'action_type': torch.randint(0, NUM_ACTIONS, (T, B), generator=g),
+        'delay': torch.randint(0, MAX_DELAY + 1, (T, B), generator=g),
+        'queued': torch.randint(0, 2, (T, B), generator=g),
+        'selected_units': sel,
+        'target_unit': (torch.randint(0, 1 << 30, (T, B), generator=g)
+                        % en_flat.view(T, B)).long(),
+        'target_location': torch.randint(0, SPATIAL_SIZE[0] * SPATIAL_SIZE[1], (T, B),
+                                         generator=g),
+    }
+    su_mask = sequence_mask(su_num.view(-1), max_len=MAX_SELECTED_UNITS_NUM).view(
+        T, B, MAX_SELECTED_UNITS_NUM)
+    behaviour_logp = {
+        'action_type': -torch.rand(T, B), 'delay': -torch.rand(T, B),
+        'queued': -torch.rand(T, B),
+        'selected_units': torch.where(su_mask, -torch.rand(T, B, MAX_SELECTED_UNITS_NUM),
+                                      torch.full((T, B, MAX_SELECTED_UNITS_NUM), -1e9)),
+        'target_unit': -torch.rand(T, B), 'target_location': -torch.rand(T, B),
+    }
+    # teacher SU logits with the availability/prev-selected mask structure
+    su_logit = torch.randn(T, B, MAX_SELECTED_UNITS_NUM, EN + 1, generator=g)
+    avail = sequence_mask(en_flat + 1, max_len=EN + 1).view(T, B, 1, EN + 1)
+    step_idx = torch.arange(MAX_SELECTED_UNITS_NUM).view(1, 1, -1, 1)
+    valid_step = su_mask.unsqueeze(-1)
+    # previously-selected mask: labels before step s disabled
+    onehot = torch.zeros(T * B, MAX_SELECTED_UNITS_NUM, EN + 1, dtype=torch.int16)
+    onehot.scatter_(2, sel.view(T * B, -1, 1), 1)
+    prev = torch.cat([onehot.new_zeros(T * B, 1, EN + 1),
+                      onehot.cumsum(1)[:, :-1]], dim=1) > 0
+    mask_full = avail & ~prev.view(T, B, MAX_SELECTED_UNITS_NUM, EN + 1) & valid_step
+    mask_full[:, :, 0].scatter_(-1, en_flat.view(T, B, 1), False)
+    su_logit = su_logit.masked_fill(~mask_full, -1e9)
+    tu_logit = torch.randn(T, B, EN, generator=g).masked_fill(
+        ~sequence_mask(en_flat, max_len=EN).view(T, B, EN), -1e9)
+    teacher_logit = {
+        'action_type': torch.randn(T, B, NUM_ACTIONS, generator=g),
+        'delay': torch.randn(T, B, MAX_DELAY + 1, generator=g),
+        'queued': torch.randn(T, B, 2, generator=g),
+        'selected_units': su_logit, 'target_unit': tu_logit,
+        'target_location': torch.randn(T, B, SPATIAL_SIZE[0] * SPATIAL_SIZE[1],
+                                       generator=g),
+    }
+    mask = {
+        'actions_mask': {k: torch.ones(T, B, dtype=torch.long)
+                         for k in ('queued', 'selected_units', 'target_location',
+                                   'target_unit')},
+        'cum_action_mask': torch.ones(T, B), 'build_order_mask': torch.ones(T, B),
+        'built_unit_mask': torch.ones(T, B),
+        'selected_units_mask': su_mask,
+        'selected_units_logits_mask': sequence_mask(en_flat + 1, max_len=EN + 1
+                                                    ).view(T, B, EN + 1),
+        'target_units_logits_mask': sequence_mask(en_flat, max_len=EN).view(T, B, EN),
+    }
+    reward = {
+        'winloss': torch.randint(-1, 2, (T, B), generator=g).float(),
+        'build_order': torch.rand(T, B, generator=g) * 2 - 1,
+        'built_unit': torch.rand(T, B, generator=g) * 2 - 1,
+        'battle': torch.rand(T, B, generator=g) * 2 - 1,
+    }
+    return {
+        'spatial_info': spatial_info, 'scalar_info': scalar_info,
+        'entity_info': entity_info, 'entity_num': entity_nums,
+        'hidden_state': [(torch.zeros(n, 384), torch.zeros(n, 384)) for _ in range(3)],
+        'action_info': action_info, 'selected_units_num': su_num,
+        'behaviour_logp': behaviour_logp, 'teacher_logit': teacher_logit,
+        'mask': mask, 'reward': reward,
+        'step': torch.randint(0, 10000, (T, B), generator=g).float(),
+        'model_last_iter': torch.zeros(B),
+        'batch_size': B, 'unroll_len': T,
+    }
+
+
 def fake_sl_batch(batch_size=4, traj_len=8, entity_num=None, seed=0):
     """SL batch: (B*T) lane-major rows + traj_lens/new_episodes
     (reference `sl_dataloader.py:__next__` / FakeDataloader)."""

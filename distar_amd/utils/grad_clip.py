@@ -1,0 +1,99 @@
+"""Gradient clipping strategies (reference `ctools/torch_utils/grad_clip.py:19-151`):
+none / max_norm (EMA-normalized) / momentum_norm (per-param EMA) /
+clip_value (second-moment clamp) / clip_const / pytorch_norm.
+
+On GPU the norm computations use torch's fused multi-tensor paths
+(`torch.nn.utils.clip_grad_norm_` with foreach=True) — one kernel per dtype
+group rather than a per-parameter loop.
+"""
+import torch
+
+
+class GradClip:
+    def __init__(self, clip_type, threshold=1.0, norm_type=2, begin_step=200,
+                 ignore_threshold=3.0):
+        assert clip_type in ('max_norm', 'clip_value', 'none', 'clip_const',
+                             'pytorch_norm', 'momentum_norm')
+        self.clip_type = clip_type
+        self.threshold = threshold
+        self.norm_type = norm_type
+        self.begin_step = begin_step
+        self.ignore_threshold = ignore_threshold
+        self.step = 0
+        self._ema = None
+        self._state = {}
+
+    def apply(self, parameters):
+        params = [p for p in parameters if p.grad is not None]
+        if not params:
+            return 0.
+        self.step += 1
+        if self.clip_type == 'none':
+            with torch.no_grad():
+                total = torch.norm(torch.stack(
+                    [torch.norm(p.grad.detach(), self.norm_type) for p in params]),
+                    self.norm_type)
+            return total.item()
+        if self.clip_type == 'pytorch_norm':
+            total = torch.nn.utils.clip_grad_norm_(params, self.threshold,
+                                                   norm_type=self.norm_type,
+                                                   foreach=True)
+            return total.item()
+        if self.clip_type == 'clip_const':
+            with torch.no_grad():
+                torch._foreach_clamp_(
+                    [p.grad for p in params], -self.threshold, self.threshold)
+                total = torch.norm(torch.stack(
+                    [torch.norm(p.grad, self.norm_type) for p in params]),
+                    self.norm_type)
+            return total.item()
+        if self.clip_type == 'max_norm':
+            # clip against an EMA of the global grad norm
+            with torch.no_grad():
+                total = torch.norm(torch.stack(
+                    [torch.norm(p.grad, self.norm_type) for p in params]),
+                    self.norm_type).item()
+                if self._ema is None:
+                    self._ema = total
+                clip_to = self._ema * self.threshold
+                if self.step > self.begin_step and total > clip_to:
+                    scale = clip_to / (total + 1e-6)
+                    torch._foreach_mul_([p.grad for p in params], scale)
+                self._ema = 0.99 * self._ema + 0.01 * min(total, self._ema * self.ignore_threshold)
+            return total
+        if self.clip_type == 'momentum_norm':
+            # per-parameter EMA-normalized clip (reference SL default)
+            with torch.no_grad():
+                total_sq = 0.
+                for i, p in enumerate(params):
+                    norm = torch.norm(p.grad, self.norm_type).item()
+                    total_sq += norm ** 2
+                    ema = self._state.get(i, norm)
+                    clip_to = ema * self.threshold
+                    if self.step > self.begin_step and norm > clip_to:
+                        p.grad.mul_(clip_to / (norm + 1e-6))
+                    self._state[i] = 0.99 * ema + 0.01 * min(norm, ema * self.ignore_threshold)
+            return total_sq ** 0.5
+        if self.clip_type == 'clip_value':
+            # Adam-like second-moment clamp
+            with torch.no_grad():
+                total_sq = 0.
+                for i, p in enumerate(params):
+                    state = self._state.setdefault(i, torch.zeros_like(p.grad))
+                    state.mul_(0.999).addcmul_(p.grad, p.grad, value=0.001)
+                    bound = state.sqrt() * self.threshold + 1e-8
+                    p.grad.clamp_(-1e9, 1e9)
+                    if self.step > self.begin_step:
+                        torch.minimum(p.grad, bound, out=p.grad)
+                        torch.maximum(p.grad, -bound, out=p.grad)
+                    total_sq += torch.norm(p.grad, self.norm_type).item() ** 2
+            return total_sq ** 0.5
+        raise KeyError(self.clip_type)
+
+
+def build_grad_clip(cfg):
+    return GradClip(cfg.get('type', 'pytorch_norm'),
+                    threshold=cfg.get('threshold', 1.0),
+                    norm_type=cfg.get('norm_type', 2),
+                    begin_step=cfg.get('begin_step', 200),
+                    ignore_threshold=cfg.get('ignore_threshold', 3.0))
