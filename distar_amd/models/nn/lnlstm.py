@@ -59,6 +59,11 @@ class LSTMLayer(nn.Module):
         self.cell = cell(*cell_args)
 
     def forward(self, input: Tensor, state: Tuple[Tensor, Tensor]) -> Tuple[Tensor, Tuple[Tensor, Tensor]]:
+        import os
+        if input.is_cuda and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1':
+            # fused HIP path: ONE kernel for the whole T-step unroll
+            from ...ops.lnlstm import fused_lnlstm_layer
+            return fused_lnlstm_layer(input, state, self.cell)
         T, B = input.shape[0], input.shape[1]
         cell = self.cell
         # One big GEMM + LN for the input half of all T steps.

@@ -1,6 +1,7 @@
 // Python bindings for the distar_amd HIP/CDNA4 kernels (_hip_ops).
 #include <torch/extension.h>
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 #include <c10/hip/HIPStream.h>
 
 extern "C" __global__ void lambda_return_kernel(
@@ -8,6 +9,18 @@ extern "C" __global__ void lambda_return_kernel(
 extern "C" __global__ void vtrace_kernel(
     const float*, const float*, const float*, const float*, const float*,
     const float*, float*, int, int);
+
+using bf16_t = __hip_bfloat16;
+extern "C" __global__ void lnlstm_forward_kernel(
+    const float*, const float*, const float*, const bf16_t*,
+    const float*, const float*, const float*, const float*,
+    float*, float*, float*, float*, int, int, int);
+extern "C" __global__ void lnlstm_backward_kernel(
+    const float*, const float*, const float*,
+    const float*, const float*, const float*, const float*, const float*,
+    const bf16_t*, const float*, const float*, const float*, const float*,
+    float*, float*, float*, float*, float*, float*, float*, float*,
+    int, int, int);
 
 namespace {
 
@@ -57,10 +70,84 @@ torch::Tensor vtrace_scan(torch::Tensor clipped_rhos, torch::Tensor clipped_cs,
   return out;
 }
 
+constexpr int kNT = 512;   // must match NT in lnlstm.hip
+
+std::vector<torch::Tensor> lnlstm_forward(
+    torch::Tensor igates, torch::Tensor h0, torch::Tensor c0,
+    torch::Tensor w_hh_bf16, torch::Tensor lnh_w, torch::Tensor lnh_b,
+    torch::Tensor lnc_w, torch::Tensor lnc_b) {
+  check_2d(igates, "igates");
+  TORCH_CHECK(w_hh_bf16.scalar_type() == torch::kBFloat16, "w_hh must be bf16");
+  TORCH_CHECK(w_hh_bf16.is_contiguous());
+  int64_t T = igates.size(0), B = igates.size(1), G = igates.size(2);
+  int64_t H = G / 4;
+  TORCH_CHECK(H <= kNT && G <= 4 * kNT, "hidden size too large for kernel");
+  auto opt = igates.options();
+  auto h_all = torch::empty({T + 1, B, H}, opt);
+  auto c_all = torch::empty({T + 1, B, H}, opt);
+  auto hgates_raw = torch::empty({T, B, G}, opt);
+  auto cellraw = torch::empty({T, B, H}, opt);
+  size_t lds = sizeof(float) * ((H + 1) / 2 + G + H + kNT);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(lnlstm_forward_kernel, dim3(B), dim3(kNT), lds,
+                     stream.stream(),
+                     igates.data_ptr<float>(), h0.data_ptr<float>(),
+                     c0.data_ptr<float>(),
+                     reinterpret_cast<const __hip_bfloat16*>(w_hh_bf16.data_ptr()),
+                     lnh_w.data_ptr<float>(), lnh_b.data_ptr<float>(),
+                     lnc_w.data_ptr<float>(), lnc_b.data_ptr<float>(),
+                     h_all.data_ptr<float>(), c_all.data_ptr<float>(),
+                     hgates_raw.data_ptr<float>(), cellraw.data_ptr<float>(),
+                     (int)T, (int)B, (int)H);
+  return {h_all, c_all, hgates_raw, cellraw};
+}
+
+std::vector<torch::Tensor> lnlstm_backward(
+    torch::Tensor dout, torch::Tensor dhT, torch::Tensor dcT,
+    torch::Tensor igates, torch::Tensor h_all, torch::Tensor c_all,
+    torch::Tensor hgates_raw, torch::Tensor cellraw, torch::Tensor w_hh_t_bf16,
+    torch::Tensor lnh_w, torch::Tensor lnh_b, torch::Tensor lnc_w,
+    torch::Tensor lnc_b) {
+  check_2d(dout, "dout");
+  TORCH_CHECK(w_hh_t_bf16.scalar_type() == torch::kBFloat16);
+  int64_t T = igates.size(0), B = igates.size(1), G = igates.size(2);
+  int64_t H = G / 4;
+  auto opt = igates.options();
+  auto digates = torch::empty({T, B, G}, opt);
+  auto dhgates_raw = torch::empty({T, B, G}, opt);
+  auto dh0 = torch::empty({B, H}, opt);
+  auto dc0 = torch::empty({B, H}, opt);
+  auto dlnh_w = torch::zeros({G}, opt);
+  auto dlnh_b = torch::zeros({G}, opt);
+  auto dlnc_w = torch::zeros({H}, opt);
+  auto dlnc_b = torch::zeros({H}, opt);
+  size_t lds = sizeof(float) * (3 * G + 2 * H + kNT);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(lnlstm_backward_kernel, dim3(B), dim3(kNT), lds,
+                     stream.stream(),
+                     dout.data_ptr<float>(),
+                     (dhT.defined() && dhT.numel()) ? dhT.data_ptr<float>() : nullptr,
+                     (dcT.defined() && dcT.numel()) ? dcT.data_ptr<float>() : nullptr,
+                     igates.data_ptr<float>(), h_all.data_ptr<float>(),
+                     c_all.data_ptr<float>(), hgates_raw.data_ptr<float>(),
+                     cellraw.data_ptr<float>(),
+                     reinterpret_cast<const __hip_bfloat16*>(w_hh_t_bf16.data_ptr()),
+                     lnh_w.data_ptr<float>(), lnh_b.data_ptr<float>(),
+                     lnc_w.data_ptr<float>(), lnc_b.data_ptr<float>(),
+                     digates.data_ptr<float>(), dhgates_raw.data_ptr<float>(),
+                     dh0.data_ptr<float>(), dc0.data_ptr<float>(),
+                     dlnh_w.data_ptr<float>(), dlnh_b.data_ptr<float>(),
+                     dlnc_w.data_ptr<float>(), dlnc_b.data_ptr<float>(),
+                     (int)T, (int)B, (int)H);
+  return {digates, dhgates_raw, dh0, dc0, dlnh_w, dlnh_b, dlnc_w, dlnc_b};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lambda_return_scan", &lambda_return_scan,
         "generalized lambda-return reverse scan (T,B)");
   m.def("vtrace_scan", &vtrace_scan, "v-trace corrected-value reverse scan");
+  m.def("lnlstm_forward", &lnlstm_forward, "fused LN-LSTM layer forward");
+  m.def("lnlstm_backward", &lnlstm_backward, "fused LN-LSTM layer backward");
 }

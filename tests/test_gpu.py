@@ -91,3 +91,57 @@ def test_rl_train_step_on_gpu():
     ld['total_loss'].backward()
     assert all(torch.isfinite(p.grad).all() for p in model.parameters()
                if p.grad is not None)
+
+
+def test_fused_lnlstm_matches_eager():
+    """Fused HIP LN-LSTM layer vs the fp32 eager reference cell: forward
+    outputs and all gradients (bf16 W_hh products => 1e-2-level tolerance)."""
+    import os
+    from distar_amd.models.nn.lnlstm import script_lnlstm
+    torch.manual_seed(0)
+    T, B, IN, H = 7, 4, 64, 384
+    lstm = script_lnlstm(IN, H, 2).cuda()
+    x = torch.randn(T, B, IN, device='cuda')
+    states = [(torch.randn(B, H, device='cuda'), torch.randn(B, H, device='cuda'))
+              for _ in range(2)]
+
+    def run():
+        xx = x.clone().requires_grad_(True)
+        out, out_states = lstm(xx, [(h.clone(), c.clone()) for h, c in states])
+        loss = (out.float() ** 2).mean() + sum(
+            (s[0].float() ** 2).mean() + (s[1].float() ** 2).mean()
+            for s in out_states)
+        lstm.zero_grad()
+        loss.backward()
+        grads = {n: p.grad.clone() for n, p in lstm.named_parameters()}
+        return out.detach().float(), grads, xx.grad.clone()
+
+    out_hip, grads_hip, xgrad_hip = run()
+    os.environ['DISTAR_AMD_DISABLE_HIP'] = '1'
+    try:
+        out_ref, grads_ref, xgrad_ref = run()
+    finally:
+        del os.environ['DISTAR_AMD_DISABLE_HIP']
+    torch.testing.assert_close(out_hip, out_ref, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(xgrad_hip, xgrad_ref, rtol=5e-2, atol=5e-2)
+    for n in grads_ref:
+        torch.testing.assert_close(grads_hip[n], grads_ref[n], rtol=5e-2, atol=5e-2,
+                                   msg=lambda m: f'{n}: {m}')
+
+
+def test_fused_lnlstm_small_hidden():
+    """The selected-units pointer LSTM shape (H=32) through the same kernel."""
+    import os
+    from distar_amd.models.nn.lnlstm import script_lnlstm
+    torch.manual_seed(1)
+    T, B, IN, H = 64, 128, 32, 32
+    lstm = script_lnlstm(IN, H, 1).cuda()
+    x = torch.randn(T, B, IN, device='cuda')
+    st = [(torch.zeros(B, H, device='cuda'), torch.zeros(B, H, device='cuda'))]
+    out, _ = lstm(x, [(h.clone(), c.clone()) for h, c in st])
+    os.environ['DISTAR_AMD_DISABLE_HIP'] = '1'
+    try:
+        ref, _ = lstm(x, [(h.clone(), c.clone()) for h, c in st])
+    finally:
+        del os.environ['DISTAR_AMD_DISABLE_HIP']
+    torch.testing.assert_close(out.float(), ref.float(), rtol=2e-2, atol=2e-2)
