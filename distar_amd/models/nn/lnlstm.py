@@ -60,7 +60,14 @@ class LSTMLayer(nn.Module):
 
     def forward(self, input: Tensor, state: Tuple[Tensor, Tensor]) -> Tuple[Tensor, Tuple[Tensor, Tensor]]:
         import os
-        if input.is_cuda and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1':
+        use_fused = input.is_cuda and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'
+        if use_fused and self.cell.hidden_size >= 128 \
+                and os.environ.get('DISTAR_AMD_NO_FUSED_CORE') == '1':
+            use_fused = False
+        if use_fused and self.cell.hidden_size < 128 \
+                and os.environ.get('DISTAR_AMD_NO_FUSED_SU') == '1':
+            use_fused = False
+        if use_fused:
             # fused HIP path: ONE kernel for the whole T-step unroll
             from ...ops.lnlstm import fused_lnlstm_layer
             return fused_lnlstm_layer(input, state, self.cell)
