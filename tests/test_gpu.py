@@ -130,8 +130,12 @@ def test_fused_lnlstm_matches_eager():
 
 
 def test_fused_lnlstm_small_hidden():
-    """The selected-units pointer LSTM shape (H=32) through the same kernel."""
-    import os
+    """The selected-units pointer LSTM shape (H=32) through the same kernel.
+
+    The eager reference here uses the SAME bf16-rounded h/W products as the
+    kernel — over a 64-step recurrence, plain-fp32-vs-bf16 divergence is
+    chaotic and elementwise bounds are meaningless (observed 4% mismatch),
+    so the comparison has to share the rounding."""
     from distar_amd.models.nn.lnlstm import script_lnlstm
     torch.manual_seed(1)
     T, B, IN, H = 64, 128, 32, 32
@@ -139,9 +143,22 @@ def test_fused_lnlstm_small_hidden():
     x = torch.randn(T, B, IN, device='cuda')
     st = [(torch.zeros(B, H, device='cuda'), torch.zeros(B, H, device='cuda'))]
     out, _ = lstm(x, [(h.clone(), c.clone()) for h, c in st])
-    os.environ['DISTAR_AMD_DISABLE_HIP'] = '1'
-    try:
-        ref, _ = lstm(x, [(h.clone(), c.clone()) for h, c in st])
-    finally:
-        del os.environ['DISTAR_AMD_DISABLE_HIP']
-    torch.testing.assert_close(out.float(), ref.float(), rtol=2e-2, atol=2e-2)
+
+    cell = lstm.layers[0].cell
+    with torch.no_grad():
+        igates = cell.layernorm_i(x.reshape(T * B, -1).float()
+                                  .mm(cell.weight_ih.t().float())).view(T, B, -1)
+        w_bf = cell.weight_hh.bfloat16().float()
+        h, c = st[0][0].clone(), st[0][1].clone()
+        refs = []
+        for t in range(T):
+            hg = h.bfloat16().float().mm(w_bf.t())
+            gates = igates[t] + cell.layernorm_h(hg)
+            i, f, g, o = gates.chunk(4, 1)
+            i, f, o = torch.sigmoid(i), torch.sigmoid(f), torch.sigmoid(o)
+            g = torch.tanh(g)
+            c = cell.layernorm_c(f * c + i * g)
+            h = o * torch.tanh(c)
+            refs.append(h)
+        ref = torch.stack(refs)
+    torch.testing.assert_close(out.float(), ref, rtol=2e-3, atol=2e-3)
