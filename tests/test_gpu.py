@@ -140,6 +140,14 @@ def test_fused_lnlstm_small_hidden():
     torch.manual_seed(1)
     T, B, IN, H = 64, 128, 32, 32
     lstm = script_lnlstm(IN, H, 1).cuda()
+    with torch.no_grad():
+        # default randn init gives |W| ~ 1: an expansive recurrence where
+        # fp32 accumulation-order noise grows ~2x per step and elementwise
+        # comparison over 64 steps is meaningless.  Scale to contractive
+        # dynamics (what trained weights look like).
+        for cell in (l.cell for l in lstm.layers):
+            cell.weight_hh.mul_(0.05)
+            cell.weight_ih.mul_(0.05)
     x = torch.randn(T, B, IN, device='cuda')
     st = [(torch.zeros(B, H, device='cuda'), torch.zeros(B, H, device='cuda'))]
     out, _ = lstm(x, [(h.clone(), c.clone()) for h, c in st])
