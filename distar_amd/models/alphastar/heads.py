@@ -379,11 +379,12 @@ class LocationHead(nn.Module):
         for i in range(len(self.res)):
             x = x + map_skip[len(map_skip) - i - 1]
             x = self.res[i](x, x) if self.use_gate else self.res[i](x)
+        from ...ops.upsample import upsample2x_bilinear
         for i, layer in enumerate(self.upsample):
             if self.cfg.upsample_type == 'nearest':
                 x = F.interpolate(x, scale_factor=2., mode='nearest')
             elif self.cfg.upsample_type == 'bilinear':
-                x = F.interpolate(x, scale_factor=2., mode='bilinear')
+                x = upsample2x_bilinear(x)
             if self.use_unet:
                 x = x + map_skip[len(map_skip) - len(self.res) - i - 1]
             x = layer(x)

@@ -70,6 +70,60 @@ torch::Tensor vtrace_scan(torch::Tensor clipped_rhos, torch::Tensor clipped_cs,
   return out;
 }
 
+extern "C" __global__ void upsample2x_fwd_f32(const float*, float*, int, int, int);
+extern "C" __global__ void upsample2x_fwd_bf16(const __hip_bfloat16*, __hip_bfloat16*, int, int, int);
+extern "C" __global__ void upsample2x_bwd_f32(const float*, float*, int, int, int);
+extern "C" __global__ void upsample2x_bwd_bf16(const __hip_bfloat16*, __hip_bfloat16*, int, int, int);
+
+torch::Tensor upsample2x(torch::Tensor input) {
+  TORCH_CHECK(input.is_cuda() && input.is_contiguous() && input.dim() == 4);
+  int64_t N = input.size(0), C = input.size(1), H = input.size(2), W = input.size(3);
+  auto out = torch::empty({N, C, H * 2, W * 2}, input.options());
+  long total = N * C * H * 2 * W * 2;
+  int threads = 256;
+  int blocks = std::min<long>((total + threads - 1) / threads, 8192);
+  auto stream = c10::hip::getCurrentHIPStream();
+  if (input.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(upsample2x_fwd_f32, dim3(blocks), dim3(threads), 0,
+                       stream.stream(), input.data_ptr<float>(),
+                       out.data_ptr<float>(), (int)(N * C), (int)H, (int)W);
+  } else if (input.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(upsample2x_fwd_bf16, dim3(blocks), dim3(threads), 0,
+                       stream.stream(),
+                       reinterpret_cast<const __hip_bfloat16*>(input.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                       (int)(N * C), (int)H, (int)W);
+  } else {
+    TORCH_CHECK(false, "upsample2x: fp32/bf16 only");
+  }
+  return out;
+}
+
+torch::Tensor upsample2x_backward(torch::Tensor gout) {
+  TORCH_CHECK(gout.is_cuda() && gout.is_contiguous() && gout.dim() == 4);
+  int64_t N = gout.size(0), C = gout.size(1), H2 = gout.size(2), W2 = gout.size(3);
+  int64_t H = H2 / 2, W = W2 / 2;
+  auto gin = torch::empty({N, C, H, W}, gout.options());
+  long total = N * C * H * W;
+  int threads = 256;
+  int blocks = std::min<long>((total + threads - 1) / threads, 8192);
+  auto stream = c10::hip::getCurrentHIPStream();
+  if (gout.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(upsample2x_bwd_f32, dim3(blocks), dim3(threads), 0,
+                       stream.stream(), gout.data_ptr<float>(),
+                       gin.data_ptr<float>(), (int)(N * C), (int)H, (int)W);
+  } else if (gout.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(upsample2x_bwd_bf16, dim3(blocks), dim3(threads), 0,
+                       stream.stream(),
+                       reinterpret_cast<const __hip_bfloat16*>(gout.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(gin.data_ptr()),
+                       (int)(N * C), (int)H, (int)W);
+  } else {
+    TORCH_CHECK(false, "upsample2x_backward: fp32/bf16 only");
+  }
+  return gin;
+}
+
 constexpr int kNT = 512;   // must match NT in lnlstm.hip
 
 std::vector<torch::Tensor> lnlstm_forward(
@@ -150,4 +204,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("vtrace_scan", &vtrace_scan, "v-trace corrected-value reverse scan");
   m.def("lnlstm_forward", &lnlstm_forward, "fused LN-LSTM layer forward");
   m.def("lnlstm_backward", &lnlstm_backward, "fused LN-LSTM layer backward");
+  m.def("upsample2x", &upsample2x, "bilinear 2x upsample (align_corners=false)");
+  m.def("upsample2x_backward", &upsample2x_backward, "bilinear 2x upsample backward");
 }

@@ -1,0 +1,112 @@
+// Bilinear 2x upsample (align_corners=false), forward + backward — the
+// LocationHead's 3-stage upsample chain.
+//
+// PyTorch's generic NCHW bilinear kernel ran fp32/scalar and was 29% of the
+// whole SL step (profiles/r01_notes.md).  For the fixed scale-2,
+// align_corners=false case every output pixel is a 4-tap filter with weights
+// from {9/16, 3/16, 1/16} and source indices derivable with shifts:
+//   src = (dst + 0.5)/2 - 0.5  ->  x0 = (dst-1)>>1, frac in {0.25, 0.75}.
+//
+// Memory-bound: one thread per output pixel over (N*C, H2, W2) with
+// row-contiguous coalesced access; fp32 math, fp32 or bf16 storage.
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+using bf16 = __hip_bfloat16;
+
+template <typename T>
+__device__ inline float ld(const T* p);
+template <> __device__ inline float ld<float>(const float* p) { return *p; }
+template <> __device__ inline float ld<bf16>(const bf16* p) { return __bfloat162float(*p); }
+
+template <typename T>
+__device__ inline void st(T* p, float v);
+template <> __device__ inline void st<float>(float* p, float v) { *p = v; }
+template <> __device__ inline void st<bf16>(bf16* p, float v) { *p = __float2bfloat16(v); }
+
+// dst index -> (src0, src1, w0, w1) along one axis for scale-2,
+// align_corners=false, with edge clamping.
+__device__ inline void taps2x(int d, int n_src, int& s0, int& s1, float& w0) {
+  // src = (d + 0.5f) * 0.5f - 0.5f
+  float src = fmaf((float)d, 0.5f, -0.25f);
+  float floor_src = floorf(src);
+  s0 = (int)floor_src;
+  float frac = src - floor_src;          // 0.25 or 0.75
+  s1 = s0 + 1;
+  if (s0 < 0) s0 = 0;
+  if (s1 > n_src - 1) s1 = n_src - 1;
+  w0 = 1.f - frac;
+}
+
+template <typename T>
+__device__ void upsample2x_fwd(const T* __restrict__ in, T* __restrict__ out,
+                               int NC, int H, int W) {
+  const int H2 = H * 2, W2 = W * 2;
+  const long total = (long)NC * H2 * W2;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    int x = idx % W2;
+    int y = (idx / W2) % H2;
+    long nc = idx / ((long)W2 * H2);
+    int x0, x1, y0, y1;
+    float wx0, wy0;
+    taps2x(x, W, x0, x1, wx0);
+    taps2x(y, H, y0, y1, wy0);
+    const T* base = in + nc * H * W;
+    float v = wy0 * (wx0 * ld(base + y0 * W + x0) + (1.f - wx0) * ld(base + y0 * W + x1))
+        + (1.f - wy0) * (wx0 * ld(base + y1 * W + x0) + (1.f - wx0) * ld(base + y1 * W + x1));
+    st(out + idx, v);
+  }
+}
+
+// backward: each SOURCE pixel gathers from the (at most) 16 destination
+// pixels whose stencil touches it — gather, not atomics.  For scale 2 each
+// source pixel is touched by a fixed 4x4 window of outputs.
+template <typename T>
+__device__ void upsample2x_bwd(const T* __restrict__ gout, T* __restrict__ gin,
+                               int NC, int H, int W) {
+  const int H2 = H * 2, W2 = W * 2;
+  const long total = (long)NC * H * W;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    int sx = idx % W;
+    int sy = (idx / W) % H;
+    long nc = idx / ((long)W * H);
+    const T* base = gout + nc * (long)H2 * W2;
+    float acc = 0.f;
+    // destination rows/cols that can reference (sy, sx): d in [2s-1, 2s+2]
+    for (int dy = sy * 2 - 1; dy <= sy * 2 + 2; ++dy) {
+      if (dy < 0 || dy >= H2) continue;
+      int y0, y1; float wy0;
+      taps2x(dy, H, y0, y1, wy0);
+      float wy = (y0 == sy ? wy0 : 0.f) + (y1 == sy ? 1.f - wy0 : 0.f);
+      if (wy == 0.f) continue;
+      for (int dx = sx * 2 - 1; dx <= sx * 2 + 2; ++dx) {
+        if (dx < 0 || dx >= W2) continue;
+        int x0, x1; float wx0;
+        taps2x(dx, W, x0, x1, wx0);
+        float wx = (x0 == sx ? wx0 : 0.f) + (x1 == sx ? 1.f - wx0 : 0.f);
+        if (wx == 0.f) continue;
+        acc += wy * wx * ld(base + (long)dy * W2 + dx);
+      }
+    }
+    st(gin + idx, acc);
+  }
+}
+
+extern "C" __global__ void upsample2x_fwd_f32(const float* in, float* out,
+                                              int NC, int H, int W) {
+  upsample2x_fwd<float>(in, out, NC, H, W);
+}
+extern "C" __global__ void upsample2x_fwd_bf16(const bf16* in, bf16* out,
+                                               int NC, int H, int W) {
+  upsample2x_fwd<bf16>(in, out, NC, H, W);
+}
+extern "C" __global__ void upsample2x_bwd_f32(const float* gout, float* gin,
+                                              int NC, int H, int W) {
+  upsample2x_bwd<float>(gout, gin, NC, H, W);
+}
+extern "C" __global__ void upsample2x_bwd_bf16(const bf16* gout, bf16* gin,
+                                               int NC, int H, int W) {
+  upsample2x_bwd<bf16>(gout, gin, NC, H, W);
+}

@@ -22,6 +22,7 @@ ARCH = os.environ.get('PYTORCH_ROCM_ARCH', 'gfx950')
 SOURCES = [
     os.path.join(HIP_DIR, 'scans.hip'),
     os.path.join(HIP_DIR, 'lnlstm.hip'),
+    os.path.join(HIP_DIR, 'upsample.hip'),
     os.path.join(HIP_DIR, 'bindings.cpp'),
 ]
 
