@@ -1,0 +1,81 @@
+"""Adapter/Coordinator transport + serialization tests (reference:
+ctools/worker/coordinator/{adapter,coordinator}.py)."""
+import threading
+
+import pytest
+import torch
+
+from distar_amd.data.adapter import Adapter
+from distar_amd.data.coordinator import Coordinator
+from distar_amd.utils.serialize import dumps, loads
+
+
+def test_serialize_roundtrip_tensor_tree():
+    data = {'a': torch.randn(4, 5), 'b': [torch.arange(3), {'c': torch.randn(2).bfloat16()}],
+            'd': 'text', 'e': 7}
+    for fs_type in ('pickle', 'nppickle', 'torch'):
+        blob = dumps(data, fs_type=fs_type)
+        out = loads(blob, fs_type=fs_type)
+        torch.testing.assert_close(out['a'], data['a'])
+        torch.testing.assert_close(out['b'][0], data['b'][0])
+        torch.testing.assert_close(out['b'][1]['c'].float(), data['b'][1]['c'].float())
+        assert out['d'] == 'text' and out['e'] == 7
+
+
+@pytest.mark.timeout(60)
+def test_adapter_push_pull_roundtrip():
+    coord = Coordinator().run()
+    try:
+        producer = Adapter(coordinator_port=coord.port)
+        consumer = Adapter(coordinator_port=coord.port)
+        payloads = [{'step': torch.full((3,), float(i))} for i in range(4)]
+        for p in payloads:
+            producer.push(p, token='MP0traj')
+        assert producer.length('MP0traj') == 4
+        out = consumer.pull('MP0traj', size=4, worker_num=2, timeout=30)
+        assert len(out) == 4
+        got = sorted(float(d['step'][0]) for d in out)
+        assert got == [0.0, 1.0, 2.0, 3.0]
+        assert consumer.length('MP0traj') == 0
+    finally:
+        coord.close()
+
+
+@pytest.mark.timeout(60)
+def test_adapter_pull_blocks_until_data():
+    coord = Coordinator().run()
+    try:
+        producer = Adapter(coordinator_port=coord.port)
+        consumer = Adapter(coordinator_port=coord.port)
+        result = []
+
+        def late_push():
+            import time
+            time.sleep(0.5)
+            producer.push({'x': torch.ones(2)}, token='t2')
+
+        t = threading.Thread(target=late_push)
+        t.start()
+        out = consumer.pull('t2', size=1, sleep_time=0.1, timeout=30)
+        t.join()
+        assert len(out) == 1
+        torch.testing.assert_close(out[0]['x'], torch.ones(2))
+    finally:
+        coord.close()
+
+
+@pytest.mark.timeout(60)
+def test_adapter_bounded_queue_drops_oldest():
+    coord = Coordinator().run()
+    try:
+        producer = Adapter(coordinator_port=coord.port, maxlen=2)
+        consumer = Adapter(coordinator_port=coord.port)
+        for i in range(4):
+            producer.push({'i': torch.tensor([i])}, token='t3')
+        # coordinator still lists all 4 metadata entries, but only the 2
+        # newest producers hold live sockets; stale ones fail + are skipped
+        out = consumer.pull('t3', size=2, sleep_time=0.05, timeout=30)
+        vals = sorted(int(d['i'][0]) for d in out)
+        assert vals == [2, 3]
+    finally:
+        coord.close()
