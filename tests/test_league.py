@@ -1,0 +1,157 @@
+"""League control-plane tests: PFSP math, payoff warm-up, player branching,
+snapshot/reset protocol, job dispatch and the HTTP API round trip."""
+import numpy as np
+import pytest
+import torch
+
+from distar_amd.league.algorithms import pfsp
+from distar_amd.league.api import create_league_server
+from distar_amd.league.elo import ELORating, TrueSkill
+from distar_amd.league.league import League
+from distar_amd.league.payoff import Payoff
+from distar_amd.utils.config import Config
+from distar_amd.utils.http import post_json
+
+
+def test_pfsp_weightings():
+    wr = np.array([0.1, 0.5, 0.9])
+    for w in ('squared', 'variance', 'normal'):
+        p = pfsp(wr, weighting=w)
+        assert abs(p.sum() - 1) < 1e-8 and (p >= 0).all()
+    sq = pfsp(wr, 'squared')
+    assert sq[0] > sq[1] > sq[2]          # prioritize opponents we lose to
+    var = pfsp(wr, 'variance')
+    assert var[1] > var[0] and var[1] > var[2]
+    assert np.allclose(pfsp(np.zeros(3)), np.full(3, 1 / 3))
+
+
+def test_payoff_warmup_then_ema():
+    p = Payoff(decay=0.9, warm_up_size=3, min_win_rate_games=2)
+    assert p.win_rate_opponent('x') == 0.5    # no games yet
+    for r in (1, 1, 0, 1):
+        p.update('x', {'winrate': r, 'game_steps': 100, 'game_iters': 10,
+                       'game_duration': 60})
+    assert p.game_count['x'] == 4
+    assert 0.5 < p.win_rate_opponent('x') < 1.0
+
+
+def test_elo_and_trueskill():
+    elo = ELORating()
+    for _ in range(20):
+        elo.update('a', 'b', 1)
+    assert elo.ratings['a'] > elo.ratings['b']
+    ts = TrueSkill()
+    for _ in range(10):
+        ts.update('a', 'b')
+    assert ts.mu['a'] > ts.mu['b']
+    assert ts.sigma['a'] < TrueSkill.SIGMA
+
+
+def _league_cfg(tmp_path):
+    return Config({
+        'common': {'experiment_name': 'test_league'},
+        'league': {
+            'active_players': {
+                'player_id': ['MP0', 'ME0', 'EP0'],
+                'checkpoint_path': ['mp0.pth', 'me0.pth', 'ep0.pth'],
+                'pipeline': ['default', 'default', 'default'],
+                'frac_id': [1, 1, 1],
+                'z_path': ['3map.json', '3map.json', '3map.json'],
+                'z_prob': [0., 0., 0.],
+                'teacher_id': ['sl', 'sl', 'sl'],
+                'teacher_path': ['sl.pth', 'sl.pth', 'sl.pth'],
+                'one_phase_step': [1000, 1000, 1000],
+                'chosen_weight': [1.0, 1.0, 1.0],
+            },
+            'save_resume_freq': 100000,
+        },
+    })
+
+
+@pytest.fixture
+def league(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    lg = League(_league_cfg(tmp_path))
+    yield lg
+    lg.close()
+
+
+def test_league_job_dispatch_and_results(league):
+    from distar_amd.league.player import MainPlayer
+    assert set(league.active_players) == {'MP0', 'ME0', 'EP0'}
+    assert isinstance(league.active_players['MP0'], MainPlayer)
+    # register + send train info: snapshot on phase completion
+    resp = league.deal_with_register_learner({'player_id': 'MP0'})
+    assert resp['ckpt_path'] == 'mp0.pth'
+    resp = league.deal_with_learner_send_train_info(
+        {'player_id': 'MP0', 'train_steps': 1500, 'checkpoint_path': 'mp0.pth'})
+    assert resp['reset_checkpoint_path'] == 'none'   # MainPlayer never resets
+    assert any(p.parent_id == 'MP0' for p in league.historical_players.values())
+    # job dispatch now has historical players to sample
+    for _ in range(8):
+        job = league.deal_with_actor_ask_for_job({'job_type': 'train'})
+        assert len(job['player_ids']) == 2
+        assert job['env_info']['map_name'] == 'KingsCove'
+        assert set(job) >= {'checkpoint_paths', 'pipelines', 'z_path',
+                            'teacher_player_ids', 'send_data_players', 'branch'}
+    # result ingestion updates payoff + elo
+    league.deal_with_actor_send_result({
+        'game_steps': 1000, 'game_iters': 100, 'game_duration': 600,
+        '0': {'player_id': 'MP0', 'opponent_id': 'ME0', 'winloss': 1,
+              'race_id': 1, 'z_type': 0},
+        '1': {'player_id': 'ME0', 'opponent_id': 'MP0', 'winloss': -1,
+              'race_id': 1, 'z_type': 0},
+    })
+    import time
+    for _ in range(100):
+        if league.elo.game_count:
+            break
+        time.sleep(0.05)
+    assert league.elo.game_count == 1
+    assert league.active_players['MP0'].payoff.game_count['ME0'] == 1
+
+
+def test_main_exploiter_resets_to_teacher(league):
+    resp = league.deal_with_learner_send_train_info(
+        {'player_id': 'ME0', 'train_steps': 1500, 'checkpoint_path': 'me0.pth'})
+    # MainExploiter always resets after snapshot; teacher ckpt doesn't exist
+    # on disk in this test so the raw path comes back
+    assert resp['reset_checkpoint_path'] == 'sl.pth'
+
+
+def test_league_resume_roundtrip(league):
+    league.deal_with_learner_send_train_info(
+        {'player_id': 'MP0', 'train_steps': 1500, 'checkpoint_path': 'mp0.pth'})
+    path = league.save_resume()
+    cfg = _league_cfg(None)
+    cfg.league.resume_path = path
+    lg2 = League(cfg)
+    try:
+        assert set(lg2.active_players) == set(league.active_players)
+        assert set(lg2.historical_players) == set(league.historical_players)
+        assert lg2.active_players['MP0'].total_agent_step == 1500
+    finally:
+        lg2.close()
+
+
+def test_league_http_api(league):
+    server = create_league_server(league, host='127.0.0.1').start()
+    try:
+        url = f'http://127.0.0.1:{server.port}'
+        resp = post_json(url + '/league/register_learner', {'player_id': 'MP0'})
+        assert resp['ckpt_path'] == 'mp0.pth'
+        league.deal_with_learner_send_train_info(
+            {'player_id': 'MP0', 'train_steps': 1500, 'checkpoint_path': 'mp0.pth'})
+        job = post_json(url + '/league/actor_ask_for_job', {'job_type': 'train'})
+        assert 'player_ids' in job and len(job['player_ids']) == 2
+        r = post_json(url + '/league/actor_send_result', {
+            'game_steps': 10, 'game_iters': 1, 'game_duration': 5,
+            '0': {'player_id': 'MP0', 'opponent_id': 'EP0', 'winloss': 1,
+                  'race_id': 1, 'z_type': 0},
+            '1': {'player_id': 'EP0', 'opponent_id': 'MP0', 'winloss': -1,
+                  'race_id': 1, 'z_type': 0}})
+        assert r['ok']
+        elo = post_json(url + '/league/show_elo')
+        assert 'ratings' in elo
+    finally:
+        server.stop()
