@@ -1,0 +1,184 @@
+"""Rollout worker (reference `distar/actor/actor.py:23-388`):
+
+  - asks the league for a job (player ids, pipelines, checkpoints, teacher,
+    Z), loads models, spawns `env_num` environment processes,
+  - per-episode loop: agents step(obs) -> env.step(actions) -> collect_data
+    -> ActorComm.send_data; result posted at episode end,
+  - model refresh every `actor_model_update_interval` seconds, job rotation
+    after `actor_ask_for_job_interval` +/- 30%,
+  - crash-tolerant: episode exceptions close the env and continue (actors
+    are stateless cattle).
+
+Environment selection: 'mock' (synthetic spec-identical episodes; default in
+this offline image) or 'sc2' (real game through envs/env.py, gated on
+s2clientprotocol).
+"""
+import random
+import time
+import traceback
+
+import torch
+
+from .agent import Agent
+from .comm import ActorComm
+from ..envs.mock_env import MockSC2Env
+from ..utils.checkpoint import CheckpointHelper
+from ..utils.config import Config, deep_merge_dicts
+from ..utils.log import TextLogger, VariableRecord
+
+DEFAULT_ACTOR_CFG = Config({
+    'actor': {
+        'env_num': 1, 'episode_num': 2, 'job_type': 'train', 'traj_len': 16,
+        'use_cuda': False, 'env_type': 'mock',
+        'actor_model_update_interval': 10,
+        'actor_ask_for_job_interval': 3600,
+        'fake_model': False,
+    },
+    'env': {'player_num': 2, 'max_episode_steps': 64},
+    'common': {'experiment_name': 'actor_default', 'type': 'train'},
+})
+
+
+class Actor:
+    def __init__(self, cfg):
+        self._whole_cfg = deep_merge_dicts(DEFAULT_ACTOR_CFG, cfg)
+        self._cfg = self._whole_cfg.actor
+        self._job_type = self._cfg.job_type
+        self._comm = ActorComm(self._whole_cfg) \
+            if self._whole_cfg.get('communication') else None
+        self._logger = TextLogger(
+            f'experiments/{self._whole_cfg.common.experiment_name}/log', 'actor')
+        self._record = VariableRecord()
+        for var in ('agent_time', 'env_time', 'collect_time', 'episode_steps'):
+            self._record.register_var(var)
+        self._ckpt_helper = CheckpointHelper()
+        self._last_model_update = 0
+        self._end = False
+        self.episodes_done = 0
+        self.results = []
+
+    # ------------------------------------------------------------------ job
+    def _setup_job(self):
+        if self._comm is not None:
+            job = self._comm.ask_for_job(self._job_type)
+        else:
+            job = {'player_ids': ['MP0', 'MP1'],
+                   'pipelines': ['default', 'default'],
+                   'checkpoint_paths': ['none', 'none'],
+                   'teacher_checkpoint_paths': ['none', 'none'],
+                   'z_path': ['3map.json', '3map.json'],
+                   'send_data_players': ['MP0'],
+                   'update_players': ['MP0'],
+                   'env_info': {'map_name': 'KingsCove'}}
+        self._job = job
+        self._agents = []
+        for i, player_id in enumerate(job['player_ids']):
+            agent = Agent(self._whole_cfg, env_id=0)
+            agent.player_id = player_id
+            ckpt = job['checkpoint_paths'][i] if i < len(job['checkpoint_paths']) else 'none'
+            if ckpt not in ('none', None) and not self._cfg.fake_model:
+                try:
+                    self._ckpt_helper.load(ckpt, agent.model, strict=False,
+                                           logger_prints=self._logger.info)
+                except FileNotFoundError:
+                    self._logger.info(f'checkpoint missing: {ckpt}, random init')
+            self._agents.append(agent)
+        return job
+
+    def _make_env(self):
+        if self._cfg.env_type == 'sc2':
+            from ..envs.env import SC2Env
+            return SC2Env(self._whole_cfg)
+        return MockSC2Env(self._whole_cfg)
+
+    def _update_models(self):
+        if self._comm is None:
+            return
+        now = time.time()
+        if now - self._last_model_update < self._cfg.actor_model_update_interval:
+            return
+        self._last_model_update = now
+        for agent in self._agents:
+            if agent.player_id in self._job.get('update_players', []):
+                payload = self._comm.pull_model(agent.player_id, timeout=2)
+                if payload:
+                    agent.model.load_state_dict(payload['model'], strict=False)
+                    agent.set_model_last_iter(payload.get('model_last_iter', 0))
+
+    # -------------------------------------------------------------- episode
+    def _run_episode(self, env):
+        obs = env.reset()
+        for i, agent in enumerate(self._agents):
+            agent.reset(map_name=self._job['env_info'].get('map_name', 'KingsCove'),
+                        race='zerg', opponent_race='zerg', obs=obs.get(i))
+        done = False
+        episode_steps = 0
+        last_obs = obs
+        while not done and not self._end:
+            t0 = time.time()
+            actions = {}
+            for i, agent in enumerate(self._agents):
+                if i in last_obs:
+                    actions[i] = agent.step(last_obs[i])[0]
+            t1 = time.time()
+            obs, rewards, done, infos = env.step(actions)
+            t2 = time.time()
+            for i, agent in enumerate(self._agents):
+                if i not in last_obs:
+                    continue
+                traj = agent.collect_data(obs.get(i), rewards.get(i, 0), done, i)
+                if traj is not None and self._comm is not None and \
+                        agent.player_id in self._job.get('send_data_players', []):
+                    self._comm.send_data(traj, agent.player_id)
+            self._record.update_var({'agent_time': t1 - t0,
+                                     'env_time': t2 - t1,
+                                     'collect_time': time.time() - t2})
+            episode_steps += 1
+            last_obs = {**last_obs, **obs}
+            self._update_models()
+        self._record.update_var({'episode_steps': episode_steps})
+        result = self._build_result(rewards)
+        if self._comm is not None:
+            self._comm.send_result(result)
+        self.results.append(result)
+        return result
+
+    def _build_result(self, rewards):
+        result = {'game_steps': self._agents[0]._game_step,
+                  'game_iters': self._agents[0]._iter_count,
+                  'game_duration': 0}
+        ids = self._job['player_ids']
+        for i, agent in enumerate(self._agents):
+            opp = ids[1 - i] if len(ids) > 1 else ids[0]
+            side = {'player_id': agent.player_id, 'opponent_id': opp,
+                    'winloss': float(rewards.get(i, 0))}
+            side.update(agent.get_stat_data())
+            result[str(i)] = side
+        return result
+
+    # ------------------------------------------------------------------ run
+    def run(self):
+        self._setup_job()
+        env = self._make_env()
+        episode_num = self._cfg.episode_num
+        job_deadline = time.time() + self._cfg.actor_ask_for_job_interval * \
+            (1 + 0.3 * (2 * random.random() - 1))
+        while not self._end and (episode_num < 0 or self.episodes_done < episode_num):
+            try:
+                self._run_episode(env)
+                self.episodes_done += 1
+            except Exception:  # noqa: BLE001 - actors are cattle
+                traceback.print_exc()
+                try:
+                    env.close()
+                except Exception:  # noqa: BLE001
+                    pass
+                env = self._make_env()
+            if time.time() > job_deadline:
+                self._setup_job()
+                job_deadline = time.time() + self._cfg.actor_ask_for_job_interval
+        env.close()
+        return self.results
+
+    def close(self):
+        self._end = True

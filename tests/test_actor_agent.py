@@ -1,0 +1,84 @@
+"""Agent + mock-env rollout integration: trajectories produced by
+`Agent.collect_data` must collate through the RL pipeline into a valid
+`rl_learner_forward` batch (the full actor->learner data contract)."""
+import pytest
+import torch
+
+from distar_amd.actor.actor import Actor
+from distar_amd.actor.agent import Agent
+from distar_amd.envs.mock_env import MockSC2Env
+from distar_amd.lib.fake_data import rl_collate
+from distar_amd.losses import ReinforcementLoss
+from distar_amd.models import Model
+from distar_amd.utils.config import Config
+
+
+@pytest.mark.timeout(600)
+def test_agent_rollout_feeds_rl_learner():
+    torch.manual_seed(0)
+    cfg = Config({'common': {'type': 'train'},
+                  'actor': {'traj_len': 3, 'job_type': 'train'},
+                  'env': {'player_num': 2, 'max_episode_steps': 100000},
+                  'agent': {}})
+    env = MockSC2Env(cfg, entity_num_range=(24, 48), seed=1)
+    agents = [Agent(cfg, env_id=0) for _ in range(2)]
+    obs = env.reset()
+    for i, agent in enumerate(agents):
+        agent.player_id = f'MP{i}'
+        agent.reset(obs=obs.get(i))
+    trajs = []
+    done = False
+    last_obs = obs
+    steps = 0
+    while not done and steps < 12:
+        actions = {i: agents[i].step(last_obs[i])[0] for i in agents and last_obs}
+        for i, a in actions.items():
+            assert set(a) >= {'func_id', 'skip_steps', 'queued', 'unit_tags',
+                              'target_unit_tag', 'location'}
+        obs, rewards, done, infos = env.step(actions)
+        for i, agent in enumerate(agents):
+            if i in last_obs:
+                out = agent.collect_data(obs.get(i), rewards.get(i, 0), done, i)
+                if out is not None:
+                    trajs.append(out)
+        last_obs = {**last_obs, **obs}
+        steps += 1
+    assert trajs, 'no trajectory produced'
+    assert len(trajs[0]) == 4, 'traj_len steps + bootstrap frame'
+
+    # feed through the learner collate + forward + loss
+    batch = rl_collate([trajs[0]])
+    batch.pop('model_last_iter', None)
+    batch.pop('aux_type', None)
+    model = Model(Config({'common': {'type': 'train'},
+                          'model': {'enable_baselines':
+                                    ['winloss', 'build_order', 'built_unit',
+                                     'battle']}}),
+                  use_value_network=True)
+    out = model.rl_learner_forward(**batch)
+    loss = ReinforcementLoss(Config({}), 'MP0').compute_loss(out)
+    assert torch.isfinite(loss['total_loss'])
+    loss['total_loss'].backward()
+
+
+@pytest.mark.timeout(600)
+def test_actor_standalone_rollout():
+    torch.manual_seed(0)
+    cfg = Config({'actor': {'episode_num': 1, 'traj_len': 4, 'env_type': 'mock'},
+                  'env': {'player_num': 2, 'max_episode_steps': 6},
+                  'common': {'experiment_name': 'test_actor', 'type': 'train'}})
+    actor = Actor(cfg)
+    results = actor.run()
+    assert len(results) == 1
+    r = results[0]
+    assert '0' in r and '1' in r
+    assert abs(r['0']['winloss']) == 1 and r['0']['winloss'] == -r['1']['winloss']
+    assert 'z_type' in r['0']
+
+
+def test_agent_stat_data():
+    cfg = Config({'common': {'type': 'train'}})
+    agent = Agent(cfg)
+    agent.reset()
+    data = agent.get_stat_data()
+    assert 'race_id' in data and 'z_type' in data

@@ -135,3 +135,7 @@ def main():
 
 if __name__ == '__main__':
     main()
+
+# Note: the strategy-statistics Z files (distar_amd/assets/z_files/*.json) are
+# replay-derived *data* copied from the reference's lib/ directory — the same
+# artifacts our bin/gen_z.py regenerates from replay packs.
