@@ -167,7 +167,8 @@ class Agent:
         if obs.get('raw_obs') is not None:
             from ..lib.features import Features
             if not hasattr(self, '_feature'):
-                self._feature = Features(self._whole_cfg, self._map_name)
+                self._feature = Features(obs['game_info_proto'], obs['raw_obs'],
+                                         self._whole_cfg)
             agent_obs = self._feature.transform_obs(
                 obs['raw_obs'], padding_spatial=True,
                 opponent_obs=obs.get('opponent_obs') if self._use_value_feature else None)

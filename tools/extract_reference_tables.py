@@ -132,6 +132,21 @@ def main():
         json.dump(golden, f, indent=0)
     print('golden keys:', len(golden['policy']), len(golden['value']))
 
+    # raw action-function table: func_id -> ability/general/type (data for
+    # reverse_raw_action and transform_action)
+    from distar.pysc2.lib import actions as pysc2_actions
+    raw_funcs = []
+    for f in pysc2_actions.RAW_FUNCTIONS:
+        raw_funcs.append({
+            'id': int(f.id), 'name': f.name,
+            'ability_id': int(f.ability_id) if f.ability_id else 0,
+            'general_id': int(f.general_id) if f.general_id else 0,
+            'function_type': f.function_type.__name__,
+        })
+    with open(f'{OUT}/raw_functions.json', 'w') as fo:
+        json.dump({'raw_functions': raw_funcs}, fo)
+    print('raw functions:', len(raw_funcs))
+
 
 if __name__ == '__main__':
     main()
