@@ -1,1 +1,3 @@
 from .fake_dataloader import FakeSLDataloader, FakeRLDataloader
+from .adapter import Adapter
+from .coordinator import Coordinator, Worker

@@ -74,7 +74,8 @@ class SaveCkptHook(Hook):
 
 class LogShowHook(Hook):
     def __init__(self, **kwargs):
-        super().__init__('log_show', priority=30, **kwargs)
+        kwargs.setdefault('priority', 30)
+        super().__init__('log_show', **kwargs)
         self.freq = self.ext_args.get('freq', 100)
 
     def __call__(self, engine):
@@ -100,7 +101,8 @@ class LogReduceHook(Hook):
     """Allreduce the scalar log buffer across ranks as ONE flat tensor."""
 
     def __init__(self, **kwargs):
-        super().__init__('log_reduce', priority=10, **kwargs)
+        kwargs.setdefault('priority', 10)
+        super().__init__('log_reduce', **kwargs)
 
     def __call__(self, engine):
         if not is_initialized() or get_world_size() == 1:

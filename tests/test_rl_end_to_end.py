@@ -1,0 +1,80 @@
+"""End-to-end RL stack on CPU: coordinator + league(API) + actor (mock env)
++ RL learner with comm hooks, exchanging real trajectories/models through the
+Adapter transport — the single-node wiring of BASELINE config 5."""
+import threading
+import time
+
+import pytest
+import torch
+
+from distar_amd.actor.actor import Actor
+from distar_amd.actor.comm import LearnerComm
+from distar_amd.data.coordinator import Coordinator
+from distar_amd.league.api import create_league_server
+from distar_amd.league.league import League
+from distar_amd.learner.rl_learner import RLLearner
+from distar_amd.utils.config import Config
+
+
+@pytest.mark.timeout(900)
+def test_rl_league_loop(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    torch.manual_seed(0)
+    coord = Coordinator().run()
+    league = League(Config({
+        'common': {'experiment_name': 'e2e'},
+        'league': {
+            'save_resume_freq': 10000,
+            'active_players': {
+                'player_id': ['MP0'], 'checkpoint_path': ['none'],
+                'pipeline': ['default'], 'frac_id': [1],
+                'z_path': ['3map.json'], 'z_prob': [0.0],
+                'teacher_id': ['sl'], 'teacher_path': ['none'],
+                'one_phase_step': [int(1e9)], 'chosen_weight': [1.0],
+            }}}))
+    api = create_league_server(league, host='127.0.0.1').start()
+    cfg = Config({
+        'common': {'experiment_name': 'e2e', 'type': 'train'},
+        'communication': {'coordinator_ip': '127.0.0.1',
+                          'coordinator_port': coord.port,
+                          'league_ip': '127.0.0.1', 'league_port': api.port,
+                          'adapter_traj_worker_num': 1,
+                          'learner_send_model_freq': 1,
+                          'learner_send_train_info_freq': 1},
+        'env': {'player_num': 2, 'max_episode_steps': 100000},
+        'actor': {'episode_num': -1, 'traj_len': 3, 'env_type': 'mock',
+                  'job_type': 'train', 'use_cuda': False},
+        'learner': {'player_id': 'MP0', 'job_type': 'train', 'use_cuda': False,
+                    'use_amp': False,
+                    'data': {'batch_size': 2, 'trajectory_length': 3,
+                             'buffer_size': 2, 'use_async_cuda': False},
+                    'hook': {'after_iter': {
+                        'log_show': {'ext_args': {'freq': 1000}}}}},
+        'model': {'enable_baselines': ['winloss', 'build_order',
+                                       'built_unit', 'battle']},
+    })
+    actor = Actor(cfg)
+    actor_thread = threading.Thread(target=actor.run, daemon=True)
+    actor_thread.start()
+    try:
+        learner = RLLearner(cfg)
+        comm = LearnerComm(cfg)
+        resp = comm.register_learner(learner)
+        assert 'ckpt_path' in resp
+        learner._setup_comm_hooks(comm)
+        learner.run(max_iterations=2)
+        assert learner.last_iter.val == 2
+        assert 'total_loss' in learner.record.var_dict
+        # model was published for actors; league got train info
+        assert comm.adapter.length('MP0model') >= 0
+        deadline = time.time() + 30
+        while league.active_players['MP0'].total_agent_step == 0 and \
+                time.time() < deadline:
+            time.sleep(0.5)
+        assert league.active_players['MP0'].total_agent_step > 0
+    finally:
+        actor.close()
+        learner._dataloader.close()
+        league.close()
+        api.stop()
+        coord.close()
