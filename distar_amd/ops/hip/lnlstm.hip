@@ -27,8 +27,25 @@
 #define LN_EPS 1e-5f
 
 using bf16 = __hip_bfloat16;
+using bf16x2 = __hip_bfloat162;
 
 __device__ inline float bf2f(bf16 v) { return __bfloat162float(v); }
+
+// dot of 8 bf16 pairs loaded as one 16-byte vector each (G13: hipcc does not
+// auto-vectorize scalar bf16 loads; 16 B/lane is the coalescing sweet spot)
+__device__ inline float dot8_bf16(const uint4 a, const uint4 b) {
+  const bf16x2* pa = reinterpret_cast<const bf16x2*>(&a);
+  const bf16x2* pb = reinterpret_cast<const bf16x2*>(&b);
+  float acc = 0.f;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    float2 fa = __bfloat1622float2(pa[i]);
+    float2 fb = __bfloat1622float2(pb[i]);
+    acc = fmaf(fa.x, fb.x, acc);
+    acc = fmaf(fa.y, fb.y, acc);
+  }
+  return acc;
+}
 
 // block-wide sum over NT threads; scratch must hold NT floats
 __device__ inline float block_sum(float v, float* scratch) {
@@ -85,11 +102,11 @@ extern "C" __global__ void lnlstm_forward_kernel(
     float hg_loc[4];           // up to ceil(G/NT) owned gate columns
     int nown = 0;
     for (int j = tid; j < G; j += NT) {
-      const bf16* wrow = w_hh + (long)j * H;
+      const uint4* wrow = reinterpret_cast<const uint4*>(w_hh + (long)j * H);
+      const uint4* hv = reinterpret_cast<const uint4*>(hs);
       float acc = 0.f;
-      for (int k = 0; k < H; k += 2) {
-        acc += bf2f(hs[k]) * bf2f(wrow[k]);
-        acc += bf2f(hs[k + 1]) * bf2f(wrow[k + 1]);
+      for (int k8 = 0; k8 < H / 8; ++k8) {
+        acc += dot8_bf16(hv[k8], wrow[k8]);
       }
       hg_out[j] = acc;
       hg_loc[nown++] = acc;
@@ -277,13 +294,20 @@ extern "C" __global__ void lnlstm_backward_kernel(
     }
     __syncthreads();
 
-    // dh_prev = dhg @ W_hh  (via W_hh^T rows)
+    // dh_prev = dhg @ W_hh  (via W_hh^T rows; 16-byte weight loads)
     for (int k = tid; k < H; k += NT) {
-      const bf16* wtrow = w_hh_t + (long)k * G;
+      const uint4* wtrow = reinterpret_cast<const uint4*>(w_hh_t + (long)k * G);
       float acc = 0.f;
-      for (int j = 0; j < G; j += 2) {
-        acc += dhg[j] * bf2f(wtrow[j]);
-        acc += dhg[j + 1] * bf2f(wtrow[j + 1]);
+      for (int j8 = 0; j8 < G / 8; ++j8) {
+        uint4 wbits = wtrow[j8];
+        const bf16x2* pw = reinterpret_cast<const bf16x2*>(&wbits);
+        const float* dv = dhg + j8 * 8;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          float2 fw = __bfloat1622float2(pw[i]);
+          acc = fmaf(dv[2 * i], fw.x, acc);
+          acc = fmaf(dv[2 * i + 1], fw.y, acc);
+        }
       }
       dh_rec[k] = acc;
     }
