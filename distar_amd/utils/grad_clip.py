@@ -41,8 +41,8 @@ class GradClip:
             return total.item()
         if self.clip_type == 'clip_const':
             with torch.no_grad():
-                torch._foreach_clamp_(
-                    [p.grad for p in params], -self.threshold, self.threshold)
+                for p in params:
+                    p.grad.clamp_(-self.threshold, self.threshold)
                 total = torch.norm(torch.stack(
                     [torch.norm(p.grad, self.norm_type) for p in params]),
                     self.norm_type)
@@ -62,18 +62,23 @@ class GradClip:
                 self._ema = 0.99 * self._ema + 0.01 * min(total, self._ema * self.ignore_threshold)
             return total
         if self.clip_type == 'momentum_norm':
-            # per-parameter EMA-normalized clip (reference SL default)
+            # per-parameter EMA-normalized clip (reference SL default).
+            # Fully vectorized: one fused norm kernel + tensorized EMA state,
+            # ZERO host syncs (the reference loops with .item() per param).
             with torch.no_grad():
-                total_sq = 0.
-                for i, p in enumerate(params):
-                    norm = torch.norm(p.grad, self.norm_type).item()
-                    total_sq += norm ** 2
-                    ema = self._state.get(i, norm)
-                    clip_to = ema * self.threshold
-                    if self.step > self.begin_step and norm > clip_to:
-                        p.grad.mul_(clip_to / (norm + 1e-6))
-                    self._state[i] = 0.99 * ema + 0.01 * min(norm, ema * self.ignore_threshold)
-            return total_sq ** 0.5
+                grads = [p.grad for p in params]
+                norms = torch.stack(torch._foreach_norm(grads, self.norm_type))
+                ema = self._state.get('ema')
+                if ema is None or ema.shape != norms.shape:
+                    ema = norms.clone()
+                clip_to = ema * self.threshold
+                if self.step > self.begin_step:
+                    scale = (clip_to / (norms + 1e-6)).clamp(max=1.0)
+                    torch._foreach_mul_(grads, list(scale.unbind()))
+                self._state['ema'] = 0.99 * ema + \
+                    0.01 * torch.minimum(norms, ema * self.ignore_threshold)
+                total = torch.norm(norms, 2)
+            return total.item()
         if self.clip_type == 'clip_value':
             # Adam-like second-moment clamp
             with torch.no_grad():
