@@ -50,7 +50,8 @@ class SLBench:
             self.model = DistModule(self.model, bucket_cap_mb=args.bucket_mb)
         self.loss = SupervisedLoss(Config({'learner': {}}))
         self.grad_clip = build_grad_clip(Config({'type': 'momentum_norm', 'threshold': 1.0}))
-        self.optimizer = torch.optim.Adam(self.model.parameters(), lr=1e-3)
+        self.optimizer = torch.optim.Adam(self.model.parameters(), lr=1e-3,
+                                          fused=torch.cuda.is_available())
         B, T = args.batch, args.traj
         self.batches = [
             to_device(fake_sl_batch_fast(B, T, seed=100 * get_rank() + i), device)
@@ -100,7 +101,8 @@ class RLBench:
         self.loss = ReinforcementLoss(Config({}), 'MP0')
         self.grad_clip = build_grad_clip(Config({'type': 'pytorch_norm', 'threshold': 1.0}))
         self.optimizer = torch.optim.Adam(self.model.parameters(), lr=1e-5,
-                                          betas=(0.0, 0.99), eps=1e-5)
+                                          betas=(0.0, 0.99), eps=1e-5,
+                                          fused=torch.cuda.is_available())
         batches = []
         for i in range(args.pool):
             d = fake_rl_learner_data_fast(args.batch, args.traj,

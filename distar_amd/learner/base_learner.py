@@ -115,7 +115,7 @@ class BaseLearner:
         cfg = self._whole_cfg.learner
         self._optimizer = torch.optim.Adam(
             self.model.parameters(), lr=cfg.learning_rate,
-            weight_decay=cfg.weight_decay)
+            weight_decay=cfg.weight_decay, fused=self._use_cuda or None)
         milestones = list(cfg.lr_decay_milestones)
         self._lr_scheduler = torch.optim.lr_scheduler.MultiStepLR(
             self._optimizer, milestones=milestones, gamma=cfg.lr_decay)

@@ -58,7 +58,7 @@ class RLLearner(BaseLearner):
     def _setup_optimizer(self):
         self._optimizer = torch.optim.Adam(
             self.model.parameters(), lr=self._whole_cfg.learner.learning_rate,
-            betas=(0.0, 0.99), eps=1e-5)
+            betas=(0.0, 0.99), eps=1e-5, fused=self._use_cuda or None)
         self._lr_scheduler = torch.optim.lr_scheduler.MultiStepLR(
             self._optimizer, milestones=[], gamma=1)
 
