@@ -46,6 +46,8 @@ class SLBench:
         torch.manual_seed(1234 + get_rank())
         self.model = Model(Config({'common': {'type': 'train'}}), temperature=1.0)
         self.model = self.model.to(device)
+        if os.environ.get('DISTAR_AMD_CHANNELS_LAST') == '1':
+            self.model = self.model.to(memory_format=torch.channels_last)
         if get_world_size() > 1:
             self.model = DistModule(self.model, bucket_cap_mb=args.bucket_mb)
         self.loss = SupervisedLoss(Config({'learner': {}}))

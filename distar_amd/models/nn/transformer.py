@@ -35,13 +35,25 @@ class Attention(nn.Module):
         qkv = self.attention_pre(x).view(B, N, 3, self.head_num, self.head_dim)
         qkv = qkv.permute(2, 0, 3, 1, 4)  # 3, B, H, N, D
         query, key, value = qkv[0], qkv[1], qkv[2]
-        score = torch.matmul(query, key.transpose(-2, -1)) / math.sqrt(self.head_dim)
-        if mask is not None:
-            score = score.masked_fill(~mask, -1e9)
-        score = F.softmax(score, dim=-1)
-        if self.dropout is not None:
-            score = self.dropout(score)
-        attention = torch.matmul(score, value)                       # B, H, N, D
+        if x.is_cuda and self.dropout is None:
+            # fused scaled-dot-product path (flash/mem-efficient on ROCm).
+            # Additive -1e9 mask keeps the reference's finite-logit semantics
+            # for fully-padded query rows (a bool mask would yield NaNs).
+            attn_mask = None
+            if mask is not None:
+                attn_mask = torch.zeros(mask.shape, dtype=query.dtype,
+                                        device=query.device)
+                attn_mask.masked_fill_(~mask, -1e9)
+            attention = F.scaled_dot_product_attention(query, key, value,
+                                                       attn_mask=attn_mask)
+        else:
+            score = torch.matmul(query, key.transpose(-2, -1)) / math.sqrt(self.head_dim)
+            if mask is not None:
+                score = score.masked_fill(~mask, -1e9)
+            score = F.softmax(score, dim=-1)
+            if self.dropout is not None:
+                score = self.dropout(score)
+            attention = torch.matmul(score, value)                   # B, H, N, D
         attention = attention.permute(0, 2, 1, 3).reshape(B, N, -1)  # B, N, H*D
         return self.project(attention)
 
