@@ -196,9 +196,43 @@ std::vector<torch::Tensor> lnlstm_backward(
   return {digates, dhgates_raw, dh0, dc0, dlnh_w, dlnh_b, dlnc_w, dlnc_b};
 }
 
+extern "C" __global__ void entity_embed_kernel(
+    const int*, const float*, const int*, const int*, const int*, const int*,
+    __hip_bfloat16*, int, int, int, int, int);
+
+torch::Tensor entity_embed(torch::Tensor int_fields, torch::Tensor float_fields,
+                           torch::Tensor kinds, torch::Tensor offsets,
+                           torch::Tensor sizes, torch::Tensor src_idx,
+                           int64_t out_dim) {
+  TORCH_CHECK(int_fields.is_cuda() && int_fields.is_contiguous());
+  TORCH_CHECK(int_fields.scalar_type() == torch::kInt32);
+  TORCH_CHECK(float_fields.scalar_type() == torch::kFloat32);
+  int64_t R = int_fields.size(0);
+  int64_t n_int = int_fields.size(1);
+  int64_t n_float = float_fields.numel() ? float_fields.size(1) : 0;
+  int64_t n_fields = kinds.size(0);
+  auto out = torch::empty({R, out_dim},
+                          int_fields.options().dtype(torch::kBFloat16));
+  int threads = 256;                      // 4 waves -> 4 rows per block
+  int blocks = std::min<int64_t>((R + 3) / 4, 8192);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(entity_embed_kernel, dim3(blocks), dim3(threads), 0,
+                     stream.stream(),
+                     int_fields.data_ptr<int>(),
+                     n_float ? float_fields.data_ptr<float>() : nullptr,
+                     kinds.data_ptr<int>(), offsets.data_ptr<int>(),
+                     sizes.data_ptr<int>(), src_idx.data_ptr<int>(),
+                     reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                     (int)R, (int)out_dim, (int)n_fields, (int)n_int,
+                     (int)n_float);
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("entity_embed", &entity_embed,
+        "fused 36-field entity embedding -> (R, 997) bf16");
   m.def("lambda_return_scan", &lambda_return_scan,
         "generalized lambda-return reverse scan (T,B)");
   m.def("vtrace_scan", &vtrace_scan, "v-trace corrected-value reverse scan");

@@ -192,3 +192,30 @@ def test_upsample2x_matches_interpolate():
         torch.testing.assert_close(x.grad.float(), x2.grad,
                                    rtol=1e-2 if dtype == torch.bfloat16 else 1e-4,
                                    atol=1e-2 if dtype == torch.bfloat16 else 1e-4)
+
+
+def test_entity_embed_kernel_matches_eager():
+    """Fused K2 entity embedding vs the 36-gather eager path."""
+    import os
+    from distar_amd.models import Model
+    from distar_amd.lib.consts import fake_step_data
+    from distar_amd.utils.config import Config
+    from distar_amd.utils.data import default_collate_with_dim, to_device
+    torch.manual_seed(0)
+    model = Model(Config({'common': {'type': 'train'}})).cuda()
+    enc = model.encoder.entity_encoder
+    obs = default_collate_with_dim(
+        [fake_step_data(train=False, entity_num=64, randomize=True)
+         for _ in range(3)])
+    ent = to_device(obs['entity_info'], 'cuda')
+    ent['last_selected_units'] = torch.zeros(3, 512, dtype=torch.int8, device='cuda')
+    ent['last_targeted_unit'] = torch.zeros(3, 512, dtype=torch.int8, device='cuda')
+    out_hip = enc.embed_fields(ent)
+    os.environ['DISTAR_AMD_DISABLE_HIP'] = '1'
+    try:
+        out_ref = enc.embed_fields(ent)
+    finally:
+        del os.environ['DISTAR_AMD_DISABLE_HIP']
+    assert out_hip.shape == out_ref.shape == (3, 512, 997)
+    torch.testing.assert_close(out_hip.float(), out_ref.float(),
+                               rtol=1e-2, atol=1e-2)
