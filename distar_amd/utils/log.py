@@ -133,6 +133,40 @@ class VariableRecord:
         return '\n'.join(rows)
 
 
+class AlphaStarVarRecord(VariableRecord):
+    """Grouped RL variable tables (reference log_helper.AlphaStarVarRecord):
+    renders the {field} x {reward, value, td, pg, upgo, entropy, kl} grid of
+    registered variables as one aligned text block."""
+
+    FIELDS = ('winloss', 'build_order', 'built_unit', 'effect', 'upgrade', 'battle')
+    COLUMNS = ('reward', 'value', 'td', 'action_type', 'delay', 'queued',
+               'selected_units', 'target_unit', 'target_location', 'total')
+
+    def get_star_text(self):
+        lines = ['  ' + f'{"field":<14s}' + ' '.join(f'{c:>14s}' for c in self.COLUMNS)]
+        for field in self.FIELDS:
+            row = [f'  {field:<14s}']
+            for col in self.COLUMNS:
+                key = f'{field}/{col}'
+                row.append(f'{self.var_dict[key].avg:14.5f}' if key in self.var_dict
+                           else ' ' * 14)
+            lines.append(' '.join(row))
+        for group in ('upgo', 'entropy', 'kl'):
+            row = [f'  {group:<14s}']
+            for col in self.COLUMNS:
+                key = f'{group}/{col}'
+                row.append(f'{self.var_dict[key].avg:14.5f}' if key in self.var_dict
+                           else ' ' * 14)
+            lines.append(' '.join(row))
+        return '\n'.join(lines)
+
+    def get_vars_text(self, keys=None):
+        grouped = {k for k in self.var_dict
+                   if '/' in k and k.split('/')[0] in self.FIELDS + ('upgo', 'entropy', 'kl')}
+        flat = super().get_vars_text([k for k in sorted(self.var_dict) if k not in grouped])
+        return self.get_star_text() + '\n' + flat
+
+
 def build_logger(cfg, name='default', rank=0):
     """-> (TextLogger, ScalarLogger, VariableRecord); loggers only on rank 0
     (reference log_helper.build_logger)."""
@@ -145,5 +179,6 @@ def build_logger(cfg, name='default', rank=0):
         scalar_logger = ScalarLogger(path, name=name)
     else:
         logger, scalar_logger = None, None
-    record = VariableRecord()
+    var_type = cfg.get('learner', {}).get('var_record_type') if hasattr(cfg, 'get') else None
+    record = AlphaStarVarRecord() if var_type == 'alphastar' else VariableRecord()
     return logger, scalar_logger, record
