@@ -231,9 +231,49 @@ class SelectedUnitsHead(nn.Module):
         final_ae = autoregressive_embedding + final_delta
         return logits, None, final_ae, selected_units_num, None
 
+    def _query_sample_hip(self, key, entity_num, autoregressive_embedding,
+                          logits_mask, key_embeddings, su_mask, uniforms=None):
+        """K7: the whole data-dependent loop as ONE HIP kernel
+        (ops/hip/su_sample.hip); final ae recomputed host-side from the
+        selected keys (cheap batched fc pair)."""
+        from ...ops.su_sample import su_sample
+        bs = autoregressive_embedding.shape[0]
+        device = autoregressive_embedding.device
+        logits, results, num = su_sample(
+            self, autoregressive_embedding.float(), key, logits_mask, su_mask,
+            entity_num, self.whole_cfg.model.temperature, uniforms=uniforms)
+        S = max(int(num.max()), 1)
+        logits = logits[:, :S]
+        results = results[:, :S]
+        # final ae = base + embed(mean of selected keys), zero-selection rows
+        # keep the raw (zero) sum like the reference
+        arange = torch.arange(bs, device=device)
+        sel_mask = (torch.arange(S, device=device).unsqueeze(0) <
+                    (num - 1).clamp(min=0).unsqueeze(1))
+        gathered = key.gather(1, results.clamp(min=0).unsqueeze(-1)
+                              .expand(-1, -1, self.key_dim))
+        sel_sum = (gathered * sel_mask.unsqueeze(-1)).sum(dim=1)
+        cnt = sel_mask.sum(dim=1, keepdim=True).float()
+        mean = torch.where(cnt > 0, sel_sum / cnt.clamp(min=1), sel_sum)
+        ae = autoregressive_embedding + self.embed_fc2(self.embed_fc1(mean))
+        extra_units = torch.zeros(bs, MAX_ENTITY_NUM + 1, device=device)
+        if self.extra_units:
+            last = (num - 1).clamp(min=0)
+            last_logits = logits[arange, last]
+            end_logit = last_logits[arange, entity_num]
+            extra_units[:, :last_logits.shape[1]] = \
+                (last_logits > end_logit.unsqueeze(1)).float()
+        return logits, results, ae, num, extra_units
+
     def _query_sample(self, key, entity_num, autoregressive_embedding, logits_mask,
-                      key_embeddings, su_mask):
+                      key_embeddings, su_mask, uniforms=None):
         """Data-dependent sampling loop (reference `action_arg_head.py:262-313`)."""
+        import os
+        if autoregressive_embedding.is_cuda and \
+                os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1':
+            return self._query_sample_hip(key, entity_num, autoregressive_embedding,
+                                          logits_mask, key_embeddings, su_mask,
+                                          uniforms)
         ae = autoregressive_embedding
         bs = ae.shape[0]
         device = ae.device
@@ -261,8 +301,14 @@ class SelectedUnitsHead(nn.Module):
             queries = lstm_output.permute(1, 0, 2)                      # B, 1, k
             step_logits = (queries * key).sum(dim=2)                    # B, N+1
             step_logits = step_logits.masked_fill(~logits_mask, -1e9)
-            units = torch.multinomial(F.softmax(step_logits / self.whole_cfg.model.temperature,
-                                                dim=-1), 1)[:, 0]
+            p = F.softmax(step_logits / self.whole_cfg.model.temperature, dim=-1)
+            if uniforms is None:
+                units = torch.multinomial(p, 1)[:, 0]
+            else:   # deterministic inverse-CDF (golden tests vs the kernel)
+                cdf = p.cumsum(-1)
+                target = uniforms[:, i:i + 1] * cdf[:, -1:]
+                units = torch.searchsorted(cdf, target).clamp(
+                    max=cdf.shape[1] - 1)[:, 0]
             result = units
             newly_ended = (result == entity_num) & ~end_flag
             selected_units_num[newly_ended] = i + 1

@@ -196,6 +196,55 @@ std::vector<torch::Tensor> lnlstm_backward(
   return {digates, dhgates_raw, dh0, dc0, dlnh_w, dlnh_b, dlnc_w, dlnc_b};
 }
 
+extern "C" __global__ void su_sample_kernel(
+    const float*, const __hip_bfloat16*, const unsigned char*,
+    const unsigned char*, const int*, const float*,
+    const __hip_bfloat16*, const float*, const __hip_bfloat16*, const float*,
+    const __hip_bfloat16*, const __hip_bfloat16*,
+    const float*, const float*, const float*, const float*, const float*,
+    const float*, const __hip_bfloat16*, const float*, const __hip_bfloat16*,
+    const float*, float, float*, int*, int*, int, int, int);
+
+std::vector<torch::Tensor> su_sample(
+    torch::Tensor ae_base, torch::Tensor keys, torch::Tensor avail,
+    torch::Tensor su_mask, torch::Tensor entity_num, torch::Tensor uniforms,
+    torch::Tensor Wq1, torch::Tensor bq1, torch::Tensor Wq2, torch::Tensor bq2,
+    torch::Tensor Wih, torch::Tensor Whh,
+    torch::Tensor lni_w, torch::Tensor lni_b,
+    torch::Tensor lnh_w, torch::Tensor lnh_b,
+    torch::Tensor lnc_w, torch::Tensor lnc_b,
+    torch::Tensor We1, torch::Tensor be1, torch::Tensor We2, torch::Tensor be2,
+    double temperature) {
+  int64_t B = ae_base.size(0), AE = ae_base.size(1), N1 = keys.size(1);
+  TORCH_CHECK(ae_base.is_cuda() && ae_base.is_contiguous());
+  TORCH_CHECK(keys.scalar_type() == torch::kBFloat16 && keys.is_contiguous());
+  auto opt = ae_base.options();
+  auto logits_out = torch::full({B, 64, N1}, -1e9, opt);
+  auto results = torch::zeros({B, 64}, opt.dtype(torch::kInt32));
+  auto num_out = torch::zeros({B}, opt.dtype(torch::kInt32));
+  size_t lds = sizeof(float) * (AE + 256 + 32 + 128 + 32 + 32 + 32 + 256 + N1 + 256)
+      + ((N1 + 63) / 64) * 64 + N1 * 32 * sizeof(__hip_bfloat16);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(su_sample_kernel, dim3(B), dim3(256), lds, stream.stream(),
+      ae_base.data_ptr<float>(),
+      reinterpret_cast<const __hip_bfloat16*>(keys.data_ptr()),
+      avail.data_ptr<unsigned char>(), su_mask.data_ptr<unsigned char>(),
+      entity_num.data_ptr<int>(), uniforms.data_ptr<float>(),
+      reinterpret_cast<const __hip_bfloat16*>(Wq1.data_ptr()), bq1.data_ptr<float>(),
+      reinterpret_cast<const __hip_bfloat16*>(Wq2.data_ptr()), bq2.data_ptr<float>(),
+      reinterpret_cast<const __hip_bfloat16*>(Wih.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(Whh.data_ptr()),
+      lni_w.data_ptr<float>(), lni_b.data_ptr<float>(),
+      lnh_w.data_ptr<float>(), lnh_b.data_ptr<float>(),
+      lnc_w.data_ptr<float>(), lnc_b.data_ptr<float>(),
+      reinterpret_cast<const __hip_bfloat16*>(We1.data_ptr()), be1.data_ptr<float>(),
+      reinterpret_cast<const __hip_bfloat16*>(We2.data_ptr()), be2.data_ptr<float>(),
+      (float)temperature,
+      logits_out.data_ptr<float>(), results.data_ptr<int>(),
+      num_out.data_ptr<int>(), (int)B, (int)N1, (int)AE);
+  return {logits_out, results, num_out};
+}
+
 extern "C" __global__ void entity_embed_kernel(
     const int*, const float*, const int*, const int*, const int*, const int*,
     __hip_bfloat16*, int, int, int, int, int);
@@ -231,6 +280,7 @@ torch::Tensor entity_embed(torch::Tensor int_fields, torch::Tensor float_fields,
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("su_sample", &su_sample, "selected-units sampling loop (one kernel)");
   m.def("entity_embed", &entity_embed,
         "fused 36-field entity embedding -> (R, 997) bf16");
   m.def("lambda_return_scan", &lambda_return_scan,

@@ -24,6 +24,7 @@ SOURCES = [
     os.path.join(HIP_DIR, 'lnlstm.hip'),
     os.path.join(HIP_DIR, 'upsample.hip'),
     os.path.join(HIP_DIR, 'entity_embed.hip'),
+    os.path.join(HIP_DIR, 'su_sample.hip'),
     os.path.join(HIP_DIR, 'bindings.cpp'),
 ]
 
