@@ -95,6 +95,7 @@ class RLBench:
         torch.manual_seed(1234 + get_rank())
         self.model = Model(
             Config({'common': {'type': 'train'},
+                    'learner': {'use_value_feature': not args.no_value_feature},
                     'model': {'enable_baselines':
                               ['winloss', 'build_order', 'built_unit', 'battle']}}),
             use_value_network=True).to(device)
@@ -109,7 +110,8 @@ class RLBench:
         for i in range(args.pool):
             d = fake_rl_learner_data_fast(args.batch, args.traj,
                                           entity_num=args.entities,
-                                          seed=100 * get_rank() + i)
+                                          seed=100 * get_rank() + i,
+                                          value_feature=not args.no_value_feature)
             d.pop('model_last_iter')
             batches.append(to_device(d, device))
         self.batches = batches
@@ -141,6 +143,9 @@ def main():
     p.add_argument('--pool', type=int, default=2)
     p.add_argument('--bucket-mb', type=int, default=64)
     p.add_argument('--no-amp', action='store_true')
+    p.add_argument('--no-value-feature', action='store_true',
+                   help='RL mode: drop the opponent-side value features '
+                        '(the reference RL config trains WITH them)')
     args = p.parse_args()
     if args.batch is None:
         args.batch = 32 if args.mode == 'sl' else 16
@@ -204,6 +209,7 @@ def main():
                 'entities': 512 if args.mode == 'sl' else args.entities,
                 'parallelism': f'dp{n_gpus}',
                 'mode': args.mode,
+                'value_feature': args.mode == 'rl' and not args.no_value_feature,
             },
         }
         print(json.dumps(result))

@@ -258,7 +258,33 @@ def fake_sl_batch_fast(batch_size=32, traj_len=64, entity_min=100, seed=0):
     }
 
 
-def fake_rl_learner_data_fast(batch_size=4, unroll_len=16, entity_num=256, seed=0):
+def fake_value_feature(n, entity_num, g):
+    """Synthetic opponent-side value features (reference
+    `lib/features.py:735-765` value_feature dict), batched (n, ...)."""
+    from .actions import NUM_CUMULATIVE_STAT_ACTIONS
+    H, W = SPATIAL_SIZE
+    return {
+        'enemy_unit_counts_bow': torch.randint(0, 3, (n, 260), generator=g).to(torch.uint8),
+        'enemy_unit_type_bool': torch.randint(0, 2, (n, 260), generator=g).to(torch.uint8),
+        'enemy_agent_statistics': torch.rand(n, 10, generator=g) * 5,
+        'enemy_upgrades': torch.randint(0, 2, (n, 90), generator=g).to(torch.uint8),
+        'unit_alliance': torch.randint(0, 2, (n, MAX_ENTITY_NUM), generator=g),
+        'unit_type': torch.randint(0, 260, (n, MAX_ENTITY_NUM), generator=g),
+        'beginning_order': torch.randint(0, 174, (n, 20), generator=g),
+        'bo_location': torch.randint(0, H * W, (n, 20), generator=g),
+        'cumulative_stat': torch.randint(0, 2, (n, NUM_CUMULATIVE_STAT_ACTIONS),
+                                         generator=g),
+        'unit_x': torch.randint(0, W, (n, MAX_ENTITY_NUM), generator=g),
+        'unit_y': torch.randint(0, H, (n, MAX_ENTITY_NUM), generator=g),
+        'total_unit_count': torch.randint(entity_num // 2, MAX_ENTITY_NUM, (n,),
+                                          generator=g),
+        'own_units_spatial': torch.randint(0, 2, (n, 1, H, W), generator=g).bool(),
+        'enemy_units_spatial': torch.randint(0, 2, (n, 1, H, W), generator=g).bool(),
+    }
+
+
+def fake_rl_learner_data_fast(batch_size=4, unroll_len=16, entity_num=256, seed=0,
+                              value_feature=False):
     """Vectorized RL learner batch (uniform entity width = batch-max padding
     outcome), time-major obs over (T+1)*B rows like the RL collate."""
     T, B, EN = unroll_len, batch_size, entity_num
@@ -333,7 +359,7 @@ def fake_rl_learner_data_fast(batch_size=4, unroll_len=16, entity_num=256, seed=
         'built_unit': torch.rand(T, B, generator=g) * 2 - 1,
         'battle': torch.rand(T, B, generator=g) * 2 - 1,
     }
-    return {
+    out = {
         'spatial_info': spatial_info, 'scalar_info': scalar_info,
         'entity_info': entity_info, 'entity_num': entity_nums,
         'hidden_state': [(torch.zeros(n, 384), torch.zeros(n, 384)) for _ in range(3)],
@@ -344,6 +370,9 @@ def fake_rl_learner_data_fast(batch_size=4, unroll_len=16, entity_num=256, seed=
         'model_last_iter': torch.zeros(B),
         'batch_size': B, 'unroll_len': T,
     }
+    if value_feature:
+        out['value_feature'] = fake_value_feature(n, EN, g)
+    return out
 
 
 def fake_sl_batch(batch_size=4, traj_len=8, entity_num=None, seed=0):

@@ -66,3 +66,23 @@ def test_rl_value_pretrain_freezes_policy():
                for p in m.value_networks.parameters())
     assert all(p.grad is None or p.grad.abs().sum() == 0
                for p in m.policy.parameters())
+
+
+def test_rl_loss_with_value_features():
+    """use_value_feature path: ValueEncoder consumes the opponent-side
+    features and feeds the critics (reference rl_user_config default)."""
+    torch.manual_seed(0)
+    m = Model(Config({'common': {'type': 'train'},
+                      'learner': {'use_value_feature': True},
+                      'model': {'enable_baselines':
+                                ['winloss', 'build_order', 'built_unit', 'battle']}}),
+              use_value_network=True)
+    from distar_amd.lib.fake_data import fake_rl_learner_data_fast
+    data = fake_rl_learner_data_fast(2, 3, entity_num=64, value_feature=True)
+    data.pop('model_last_iter')
+    out = m.rl_learner_forward(**data)
+    ld = ReinforcementLoss(Config({}), 'MP0').compute_loss(out)
+    assert torch.isfinite(ld['total_loss'])
+    ld['total_loss'].backward()
+    assert any(p.grad is not None and torch.isfinite(p.grad).all()
+               for p in m.value_encoder.parameters())
