@@ -298,6 +298,8 @@ def fake_rl_learner_data_fast(batch_size=4, unroll_len=16, entity_num=256, seed=
     en_flat = entity_nums[:T * B]
     r[torch.arange(EN).unsqueeze(0) >= en_flat.unsqueeze(1)] = 2.0
     sel = r.argsort(dim=1)[:, :MAX_SELECTED_UNITS_NUM]
+    if sel.shape[1] < MAX_SELECTED_UNITS_NUM:
+        sel = torch.nn.functional.pad(sel, (0, MAX_SELECTED_UNITS_NUM - sel.shape[1]))
     sel.scatter_(1, (su_num.view(-1) - 1).unsqueeze(1), en_flat.unsqueeze(1))
     sel = sel.view(T, B, MAX_SELECTED_UNITS_NUM).long()
     action_info = {
