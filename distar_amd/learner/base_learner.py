@@ -117,8 +117,16 @@ class BaseLearner:
             self.model.parameters(), lr=cfg.learning_rate,
             weight_decay=cfg.weight_decay, fused=self._use_cuda or None)
         milestones = list(cfg.lr_decay_milestones)
-        self._lr_scheduler = torch.optim.lr_scheduler.MultiStepLR(
+        decay = torch.optim.lr_scheduler.MultiStepLR(
             self._optimizer, milestones=milestones, gamma=cfg.lr_decay)
+        warmup_iters = int(cfg.warmup_iters)
+        if warmup_iters > 0:
+            warmup = torch.optim.lr_scheduler.LinearLR(
+                self._optimizer, start_factor=1e-3, total_iters=warmup_iters)
+            self._lr_scheduler = torch.optim.lr_scheduler.SequentialLR(
+                self._optimizer, [warmup, decay], milestones=[warmup_iters])
+        else:
+            self._lr_scheduler = decay
 
     def _train(self, data):
         raise NotImplementedError

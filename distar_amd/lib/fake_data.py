@@ -229,11 +229,8 @@ def fake_sl_batch_fast(batch_size=32, traj_len=64, entity_min=100, seed=0):
     entity_num = torch.randint(entity_min, MAX_ENTITY_NUM, (n,), generator=g)
     su_num = torch.minimum(
         torch.randint(1, MAX_SELECTED_UNITS_NUM, (n,), generator=g), entity_num)
-    # distinct selections per row, end token at position su_num-1
-    sel = torch.rand(n, MAX_ENTITY_NUM, generator=g).argsort(dim=1)
-    sel = sel % entity_num.unsqueeze(1)
-    # argsort of random is a permutation of [0,512); mod en breaks distinctness,
-    # so instead take a permutation restricted per-row via argsort of masked rand
+    # distinct selections per row (argsort of masked rand restricts the
+    # permutation to each row's valid entities), end token at su_num-1
     r = torch.rand(n, MAX_ENTITY_NUM, generator=g)
     r[torch.arange(MAX_ENTITY_NUM).unsqueeze(0) >= entity_num.unsqueeze(1)] = 2.0
     sel = r.argsort(dim=1)[:, :MAX_SELECTED_UNITS_NUM]
