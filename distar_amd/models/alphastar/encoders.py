@@ -170,22 +170,53 @@ class SpatialEncoder(nn.Module):
             self.gap = nn.AdaptiveAvgPool2d((1, 1))
             self.fc = fc_block(dims[-1], self.cfg.fc_dim, activation=self.act)
 
+    def _assemble_const_planes(self, x, spatial_y, spatial_x, dtype, device):
+        """The 24 input channels that carry no gradient (height map, 6
+        one-hot categorical maps, 6 effect planes) built WITHOUT per-field
+        embedding gathers or a 56-channel concat: channel-scatter writes into
+        one buffer (hot positions only, ~100 MB instead of ~20 GB of
+        materialized one-hot fp32 at the SL bench shape)."""
+        B = x['height_map'].shape[0]
+        off = 0
+        with torch.no_grad():
+            const = torch.zeros(B, 24, spatial_y, spatial_x, dtype=dtype,
+                                device=device)
+            for k, item in self.cfg.module.items():
+                if item['arc'] == 'other':          # height_map
+                    const[:, off] = x[k].to(dtype) / 256
+                    off += 1
+                elif item['arc'] == 'one_hot':
+                    idx = (x[k].long() + off).unsqueeze(1)
+                    const.scatter_(1, idx.clamp_(max=off + item['num_embeddings'] - 1),
+                                   torch.ones((), dtype=dtype, device=device)
+                                   .expand(B, 1, spatial_y, spatial_x))
+                    off += item['num_embeddings']
+                elif item['arc'] == 'scatter':      # effect plane
+                    flat = const[:, off].reshape(B, spatial_y * spatial_x)
+                    pos = x[k].long().clamp(0, spatial_y * spatial_x - 1)
+                    flat.scatter_(1, pos, torch.ones((), dtype=dtype, device=device)
+                                  .expand(B, pos.shape[1]))
+                    off += 1
+        return const, off
+
     def forward(self, x: Dict[str, Tensor], scatter_map: Tensor) -> Tuple[Tensor, List[Tensor]]:
         spatial_y = self.whole_cfg.model.spatial_y
         spatial_x = self.whole_cfg.model.spatial_x
-        embeddings = []
-        for k, item in self.cfg.module.items():
-            if item['arc'] == 'one_hot':
-                emb = self.encode_modules[k](x[k].long()).permute(0, 3, 1, 2)
-                embeddings.append(emb)
-            elif item['arc'] == 'other':
-                assert k == 'height_map'
-                embeddings.append(x[k].unsqueeze(1).float() / 256)
-            elif item['arc'] == 'scatter':
-                embeddings.append(spatial_effect_plane(x[k].shape[0], x[k], spatial_y, spatial_x))
-        embeddings.append(scatter_map)
-        out = torch.cat(embeddings, dim=1)
-        out = self.project(out)
+        proj_dtype = torch.bfloat16 if torch.is_autocast_enabled() \
+            else self.project[0].weight.dtype
+        const, n_const = self._assemble_const_planes(
+            x, spatial_y, spatial_x, proj_dtype, scatter_map.device)
+        # split the 1x1 projection: conv(cat([const, scatter])) ==
+        # conv(const) + conv(scatter) with the weight sliced — skips the
+        # 56-channel concat while keeping project.0.weight whole (checkpoint
+        # layout) and differentiable wrt both weight slices and scatter_map
+        conv = self.project[0]
+        w = conv.weight
+        out = torch.nn.functional.conv2d(const, w[:, :n_const]) + \
+            torch.nn.functional.conv2d(scatter_map.to(const.dtype), w[:, n_const:],
+                                       bias=conv.bias)
+        for layer in list(self.project)[1:]:        # norm/act of the block
+            out = layer(out)
         map_skip = []
         for i in range(len(self.downsample)):
             map_skip.append(out)
