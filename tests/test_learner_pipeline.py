@@ -71,3 +71,24 @@ def test_offline_sl_dataloader_lanes(tmp_path):
         assert not any(batch2['new_episodes'])
     finally:
         loader.close()
+
+
+@pytest.mark.timeout(900)
+def test_auto_checkpoint_on_crash(tmp_path):
+    """auto_checkpoint saves an emergency checkpoint when the train loop
+    raises (reference checkpoint_helper.py:325-369)."""
+
+    class Boom(RuntimeError):
+        pass
+
+    class CrashingLearner(SLLearner):
+        def _train(self, data):
+            if self.last_iter.val >= 1:
+                raise Boom('injected fault')
+            super()._train(data)
+
+    learner = CrashingLearner(_sl_cfg(tmp_path))
+    with pytest.raises(Boom):
+        learner.run(max_iterations=5)
+    ckpts = glob.glob(str(tmp_path / 'test_sl' / 'checkpoint' / '*.pth.tar'))
+    assert ckpts, 'no emergency checkpoint saved'

@@ -86,3 +86,26 @@ def test_rl_loss_with_value_features():
     ld['total_loss'].backward()
     assert any(p.grad is not None and torch.isfinite(p.grad).all()
                for p in m.value_encoder.parameters())
+
+
+def test_rl_loss_with_dapo():
+    """DAPO successive-model KL (reference rl_loss.py:165-172), enabled for
+    MainPlayer learners."""
+    torch.manual_seed(0)
+    m = Model(Config({'common': {'type': 'train'},
+                      'model': {'enable_baselines': ['winloss']}}),
+              use_value_network=True)
+    from distar_amd.lib.fake_data import fake_rl_learner_data_fast
+    data = fake_rl_learner_data_fast(2, 3, entity_num=64)
+    data.pop('model_last_iter')
+    out = m.rl_learner_forward(**data)
+    out['successive_logit'] = {k: v.detach().clone()
+                               for k, v in out['target_logit'].items()}
+    loss = ReinforcementLoss(Config({'use_dapo': True}), 'MP0')
+    assert loss.use_dapo
+    ld = loss.compute_loss(out)
+    assert 'battle/total' in ld and torch.isfinite(ld['total_loss'])
+    ld['total_loss'].backward()
+    # non-main players silently disable dapo
+    loss2 = ReinforcementLoss(Config({'use_dapo': True}), 'EP0')
+    assert not loss2.use_dapo
