@@ -25,7 +25,9 @@ class Coordinator:
             '/request_datum': self._request_datum,
             '/queue_length': self._queue_length,
             '/remove_server': self._remove_server,
+            '/start_worker': self._start_worker,
         }, host=host, port=self.port)
+        self._workers = {}
 
     # --------------------------------------------------------------- routes
     def _register_datum(self, body):
@@ -58,6 +60,17 @@ class Coordinator:
                 self._strikes[key] = 0
                 return {'removed': True}
         return {'removed': False}
+
+    def _start_worker(self, body):
+        """Spawn a per-token Worker broker to offload a hot token
+        (reference coordinator.py:156-165); returns its address."""
+        token = body.get('token', 'default')
+        with self._lock:
+            if token not in self._workers:
+                worker = Worker(host='0.0.0.0')
+                worker.run()
+                self._workers[token] = worker
+            return {'ip': '127.0.0.1', 'port': self._workers[token].port}
 
     # ------------------------------------------------------------------ api
     def run(self, daemon=True):

@@ -18,7 +18,7 @@ import time
 from collections import defaultdict
 from shutil import copyfile
 
-from .elo import ELORating
+from .elo import ELORating, TrueSkill
 from .player import (ActivePlayer, AdaptiveEvolutionaryExploiterPlayer,
                      ExpertExploiterPlayer, ExpertPlayer, ExploiterPlayer,
                      HistoricalPlayer, MainExploiterPlayer, MainPlayer)
@@ -76,6 +76,7 @@ class League:
         self._stat_warm_up_size = self.cfg.stat_warm_up_size
         self._payoff_min_win_rate_games = self.cfg.payoff_min_win_rate_games
         self.elo = ELORating()
+        self.trueskill = TrueSkill()
         self.api_info = defaultdict(list)
         self._init_league()
         self.save_resume_freq = self.cfg.save_resume_freq
@@ -387,6 +388,10 @@ class League:
             with self._lock:
                 self.elo.update(first['player_id'], first['opponent_id'],
                                 first['winloss'])
+                if first['winloss'] > 0:
+                    self.trueskill.update(first['player_id'], first['opponent_id'])
+                elif first['winloss'] < 0:
+                    self.trueskill.update(first['opponent_id'], first['player_id'])
             if self.elo.game_count % 100 == 0:
                 self.logger.info(self.elo.elo_text())
         for side in sides.values():
@@ -411,7 +416,7 @@ class League:
                             f'league_resume_{int(time.time())}.pkl')
         state = {'active_players': self.active_players,
                  'historical_players': self.historical_players,
-                 'elo': self.elo}
+                 'elo': self.elo, 'trueskill': self.trueskill}
         with self._lock:
             save_file(path, state, fs_type='nppickle')
         self.logger.info(f'saved league resume: {path}')
@@ -422,6 +427,7 @@ class League:
         self.active_players = state['active_players']
         self.historical_players = state['historical_players']
         self.elo = state['elo']
+        self.trueskill = state.get('trueskill', TrueSkill())
 
     def _save_resume_thread(self):
         last = time.time()
