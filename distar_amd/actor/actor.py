@@ -13,6 +13,7 @@ Environment selection: 'mock' (synthetic spec-identical episodes; default in
 this offline image) or 'sc2' (real game through envs/env.py, gated on
 s2clientprotocol).
 """
+import os
 import random
 import time
 import traceback
@@ -72,16 +73,32 @@ class Actor:
                    'env_info': {'map_name': 'KingsCove'}}
         self._job = job
         self._agents = []
+        pipelines = job.get('pipelines', ['default'] * len(job['player_ids']))
         for i, player_id in enumerate(job['player_ids']):
-            agent = Agent(self._whole_cfg, env_id=0)
+            pipeline = pipelines[i] if i < len(pipelines) else 'default'
+            if pipeline in ('default', 'bot'):
+                agent = Agent(self._whole_cfg, env_id=0)
+            else:
+                from ..utils.import_helper import import_pipeline_agent
+                agent = import_pipeline_agent(pipeline)(self._whole_cfg, env_id=0)
             agent.player_id = player_id
             ckpt = job['checkpoint_paths'][i] if i < len(job['checkpoint_paths']) else 'none'
-            if ckpt not in ('none', None) and not self._cfg.fake_model:
+            if ckpt not in ('none', None) and not self._cfg.fake_model and \
+                    getattr(agent, 'HAS_MODEL', False):
                 try:
                     self._ckpt_helper.load(ckpt, agent.model, strict=False,
                                            logger_prints=self._logger.info)
                 except FileNotFoundError:
                     self._logger.info(f'checkpoint missing: {ckpt}, random init')
+            teacher = job.get('teacher_checkpoint_paths', [])
+            tpath = teacher[i] if i < len(teacher) else 'none'
+            if tpath not in ('none', None) and getattr(agent, 'HAS_TEACHER', False) \
+                    and not self._cfg.fake_model and os.path.isfile(str(tpath)):
+                from ..models.alphastar.model import Model as _Model
+                agent.teacher_model = _Model(self._whole_cfg)
+                agent.teacher_model.eval()
+                self._ckpt_helper.load(tpath, agent.teacher_model, strict=False,
+                                       logger_prints=self._logger.info)
             self._agents.append(agent)
         return job
 

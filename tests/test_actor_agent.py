@@ -82,3 +82,40 @@ def test_agent_stat_data():
     agent.reset()
     data = agent.get_stat_data()
     assert 'race_id' in data and 'z_type' in data
+
+
+@pytest.mark.timeout(600)
+def test_actor_respects_pipeline_field():
+    """Jobs carrying a non-default pipeline load that agent package through
+    the actor's own job setup (docs/agent.md contract)."""
+
+    class StubComm:
+        def ask_for_job(self, job_type):
+            return {'player_ids': ['MP0', 'TPL'],
+                    'pipelines': ['default', 'template'],
+                    'checkpoint_paths': ['none', 'none'],
+                    'teacher_checkpoint_paths': ['none', 'none'],
+                    'z_path': ['3map.json', '3map.json'],
+                    'send_data_players': [], 'update_players': [],
+                    'env_info': {'map_name': 'KingsCove'}}
+
+        def send_result(self, result):
+            return {'ok': True}
+
+        def send_data(self, *a, **k):
+            pass
+
+        def pull_model(self, *a, **k):
+            return None
+
+    cfg = Config({'actor': {'episode_num': 1, 'env_type': 'mock'},
+                  'env': {'player_num': 2, 'max_episode_steps': 4},
+                  'common': {'experiment_name': 'test_actor_tpl', 'type': 'train'}})
+    actor = Actor(cfg)
+    actor._comm = StubComm()
+    results = actor.run()
+    assert len(results) == 1
+    from distar_amd.agents.template.agent import Agent as TemplateAgent
+    from distar_amd.actor.agent import Agent as DefaultAgent
+    assert isinstance(actor._agents[0], DefaultAgent)
+    assert isinstance(actor._agents[1], TemplateAgent)
