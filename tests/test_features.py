@@ -98,3 +98,31 @@ def test_battle_score():
     _, raw_ob, _ = make_features()
     # killed minerals 150 + 1.5 * killed vespene 150 (same dummy category sums)
     assert compute_battle_score(raw_ob) == 150. + 1.5 * 150.
+
+
+def test_transform_obs_value_features():
+    """Opponent-side value features (reference features.py:735-765) flow into
+    the critic encoder."""
+    feat, raw_ob, units = make_features()
+    opp_units = [D.unit(tag=900 + i, unit_type=105, alliance=1, x=100 + i, y=80)
+                 for i in range(4)]
+    opp_ob = D.raw_observation(opp_units, player_id=2)
+    out = feat.transform_obs(raw_ob, padding_spatial=True, opponent_obs=opp_ob)
+    vf = out['value_feature']
+    assert int(vf['total_unit_count']) == 4 + 3     # enemy units + own units
+    assert vf['unit_x'].shape[0] == 512
+    assert vf['own_units_spatial'].shape == (1, 152, 160)
+    assert int(vf['enemy_unit_counts_bow'].sum()) == 4
+    # feeds the ValueEncoder (with the behavior-Z keys the agent appends)
+    import torch
+    from distar_amd.models.alphastar.encoders import ValueEncoder
+    from distar_amd.models.alphastar.model import alphastar_model_default_config
+    from distar_amd.utils.data import default_collate_with_dim
+    enc = ValueEncoder(alphastar_model_default_config)
+    vf = dict(vf)
+    vf.update({'beginning_order': torch.zeros(20, dtype=torch.long),
+               'bo_location': torch.zeros(20, dtype=torch.long),
+               'cumulative_stat': torch.zeros(167, dtype=torch.long)})
+    batch = default_collate_with_dim([vf, vf])
+    out_v = enc(batch)
+    assert out_v.shape == (2, 544)

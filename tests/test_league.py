@@ -155,3 +155,36 @@ def test_league_http_api(league):
         assert 'ratings' in elo
     finally:
         server.stop()
+
+
+def test_vs_bot_and_ladder_jobs(league):
+    league.cfg.vs_bot = True
+    league.cfg.bot_probs = [1, 1]
+    job = league.deal_with_actor_ask_for_job({'job_type': 'train'})
+    assert job['branch'] == 'train_bot'
+    assert job['env_info']['player_ids'][1].startswith('bot')
+    league.cfg.vs_bot = False
+    # ladder needs historical players
+    league.deal_with_learner_send_train_info(
+        {'player_id': 'MP0', 'train_steps': 1500, 'checkpoint_path': 'mp0.pth'})
+    league.deal_with_learner_send_train_info(
+        {'player_id': 'MP0', 'train_steps': 1500, 'checkpoint_path': 'mp0.pth'})
+    job = league.deal_with_actor_ask_for_job({'job_type': 'ladder'})
+    assert job['branch'] == 'ladder'
+    assert job['send_data_players'] == []
+    assert len(job['player_ids']) == 2
+
+
+def test_trueskill_updates_on_results(league):
+    import time
+    for _ in range(5):
+        league.deal_with_actor_send_result({
+            'game_steps': 10, 'game_iters': 1, 'game_duration': 5,
+            '0': {'player_id': 'MP0', 'opponent_id': 'EP0', 'winloss': 1,
+                  'race_id': 1, 'z_type': 0},
+            '1': {'player_id': 'EP0', 'opponent_id': 'MP0', 'winloss': -1,
+                  'race_id': 1, 'z_type': 0}})
+    deadline = time.time() + 20
+    while league.elo.game_count < 5 and time.time() < deadline:
+        time.sleep(0.05)
+    assert league.trueskill.mu['MP0'] > league.trueskill.mu['EP0']
