@@ -8,7 +8,9 @@ detect the codec.  'nppickle' converts torch tensors to numpy first —
 cheaper to pickle and what the RL trajectory path uses.
 """
 import io
+import os
 import pickle
+import sys
 import zlib
 
 import numpy as np
@@ -16,6 +18,26 @@ import torch
 
 _MAGIC_ZLIB = b'DAZ1'
 _MAGIC_RAW = b'DAR0'
+
+# native GIL-released zlib codec (ops/cpp/native_codec.cpp); the pure-python
+# zlib module is the fallback when the .so has not been built
+_OPS_DIR = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), 'ops')
+if _OPS_DIR not in sys.path:
+    sys.path.insert(0, _OPS_DIR)
+try:
+    import _native_codec
+
+    def _compress(payload, level):
+        return _native_codec.compress(payload, level)
+
+    def _decompress(payload):
+        return _native_codec.decompress(payload)
+except ImportError:
+    def _compress(payload, level):
+        return zlib.compress(payload, level)
+
+    def _decompress(payload):
+        return zlib.decompress(payload)
 
 
 def _tensor_to_np(data):
@@ -57,14 +79,14 @@ def dumps(data, fs_type='nppickle', compress=True, level=1):
     else:
         raise KeyError(fs_type)
     if compress:
-        return _MAGIC_ZLIB + zlib.compress(payload, level)
+        return _MAGIC_ZLIB + _compress(payload, level)
     return _MAGIC_RAW + payload
 
 
 def loads(blob, fs_type='nppickle'):
     magic, payload = blob[:4], blob[4:]
     if magic == _MAGIC_ZLIB:
-        payload = zlib.decompress(payload)
+        payload = _decompress(payload)
     elif magic != _MAGIC_RAW:
         payload = blob      # uncompressed legacy payload
     if fs_type in ('pickle', 'cPickle'):

@@ -29,7 +29,22 @@ SOURCES = [
 ]
 
 
+def build_native_codec(verbose=True):
+    """CPU-only pybind codec (no hipcc needed): g++ + zlib."""
+    import pybind11
+    src = os.path.join(OPS, 'cpp', 'native_codec.cpp')
+    out = os.path.join(OPS, '_native_codec.so')
+    py_inc = sysconfig.get_paths()['include']
+    cmd = ['g++', '-O3', '-std=c++17', '-shared', '-fPIC', src, '-o', out,
+           f'-I{pybind11.get_include()}', f'-I{py_inc}', '-lz']
+    if verbose:
+        print(' '.join(cmd))
+    subprocess.check_call(cmd, cwd=ROOT)
+    return out
+
+
 def build(verbose=True):
+    build_native_codec(verbose=verbose)
     sources = [s for s in SOURCES if os.path.exists(s)]
     torch_inc = cpp_ext.include_paths()
     torch_lib = cpp_ext.library_paths()
