@@ -79,3 +79,42 @@ def test_adapter_bounded_queue_drops_oldest():
         assert vals == [2, 3]
     finally:
         coord.close()
+
+
+@pytest.mark.timeout(120)
+def test_rl_dataloader_pulls_and_collates():
+    """RLDataLoader end-to-end on CPU: trajectories pushed through the
+    Adapter come out as collated learner batches."""
+    import random
+    from distar_amd.data.rl_dataloader import RLDataLoader
+    from distar_amd.lib.fake_data import fake_obs_step, fake_rl_step
+    from distar_amd.utils.config import Config
+    torch.manual_seed(0)
+    coord = Coordinator().run()
+    loader = None
+    try:
+        producer = Adapter(coordinator_port=coord.port)
+        rng = random.Random(0)
+        for _ in range(3):
+            steps = [fake_rl_step(48, rng) for _ in range(2)]
+            last = fake_obs_step(entity_num=48)
+            last['hidden_state'] = [(torch.zeros(384), torch.zeros(384))
+                                    for _ in range(3)]
+            for k in last['entity_info']:
+                last['entity_info'][k] = last['entity_info'][k][:48]
+            producer.push(steps + [last], token='MP0traj', fs_type='nppickle')
+        cfg = Config({'learner': {'player_id': 'MP0',
+                                  'data': {'batch_size': 2, 'buffer_size': 2,
+                                           'use_async_cuda': False}},
+                      'communication': {'coordinator_ip': '127.0.0.1',
+                                        'coordinator_port': coord.port,
+                                        'adapter_traj_worker_num': 1}})
+        loader = RLDataLoader(cfg)
+        batch = next(loader)
+        assert batch['batch_size'] == 2 and batch['unroll_len'] == 2
+        assert batch['entity_num'].shape[0] == (2 + 1) * 2   # (T+1)*B rows
+        assert batch['action_info']['action_type'].shape == (2, 2)
+    finally:
+        if loader is not None:
+            loader.close()
+        coord.close()
