@@ -108,6 +108,24 @@ class Actor:
             return SC2Env(self._whole_cfg)
         return MockSC2Env(self._whole_cfg)
 
+    def _start_batch_inference(self):
+        """Shared-slab batched inference (reference actor.py:268-299): one
+        device model serves every env worker; agents switch to writing their
+        obs into the slab and polling the signal."""
+        import threading
+        from .batch_inference import BatchInferenceServer
+        device = 'cuda' if (self._cfg.use_cuda and torch.cuda.is_available()) \
+            else 'cpu'
+        agent0 = self._agents[0]
+        server = BatchInferenceServer(agent0.model.to(device),
+                                      env_num=len(self._agents), device=device)
+        for env_id, agent in enumerate(self._agents):
+            agent.attach_batch_inference(server, env_id)
+        self._batch_server = server
+        self._batch_thread = threading.Thread(target=server.run, daemon=True)
+        self._batch_thread.start()
+        return server
+
     def _update_models(self):
         if self._comm is None:
             return
@@ -176,6 +194,8 @@ class Actor:
     # ------------------------------------------------------------------ run
     def run(self):
         self._setup_job()
+        if self._cfg.get('gpu_batch_inference', False):
+            self._start_batch_inference()
         env = self._make_env()
         episode_num = self._cfg.episode_num
         job_deadline = time.time() + self._cfg.actor_ask_for_job_interval * \

@@ -239,6 +239,44 @@ class Agent:
             return {key: self.decollate_output(v, key) for key, v in output.items()}
         return output
 
+    def attach_batch_inference(self, server, env_id):
+        """Switch to shared-slab batched inference (reference
+        agent.py:128-141): obs go into the server's input slab; outputs are
+        read back per slot after the server tick."""
+        self._batch_server = server
+        self._env_id = env_id
+
+    def _batched_infer(self, agent_obs):
+        import time as _time
+        from .batch_inference import copy_input_data
+        server = self._batch_server
+        copy_input_data(server.shared_input, agent_obs, data_idx=self._env_id)
+        server.signals[self._env_id] += 1
+        while int(server.signals[self._env_id]) != 0:
+            _time.sleep(0.001)
+        out = server.shared_output
+        idx = self._env_id
+        result = {
+            'action_info': {k: v[idx].clone() for k, v in out['action_info'].items()},
+            'action_logp': {k: v[idx].clone() for k, v in out['action_logp'].items()},
+            'logit': {k: v[idx].clone() for k, v in out['logit'].items()},
+            'selected_units_num': out['selected_units_num'][idx].clone(),
+            'entity_num': out['entity_num'][idx].clone(),
+            'extra_units': out['extra_units'][idx].clone(),
+            'hidden_state': [(out['hidden_state'][l][0][idx].clone(),
+                              out['hidden_state'][l][1][idx].clone())
+                             for l in range(len(out['hidden_state']))],
+        }
+        en, su = int(result['entity_num']), int(result['selected_units_num'])
+        result['logit']['selected_units'] = \
+            result['logit']['selected_units'][:max(su, 1), :en + 1]
+        result['logit']['target_unit'] = result['logit']['target_unit'][:en]
+        result['action_info']['selected_units'] = \
+            result['action_info']['selected_units'][:max(su, 1)]
+        result['action_logp']['selected_units'] = \
+            result['action_logp']['selected_units'][:max(su, 1)]
+        return result
+
     def step(self, observation):
         if 'eval' in self._job_type and self._iter_count > 0:
             self._update_fake_reward(int(self._last_action_type),
@@ -246,6 +284,12 @@ class Agent:
         model_input = self._pre_process(observation)
         self._stat_api.update(int(self._last_action_type),
                               (observation.get('action_result') or [1])[0])
+        if getattr(self, '_batch_server', None) is not None:
+            obs_for_slab = dict(self._observation)
+            output = self._batched_infer(obs_for_slab)
+            action = self._post_process(output)
+            self._iter_count += 1
+            return action
         with torch.no_grad():
             model_output = self.model.compute_logp_action(**model_input)
         action = self._post_process(self.decollate_output(model_output))

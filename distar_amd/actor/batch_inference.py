@@ -87,10 +87,18 @@ class BatchInferenceServer:
             self.teacher_signals = torch.zeros(env_num, dtype=torch.long).share_memory_()
         self._stop = False
 
-    def serve_once(self, signals, shared_input, shared_output, forward):
-        """One tick: when every live slot has signalled, run the batch."""
-        if int((signals > 0).sum()) < self.env_num:
+    def serve_once(self, signals, shared_input, shared_output, forward,
+                   force=False):
+        """One tick when every slot has signalled — or, with ``force``, when
+        at least one has (partial batches keep a serial/in-process actor
+        from deadlocking; un-signalled slots' outputs are recomputed stale
+        values nobody reads)."""
+        pending = signals > 0
+        if not bool(pending.any()):
             return False
+        if not force and int(pending.sum()) < self.env_num:
+            return False
+        ticked = pending.clone()
         batch = {k: v for k, v in shared_input.items() if k != 'hidden_state'}
         batch = to_device(batch, self.device)
         batch['hidden_state'] = [
@@ -102,19 +110,30 @@ class BatchInferenceServer:
                                              enabled=use_amp):
             output = forward(**batch)
         copy_output_data(shared_output, output, self.env_num)
-        signals.zero_()
+        signals[ticked] = 0
         return True
 
-    def run(self, poll_interval=0.002):
+    def run(self, poll_interval=0.002, partial_after=0.01):
+        pending_since = None
         while not self._stop:
+            any_pending = bool((self.signals > 0).any())
+            force = False
+            if any_pending:
+                if pending_since is None:
+                    pending_since = time.time()
+                force = time.time() - pending_since > partial_after
             ticked = self.serve_once(self.signals, self.shared_input,
                                      self.shared_output,
-                                     self.model.compute_logp_action)
+                                     self.model.compute_logp_action,
+                                     force=force)
+            if ticked:
+                pending_since = None
             if self.teacher_model is not None:
                 ticked |= self.serve_once(self.teacher_signals,
                                           self.teacher_input,
                                           self.teacher_output,
-                                          self.teacher_model.compute_teacher_logit)
+                                          self.teacher_model.compute_teacher_logit,
+                                          force=force)
             if not ticked:
                 time.sleep(poll_interval)
 

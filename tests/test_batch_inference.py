@@ -38,3 +38,22 @@ def test_batch_inference_slab_roundtrip():
     assert out['action_info']['action_type'].shape == (env_num,)
     assert torch.isfinite(out['logit']['action_type']).all()
     assert (out['selected_units_num'] <= 64).all()
+
+
+@pytest.mark.timeout(900)
+def test_actor_batched_inference_mode():
+    """Actor with gpu_batch_inference: agents route through the shared slab
+    server (partial-batch ticks keep the serial in-process loop live)."""
+    torch.manual_seed(0)
+    from distar_amd.actor.actor import Actor
+    cfg = Config({'actor': {'episode_num': 1, 'env_type': 'mock',
+                            'gpu_batch_inference': True, 'use_cuda': False,
+                            'traj_len': 4},
+                  'env': {'player_num': 2, 'max_episode_steps': 4},
+                  'common': {'experiment_name': 'test_actor_slab',
+                             'type': 'train'}})
+    actor = Actor(cfg)
+    results = actor.run()
+    actor._batch_server.stop()
+    assert len(results) == 1
+    assert all(agent._batch_server is not None for agent in actor._agents)
