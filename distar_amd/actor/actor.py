@@ -141,9 +141,10 @@ class Actor:
                     agent.set_model_last_iter(payload.get('model_last_iter', 0))
 
     # -------------------------------------------------------------- episode
-    def _run_episode(self, env):
+    def _run_episode(self, env, agents=None):
+        agents = agents if agents is not None else self._agents
         obs = env.reset()
-        for i, agent in enumerate(self._agents):
+        for i, agent in enumerate(agents):
             agent.reset(map_name=self._job['env_info'].get('map_name', 'KingsCove'),
                         race='zerg', opponent_race='zerg', obs=obs.get(i))
         done = False
@@ -152,13 +153,13 @@ class Actor:
         while not done and not self._end:
             t0 = time.time()
             actions = {}
-            for i, agent in enumerate(self._agents):
+            for i, agent in enumerate(agents):
                 if i in last_obs:
                     actions[i] = agent.step(last_obs[i])[0]
             t1 = time.time()
             obs, rewards, done, infos = env.step(actions)
             t2 = time.time()
-            for i, agent in enumerate(self._agents):
+            for i, agent in enumerate(agents):
                 if i not in last_obs:
                     continue
                 traj = agent.collect_data(obs.get(i), rewards.get(i, 0), done, i)
@@ -172,18 +173,19 @@ class Actor:
             last_obs = {**last_obs, **obs}
             self._update_models()
         self._record.update_var({'episode_steps': episode_steps})
-        result = self._build_result(rewards)
+        result = self._build_result(rewards, agents)
         if self._comm is not None:
             self._comm.send_result(result)
         self.results.append(result)
         return result
 
-    def _build_result(self, rewards):
-        result = {'game_steps': self._agents[0]._game_step,
-                  'game_iters': self._agents[0]._iter_count,
+    def _build_result(self, rewards, agents=None):
+        agents = agents if agents is not None else self._agents
+        result = {'game_steps': agents[0]._game_step,
+                  'game_iters': agents[0]._iter_count,
                   'game_duration': 0}
         ids = self._job['player_ids']
-        for i, agent in enumerate(self._agents):
+        for i, agent in enumerate(agents):
             opp = ids[1 - i] if len(ids) > 1 else ids[0]
             side = {'player_id': agent.player_id, 'opponent_id': opp,
                     'winloss': float(rewards.get(i, 0))}
@@ -192,17 +194,15 @@ class Actor:
         return result
 
     # ------------------------------------------------------------------ run
-    def run(self):
-        self._setup_job()
-        if self._cfg.get('gpu_batch_inference', False):
-            self._start_batch_inference()
+    def _env_loop(self, agents, episode_num):
+        """One environment worker: episodes until quota/stop (the reference
+        forks `env_num` processes, `actor.py:301-319`; here each worker is a
+        thread sharing the job's model weights — inference is no-grad and
+        per-agent recurrent state lives on the Agent)."""
         env = self._make_env()
-        episode_num = self._cfg.episode_num
-        job_deadline = time.time() + self._cfg.actor_ask_for_job_interval * \
-            (1 + 0.3 * (2 * random.random() - 1))
         while not self._end and (episode_num < 0 or self.episodes_done < episode_num):
             try:
-                self._run_episode(env)
+                self._run_episode(env, agents)
                 self.episodes_done += 1
             except Exception:  # noqa: BLE001 - actors are cattle
                 traceback.print_exc()
@@ -211,10 +211,59 @@ class Actor:
                 except Exception:  # noqa: BLE001
                     pass
                 env = self._make_env()
-            if time.time() > job_deadline:
-                self._setup_job()
-                job_deadline = time.time() + self._cfg.actor_ask_for_job_interval
         env.close()
+
+    def _clone_agents(self, env_id):
+        """Per-env agent wrappers sharing the job's model objects."""
+        clones = []
+        for agent in self._agents:
+            clone = type(agent)(self._whole_cfg, env_id=env_id)
+            clone.player_id = agent.player_id
+            if getattr(agent, 'HAS_MODEL', False):
+                clone.model = agent.model
+                clone.teacher_model = agent.teacher_model
+            clones.append(clone)
+        return clones
+
+    def run(self):
+        import threading
+        self._setup_job()
+        if self._cfg.get('gpu_batch_inference', False):
+            self._start_batch_inference()
+        env_num = self._cfg.env_num
+        episode_num = self._cfg.episode_num
+        job_deadline = time.time() + self._cfg.actor_ask_for_job_interval * \
+            (1 + 0.3 * (2 * random.random() - 1))
+        if env_num <= 1:
+            env = self._make_env()
+            while not self._end and \
+                    (episode_num < 0 or self.episodes_done < episode_num):
+                try:
+                    self._run_episode(env)
+                    self.episodes_done += 1
+                except Exception:  # noqa: BLE001 - actors are cattle
+                    traceback.print_exc()
+                    try:
+                        env.close()
+                    except Exception:  # noqa: BLE001
+                        pass
+                    env = self._make_env()
+                if time.time() > job_deadline:
+                    self._setup_job()
+                    job_deadline = time.time() + self._cfg.actor_ask_for_job_interval
+            env.close()
+            return self.results
+        workers = []
+        for env_id in range(env_num):
+            agents = self._clone_agents(env_id) if env_id else self._agents
+            t = threading.Thread(target=self._env_loop,
+                                 args=(agents, episode_num), daemon=True)
+            t.start()
+            workers.append(t)
+        while any(t.is_alive() for t in workers):
+            for t in workers:
+                t.join(timeout=0.5)
+            self._update_models()
         return self.results
 
     def close(self):

@@ -119,3 +119,20 @@ def test_actor_respects_pipeline_field():
     from distar_amd.actor.agent import Agent as DefaultAgent
     assert isinstance(actor._agents[0], DefaultAgent)
     assert isinstance(actor._agents[1], TemplateAgent)
+
+
+@pytest.mark.timeout(900)
+def test_actor_multi_env_workers():
+    """env_num > 1: parallel environment workers sharing the job's models
+    (reference forks env_num processes, actor.py:301-319)."""
+    torch.manual_seed(0)
+    cfg = Config({'actor': {'episode_num': 3, 'env_num': 2, 'env_type': 'mock',
+                            'traj_len': 4},
+                  'env': {'player_num': 2, 'max_episode_steps': 4},
+                  'common': {'experiment_name': 'test_actor_multi',
+                             'type': 'train'}})
+    actor = Actor(cfg)
+    results = actor.run()
+    assert len(results) >= 3
+    for r in results:
+        assert '0' in r and '1' in r
