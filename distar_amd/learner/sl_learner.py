@@ -28,6 +28,15 @@ class SLLearner(BaseLearner):
             zero = zero.cuda()
         self.hidden_state = [(zero, zero) for _ in range(self.num_layers)]
         self.ignore_step = 0
+        # debug mode (reference sl_learner.py:25-29,55-60): EMA-track the five
+        # per-head losses; a 10x spike after iteration 200 snapshots the bad
+        # batch + checkpoint for offline repro
+        self.debug = self._whole_cfg.learner.get('debug', False)
+        if self.debug:
+            self.debug_loss = {k: 0. for k in (
+                'action_type_loss', 'delay_loss', 'selected_units_loss_norm',
+                'target_unit_loss', 'target_location_loss')}
+            self.debug_min_iter = self._whole_cfg.learner.get('debug_min_iter', 200)
 
     def reset_hidden_state(self, new_episodes):
         for l in range(self.num_layers):
@@ -60,6 +69,8 @@ class SLLearner(BaseLearner):
                     data['selected_units_num'], data['entity_num'], infer_action_info)
             loss = log_vars['total_loss']
         self._log_buffer['forward_time'] = self._timer.value
+        if self.debug:
+            self._debug_check(data, log_vars, logits)
 
         with self._timer:
             if self.ignore_step > 5:
@@ -78,3 +89,23 @@ class SLLearner(BaseLearner):
         self._log_buffer['backward_time'] = self._timer.value
         self._log_buffer.update({k: (v.item() if isinstance(v, torch.Tensor) else v)
                                  for k, v in log_vars.items()})
+
+    def _debug_check(self, data, log_vars, logits):
+        """Snapshot spiking batches (reference sl_learner.py:55-60)."""
+        for k in self.debug_loss:
+            prev = self.debug_loss[k]
+            cur = float(log_vars[k])
+            self.debug_loss[k] = prev * 0.95 + cur * 0.05
+            if prev > 0 and cur > prev * 10 and self.last_iter.val > self.debug_min_iter:
+                self.save_checkpoint()
+                import os
+                path = os.path.join(
+                    self._exp_dir,
+                    f'debug_{k}_iter_{self.last_iter.val}_rank_{self._rank}.pth')
+                torch.save({'data': data, 'hidden_state': self.hidden_state,
+                            'log_vars': {n: float(v) for n, v in log_vars.items()},
+                            'logits': {n: (v.detach().cpu() if isinstance(v, torch.Tensor) else v)
+                                       for n, v in logits.items()}},
+                           path)
+                self.info(f'[debug] {k} spiked {cur:.3g} (ema {prev:.3g}); '
+                          f'snapshot -> {path}')

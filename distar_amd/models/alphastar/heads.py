@@ -384,6 +384,7 @@ class LocationHead(nn.Module):
                                   activation=build_activation(self.cfg.activation),
                                   norm_type=None)
         self.res_dim = self.cfg.res_dim
+        self.use_film = self.cfg.get('film', False)
         self.use_gate = self.cfg.get('gate', False)
         self.use_unet = self.cfg.get('unet', False)
         spatial_y = self.whole_cfg.model.spatial_y
@@ -391,6 +392,15 @@ class LocationHead(nn.Module):
         self.project_embed = fc_block(
             self.cfg.input_dim, (spatial_y // 8) * (spatial_x // 8) * 4,
             activation=build_activation(self.cfg.activation))
+        if self.use_film:
+            # FiLM conditioning of each res stage on the LSTM embedding
+            # (reference action_arg_head.py:385-402)
+            from ..nn.blocks import FiLMedResBlock
+            self.film_fc = fc_block(self.cfg.input_dim, self.res_dim,
+                                    activation=build_activation(self.cfg.activation))
+            self.film_gamma = nn.ModuleList()
+            self.film_beta = nn.ModuleList()
+            self.film = nn.ModuleList()
         self.res = nn.ModuleList()
         for _ in range(self.cfg.res_num):
             if self.use_gate:
@@ -400,6 +410,14 @@ class LocationHead(nn.Module):
             else:
                 self.res.append(ResBlock(self.res_dim, build_activation(self.cfg.activation),
                                          norm_type=None))
+            if self.use_film:
+                g = nn.Linear(self.res_dim, self.res_dim)
+                b = nn.Linear(self.res_dim, self.res_dim)
+                nn.init.xavier_uniform_(g.weight)
+                nn.init.xavier_uniform_(b.weight)
+                self.film_gamma.append(g)
+                self.film_beta.append(b)
+                self.film.append(FiLMedResBlock(self.res_dim, with_cond=[True]))
         self.upsample = nn.ModuleList()
         dims = [self.res_dim] + list(self.cfg.upsample_dims)
         assert self.cfg.upsample_type in ('deconv', 'nearest', 'bilinear')
@@ -422,9 +440,13 @@ class LocationHead(nn.Module):
                                      spatial_y // 8, spatial_x // 8)
         cat_feature = torch.cat([reshaped, map_skip[-1]], dim=1)
         x = self.conv1(self.act(cat_feature))
+        film_embedding = self.film_fc(embedding) if self.use_film else None
         for i in range(len(self.res)):
             x = x + map_skip[len(map_skip) - i - 1]
             x = self.res[i](x, x) if self.use_gate else self.res[i](x)
+            if self.use_film:
+                x = self.film[i](x, gammas=self.film_gamma[i](film_embedding),
+                                 betas=self.film_beta[i](film_embedding))
         from ...ops.upsample import upsample2x_bilinear
         for i, layer in enumerate(self.upsample):
             if self.cfg.upsample_type == 'nearest':

@@ -231,6 +231,49 @@ class GatedResBlock(nn.Module):
         return self.act(x + residual)
 
 
+class FiLM(nn.Module):
+    """Feature-wise linear modulation (reference `module_utils.py:234-241`):
+    per-channel affine conditioning ``gamma * x + beta`` with (N, C) gammas
+    and betas broadcast over the spatial axes."""
+
+    def forward(self, x, gammas, betas):
+        return torch.addcmul(betas[:, :, None, None], gammas[:, :, None, None], x)
+
+
+class FiLMedResBlock(nn.Module):
+    """Conditioned residual conv block (reference `module_utils.py:244-353`).
+
+    Only the configuration the reference actually instantiates is kept
+    (`action_arg_head.py:400`: ``FiLMedResBlock(dim, with_cond=[True])``,
+    i.e. 3x3 input projection + 3x3 conv + conv-film + residual relu);
+    parameter names match for checkpoint-key parity.
+    """
+
+    def __init__(self, in_dim, out_dim=None, with_residual=True,
+                 with_cond=(True,), kernel_size=3, with_input_proj=3):
+        super().__init__()
+        out_dim = out_dim or in_dim
+        assert kernel_size % 2 == 1 and with_input_proj % 2 == 1
+        self.with_residual = with_residual
+        self.with_cond = list(with_cond)
+        self.input_proj = nn.Conv2d(in_dim, in_dim, kernel_size=with_input_proj,
+                                    padding=with_input_proj // 2)
+        self.conv1 = nn.Conv2d(in_dim, out_dim, kernel_size=kernel_size,
+                               padding=kernel_size // 2)
+        self.film = FiLM() if self.with_cond[0] else None
+        for m in (self.input_proj, self.conv1):
+            nn.init.kaiming_normal_(m.weight)
+
+    def forward(self, x, gammas=None, betas=None):
+        x = F.relu(self.input_proj(x))
+        out = self.conv1(x)
+        if self.film is not None:
+            out = self.film(out, gammas, betas)
+        if self.with_residual:
+            out = F.relu(x + out)
+        return out
+
+
 # ------------------------------------------------------------------ helpers
 
 def sequence_mask(lengths: torch.Tensor, max_len=None) -> torch.Tensor:

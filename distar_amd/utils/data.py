@@ -124,3 +124,69 @@ def flat(data):
     if isinstance(data, Sequence) and not isinstance(data, str):
         return [flat(v) for v in data]
     raise TypeError(type(data))
+
+
+class CudaFetcher:
+    """Background H2D prefetcher (reference `ctools/torch_utils/data_helper.py:203-231`):
+    a producer thread pulls from ``data_source``, stages each batch to
+    ``device`` on its own HIP stream (pinned-host copies overlap compute on
+    the default stream), and parks results in a bounded queue.
+
+    The RL learner path uses the richer ``data.rl_dataloader._DeviceStager``
+    (per-batch events); this class is the drop-in generic fetcher for any
+    iterator."""
+
+    def __init__(self, data_source, device, queue_size=4, sleep=0.1):
+        import queue as _queue
+        import threading
+        self._source = data_source
+        self._queue = _queue.Queue(maxsize=queue_size)
+        self._device = device
+        self._sleep = sleep
+        self._end_flag = True
+        self._stream = None
+        self._thread = threading.Thread(target=self._producer, daemon=True)
+
+    def __iter__(self):
+        return self
+
+    def __next__(self):
+        return self._queue.get()
+
+    def run(self):
+        self._end_flag = False
+        self._thread.start()
+        return self
+
+    def close(self):
+        self._end_flag = True
+
+    def _producer(self):
+        import time as _time
+        use_cuda = str(self._device).startswith('cuda') and torch.cuda.is_available()
+        if use_cuda and self._stream is None:
+            self._stream = torch.cuda.Stream()
+        ctx = torch.cuda.stream(self._stream) if use_cuda else _NullCtx()
+        with ctx:
+            while not self._end_flag:
+                if self._queue.full():
+                    _time.sleep(self._sleep)
+                    continue
+                try:
+                    data = next(self._source)
+                except StopIteration:
+                    self._end_flag = True
+                    break
+                data = to_device(data, self._device,
+                                 non_blocking=use_cuda)
+                if use_cuda:
+                    self._stream.synchronize()
+                self._queue.put(data)
+
+
+class _NullCtx:
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        return False
