@@ -61,7 +61,8 @@ class Actor:
     # ------------------------------------------------------------------ job
     def _setup_job(self):
         if self._comm is not None:
-            job = self._comm.ask_for_job(self._job_type)
+            job = self._comm.ask_for_job(
+                self._job_type, player_id=self._cfg.get('job_player_id'))
         else:
             job = {'player_ids': ['MP0', 'MP1'],
                    'pipelines': ['default', 'default'],

@@ -26,9 +26,12 @@ class ActorComm:
         self.adapter = adapter or Adapter(cfg=cfg)
         self.job = {}
 
-    def ask_for_job(self, job_type='train'):
+    def ask_for_job(self, job_type='train', player_id=None):
+        req = {'job_type': job_type}
+        if player_id:
+            req['player_id'] = player_id
         self.job = post_json(self._league_url + '/league/actor_ask_for_job',
-                             {'job_type': job_type}, retries=10, backoff=1.0)
+                             req, retries=10, backoff=1.0)
         return self.job
 
     def send_result(self, result_info):

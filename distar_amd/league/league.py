@@ -237,7 +237,13 @@ class League:
         if job_type == 'ladder':
             branch, job_info = self._get_ladder_job_info()
         else:
-            player = self.choose_active_player()
+            pinned = request_info.get('player_id')
+            if pinned and pinned in self.active_players:
+                # targeted job (test/ops facility beyond the reference's
+                # weighted sampling): train THIS active player
+                player = self.active_players[pinned]
+            else:
+                player = self.choose_active_player()
             if self.cfg.get('vs_bot', False):
                 branch, job_info = self._get_vs_bot_job_info(player)
             else:

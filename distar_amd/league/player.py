@@ -233,9 +233,12 @@ class MainPlayer(ActivePlayer):
             return branch, [self], [opponent]
         if branch == 'pfsp':
             keys = _hist_nonbot(historical_players, pfsp_train_bot)
-            assert keys, 'pfsp branch needs historical players'
+            if not keys:        # before the first snapshot: self-play
+                return 'sp', [self], [self]
             return branch, [self], [self._pfsp_pick(historical_players, keys, 'squared')]
         if branch == 'eval':
+            if not historical_players:
+                return 'sp', [self], [self]
             pid = random.choice(list(historical_players.keys()))
             return branch, [self], [historical_players[pid]]
         raise NotImplementedError(branch)
@@ -265,10 +268,15 @@ class ExploiterPlayer(ActivePlayer):
         branch_probs = branch_probs_dict[self._name]
         branch = random.choices(list(branch_probs.keys()),
                                 weights=list(branch_probs.values()), k=1)[0]
+        mains = [p for p in active_players.values() if isinstance(p, MainPlayer)]
         if branch == 'pfsp':
             keys = _hist_nonbot(historical_players, pfsp_train_bot)
+            if not keys and mains:  # before the first snapshot: vs active main
+                return 'pfsp', [self], [random.choice(mains)]
             return branch, [self], [self._pfsp_pick(historical_players, keys, 'normal')]
         if branch == 'eval':
+            if not historical_players and mains:
+                return 'eval', [self], [random.choice(mains)]
             pid = random.choice(list(historical_players.keys()))
             return branch, [self], [historical_players[pid]]
         raise NotImplementedError(branch)
@@ -297,10 +305,15 @@ class ExpertExploiterPlayer(ExploiterPlayer):
         branch_probs = branch_probs_dict[self._name]
         branch = random.choices(list(branch_probs.keys()),
                                 weights=list(branch_probs.values()), k=1)[0]
+        mains = [p for p in active_players.values() if isinstance(p, MainPlayer)]
         if branch == 'pfsp':
             keys = _hist_nonbot(historical_players, pfsp_train_bot)
+            if not keys and mains:  # before the first snapshot: vs active main
+                return 'pfsp', [self], [random.choice(mains)]
             return branch, [self], [self._pfsp_pick(historical_players, keys, 'normal')]
         if branch == 'eval':
+            if not historical_players and mains:
+                return 'eval', [self], [random.choice(mains)]
             pid = random.choice(list(historical_players.keys()))
             return branch, [self], [historical_players[pid]]
         raise NotImplementedError(branch)
