@@ -24,6 +24,7 @@ class ActorComm:
         self._league_url = 'http://{}:{}'.format(
             comm.get('league_ip', '127.0.0.1'), comm.get('league_port', 0))
         self.adapter = adapter or Adapter(cfg=cfg)
+        self._traj_compress = comm.get('traj_compress', False)
         self.job = {}
 
     def ask_for_job(self, job_type='train', player_id=None):
@@ -50,7 +51,11 @@ class ActorComm:
         token = player_id + 'traj'
         if self.adapter.full(token):
             print(f'[ActorComm] trajectory backlog full for {token}')
-        self.adapter.push(traj_data, token=token, fs_type='nppickle')
+        # zlib-1 costs ~5-10x nppickle dump time for 1.63x size (see
+        # profiles/r01_serialize.md) -> uncompressed by default; turn on for
+        # bandwidth-constrained actor fleets via communication.traj_compress
+        self.adapter.push(traj_data, token=token, fs_type='nppickle',
+                          compress=self._traj_compress)
 
 
 class LearnerComm:
@@ -61,6 +66,7 @@ class LearnerComm:
             comm.get('league_ip', '127.0.0.1'), comm.get('league_port', 0))
         self.adapter = adapter or Adapter(cfg=cfg)
         self.player_id = cfg.get('learner', {}).get('player_id', 'MP0')
+        self._traj_compress = comm.get('traj_compress', False)
         self._send_model_count = 0
 
     def register_learner(self, learner):
@@ -85,7 +91,8 @@ class LearnerComm:
         state_dict = {k: v.detach().cpu() for k, v in
                       self.strip_value_keys(model.state_dict()).items()}
         payload = {'model': state_dict, 'model_last_iter': learner.last_iter.val}
-        self.adapter.push(payload, token=self.player_id + 'model', fs_type='pickle')
+        self.adapter.push(payload, token=self.player_id + 'model',
+                          fs_type='pickle', compress=self._traj_compress)
         self._send_model_count += 1
 
     def send_train_info(self, learner):
