@@ -90,7 +90,7 @@ def test_actor_respects_pipeline_field():
     the actor's own job setup (docs/agent.md contract)."""
 
     class StubComm:
-        def ask_for_job(self, job_type):
+        def ask_for_job(self, job_type, player_id=None):
             return {'player_ids': ['MP0', 'TPL'],
                     'pipelines': ['default', 'template'],
                     'checkpoint_paths': ['none', 'none'],
