@@ -12,8 +12,13 @@ def image(data, bpp=8):
               data=data.tobytes())
 
 
+def passenger(tag, unit_type=105, health=35, health_max=35):
+    return NS(tag=tag, unit_type=unit_type, health=health, health_max=health_max,
+              shield=0, shield_max=0, energy=0, energy_max=0)
+
+
 def unit(tag, unit_type=86, alliance=1, x=30.0, y=30.0, orders=(), buffs=(),
-         health=100, health_max=100):
+         health=100, health_max=100, passengers=()):
     return NS(tag=tag, unit_type=unit_type, alliance=alliance,
               cargo_space_taken=0, build_progress=1.0, health_max=health_max,
               shield_max=0, energy_max=0, display_type=1, owner=1,
@@ -24,7 +29,7 @@ def unit(tag, unit_type=86, alliance=1, x=30.0, y=30.0, orders=(), buffs=(),
               is_hallucination=False, buff_ids=list(buffs), add_on_tag=0,
               is_active=True, attack_upgrade_level=0, armor_upgrade_level=0,
               shield_upgrade_level=0, health=health, shield=0, energy=0,
-              passengers=[])
+              passengers=list(passengers))
 
 
 def score():

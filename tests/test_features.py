@@ -126,3 +126,54 @@ def test_transform_obs_value_features():
     batch = default_collate_with_dim([vf, vf])
     out_v = enc(batch)
     assert out_v.shape == (2, 544)
+
+
+def test_transform_obs_passengers_become_entities():
+    """Units riding in transports appear as in-cargo entity rows (reference
+    features.py:544-560): carried units share the carrier's position/owner and
+    set the is_in_cargo field."""
+    gi = D.game_info()
+    carrier = D.unit(tag=200, unit_type=86, x=40, y=40,
+                     passengers=[D.passenger(tag=300), D.passenger(tag=301)])
+    raw_ob = D.raw_observation([carrier])
+    feat = Features(gi, raw_ob, Config({}))
+    out = feat.transform_obs(raw_ob, padding_spatial=True)
+    n = int(out['entity_num'])
+    assert n == 3                                   # carrier + 2 passengers
+    cargo = out['entity_info']['is_in_cargo'][:n]
+    assert cargo.tolist() == [0, 1, 1]
+    # passengers inherit the carrier's location
+    assert out['entity_info']['x'][1] == out['entity_info']['x'][0]
+    assert out['entity_info']['y'][2] == out['entity_info']['y'][0]
+
+
+def test_transform_obs_effects_encode_locations():
+    """Active effects land in the effect_* spatial rasters (reference
+    features.py:479-487), except own liberator/lurker zones."""
+    feat, _, _ = make_features()
+    psistorm, liberator = 1, 9
+    raw_ob = D.raw_observation(
+        [D.unit(tag=100)],
+        effects=[(psistorm, 2, 50, 60), (liberator, 1, 30, 30),
+                 (liberator, 2, 70, 80)])
+    out = feat.transform_obs(raw_ob, padding_spatial=True)
+    storm = out['spatial_info']['effect_PsiStorm']
+    assert storm.shape == (EFFECT_LEN,)
+    loc = int(50) + int(feat.map_size.y - 60) * 160
+    assert loc in storm.tolist()
+    # own (owner==1) LiberatorDefenderZone is filtered, enemy kept
+    lib = out['spatial_info']['effect_LiberatorDefenderZone'].tolist()
+    own_loc = 30 + int(feat.map_size.y - 30) * 160
+    enemy_loc = 70 + int(feat.map_size.y - 80) * 160
+    assert enemy_loc in lib and own_loc not in lib
+
+
+def test_transform_obs_upgrades_one_hot():
+    """Researched upgrade ids scatter into the NUM_UPGRADES one-hot
+    (reference features.py:505-512)."""
+    feat, _, _ = make_features()
+    raw_ob = D.raw_observation([D.unit(tag=100)], upgrades=(1, 3))
+    out = feat.transform_obs(raw_ob, padding_spatial=True)
+    up = out['scalar_info']['upgrades']
+    assert int(up.sum()) == 2
+    assert up[1] == 1 and up[3] == 1
