@@ -57,3 +57,9 @@ class FakeRLDataloader:
 
     def __next__(self):
         return dict(next(self._cycle))
+
+    def close(self):
+        pass
+
+    def reset_comm(self):
+        pass

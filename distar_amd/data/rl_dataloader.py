@@ -119,3 +119,8 @@ class RLDataLoader:
 
     def close(self):
         self._stop = True
+
+    def reset_comm(self):
+        """Rebuild the Adapter (drops stale producer metadata after a
+        league reset; reference rl_learner.py reset_comm endpoint)."""
+        self.adapter = Adapter(cfg=self._whole_cfg)

@@ -51,6 +51,8 @@ class RLLearner(BaseLearner):
             self._whole_cfg.learner.get('value_pretrain_iters', -1)
         self._reset_value_flag = False
         self._update_config_flag = False
+        self._config_update = None
+        self._debug_server = None
 
     def _setup_loss(self):
         self._loss = ReinforcementLoss(self._whole_cfg.learner, self._player_id)
@@ -92,7 +94,54 @@ class RLLearner(BaseLearner):
             m.only_update_baseline = False
             self.info('value pretrain done, policy updates enabled')
 
+    # ---------------------------------------------------------------- debug
+    def start_debug_server(self, host='127.0.0.1', port=None):
+        """Live-control endpoints (reference rl_learner.py:263-287 serves a
+        Flask debug app): POST /learner/update_config {overrides} deep-merges
+        config and rebuilds the loss at the next iteration boundary;
+        /learner/reset_value reinitializes + rebroadcasts the value networks;
+        /learner/reset_comm restarts the dataloader's adapter pulls."""
+        from ..utils.http import JsonHttpServer
+
+        def _update_config(body):
+            self._config_update = body.get('overrides', {})
+            self._update_config_flag = True
+            return {'done': True}
+
+        def _reset_value(body):
+            self._reset_value_flag = True
+            return {'done': True}
+
+        def _reset_comm(body):
+            loader = getattr(self, '_dataloader', None)
+            if loader is not None and hasattr(loader, 'reset_comm'):
+                loader.reset_comm()
+            return {'done': True}
+
+        self._debug_server = JsonHttpServer(
+            {'/learner/update_config': _update_config,
+             '/learner/reset_value': _reset_value,
+             '/learner/reset_comm': _reset_comm}, host=host, port=port)
+        self._debug_server.start()
+        self.info(f'debug server on {host}:{self._debug_server.port}')
+        return self._debug_server
+
+    def _apply_debug_flags(self):
+        if self._update_config_flag:
+            self._update_config_flag = False
+            from ..utils.config import Config, deep_merge_dicts
+            if self._config_update:
+                self._whole_cfg = deep_merge_dicts(
+                    self._whole_cfg, Config(self._config_update))
+            self._setup_loss()
+            self.info(f'config updated: {self._config_update}')
+            self._config_update = None
+        if self._reset_value_flag:
+            self._reset_value_flag = False
+            self.reset_value()
+
     def _train(self, data):
+        self._apply_debug_flags()
         with self._timer:
             self.step_value_pretrain()
             model_last_iter = data.pop('model_last_iter', None)

@@ -121,3 +121,35 @@ def test_sl_learner_learns_fixed_batch(tmp_path):
         opt.step()
         losses.append(float(ld['total_loss'].detach()))
     assert losses[-1] < losses[0] * 0.5, losses
+
+
+@pytest.mark.timeout(300)
+def test_rl_learner_debug_endpoints(tmp_path, monkeypatch):
+    """Live-control HTTP endpoints (reference rl_learner.py:263-287):
+    update_config rebuilds the loss with merged overrides, reset_value
+    reinitializes value networks at the next iteration."""
+    monkeypatch.chdir(tmp_path)
+    import requests
+    from distar_amd.learner.rl_learner import RLLearner
+    torch.manual_seed(0)
+    cfg = Config({'learner': {'player_id': 'MP0', 'job_type': 'fake',
+                              'use_cuda': False, 'use_amp': False,
+                              'data': {'batch_size': 2, 'trajectory_length': 3},
+                              'hook': {'after_iter': {
+                                  'log_show': {'ext_args': {'freq': 1000}}}}},
+                  'common': {'experiment_name': 'test_debug_ep', 'type': 'train'},
+                  'model': {'enable_baselines': ['winloss']}})
+    learner = RLLearner(cfg)
+    srv = learner.start_debug_server()
+    url = f'http://127.0.0.1:{srv.port}'
+    old_weight = float(learner._loss.loss_weights.kl)
+    r = requests.post(f'{url}/learner/update_config', json={
+        'overrides': {'learner': {'loss_weights': {'kl': old_weight * 2}}}})
+    assert r.json().get('done')
+    r = requests.post(f'{url}/learner/reset_value', json={})
+    assert r.json().get('done')
+    learner.run(max_iterations=1)
+    assert float(learner._loss.loss_weights.kl) == old_weight * 2
+    assert not learner._reset_value_flag
+    srv.stop()
+    learner._dataloader.close()
