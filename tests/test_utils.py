@@ -1,0 +1,145 @@
+"""Core-utility parity tests: metrics (reference `ctools/torch_utils/metric.py`),
+grad-clip family (`ctools/torch_utils/grad_clip.py`), checkpoint helper
+(`ctools/utils/checkpoint_helper.py`), log records (`ctools/utils/log_helper.py`)."""
+import torch
+
+from distar_amd.utils.checkpoint import CheckpointHelper, CountVar
+from distar_amd.utils.grad_clip import GradClip
+from distar_amd.utils.metric import (hamming_distance, l2_distance,
+                                     levenshtein_distance)
+
+
+# ------------------------------------------------------------------- metrics
+
+def test_levenshtein_basic():
+    a = torch.tensor([1, 2, 3, 4])
+    b = torch.tensor([1, 3, 4, 5])
+    # delete 2, insert 5 -> distance 2
+    assert float(levenshtein_distance(a, b)) == 2.0
+    assert float(levenshtein_distance(a, a)) == 0.0
+    assert float(levenshtein_distance(a[:0], b)) == 4.0
+
+
+def test_levenshtein_location_cost():
+    """Equal build-order actions still pay a distance-scaled location cost
+    (reference metric.py:14-60 extra_fn; used by the build-order pseudo-reward
+    agent.py update_fake_reward)."""
+    a = torch.tensor([7, 8])
+    b = torch.tensor([7, 8])
+    loc_a = torch.tensor([0, 0])                      # top-left corner
+    loc_b = torch.tensor([0, 159 + 160 * 151])        # opposite corner
+    d = levenshtein_distance(a, b, loc_a, loc_b,
+                             extra_fn=lambda x, y: l2_distance(x, y))
+    assert float(levenshtein_distance(a, b)) == 0.0
+    assert 0.0 < float(d) <= 2 * 0.8                  # capped at max_val each
+
+
+def test_hamming_and_l2():
+    p = torch.tensor([[1, 0, 1, 1], [0, 0, 0, 0]])
+    t = torch.tensor([[1, 1, 1, 0], [0, 0, 0, 0]])
+    assert hamming_distance(p, t).tolist() == [2.0, 0.0]
+    same = l2_distance(torch.tensor(5), torch.tensor(5))
+    far = l2_distance(torch.tensor(0), torch.tensor(159 + 160 * 151))
+    assert float(same) == 0.0 and abs(float(far) - 0.8) < 1e-6   # clamped
+
+
+# ----------------------------------------------------------------- grad clip
+
+def _grads(seed=0, scale=100.0):
+    torch.manual_seed(seed)
+    params = [torch.nn.Parameter(torch.randn(16)) for _ in range(3)]
+    for p in params:
+        p.grad = torch.randn_like(p) * scale
+    return params
+
+
+def test_grad_clip_family_finite():
+    for clip_type in ('none', 'pytorch_norm', 'clip_const', 'max_norm',
+                      'momentum_norm', 'clip_value'):
+        params = _grads()
+        gc = GradClip(clip_type, threshold=1.0, begin_step=0)
+        total = gc.apply(params)
+        assert total >= 0 and total == total, clip_type
+        assert all(torch.isfinite(p.grad).all() for p in params), clip_type
+
+
+def test_grad_clip_pytorch_norm_clips():
+    params = _grads(scale=100.0)
+    GradClip('pytorch_norm', threshold=1.0).apply(params)
+    total = torch.norm(torch.stack([p.grad.norm() for p in params]))
+    assert float(total) <= 1.0 + 1e-4
+
+
+def test_grad_clip_momentum_norm_scales_spike():
+    """A 100x grad spike after warm-up is pulled back to ~EMA*threshold."""
+    gc = GradClip('momentum_norm', threshold=1.0, begin_step=2)
+    params = _grads(scale=1.0)
+    base = [p.grad.norm().item() for p in params]
+    for _ in range(5):                    # build EMA state at scale 1
+        for i, p in enumerate(_grads(scale=1.0)):
+            params[i].grad = p.grad
+        gc.apply(params)
+    for p in params:
+        p.grad = p.grad * 100.0
+    gc.apply(params)
+    for p, b in zip(params, base):
+        assert p.grad.norm().item() < 10 * b
+
+
+def test_grad_clip_const_bounds_elements():
+    params = _grads(scale=100.0)
+    GradClip('clip_const', threshold=0.5).apply(params)
+    assert all(p.grad.abs().max() <= 0.5 for p in params)
+
+
+# ---------------------------------------------------------------- checkpoint
+
+def test_checkpoint_prefix_and_mask(tmp_path):
+    """prefix add/strip + state_dict_mask partial load (reference
+    checkpoint_helper.py:85-279)."""
+    m1 = torch.nn.Sequential(torch.nn.Linear(4, 4), torch.nn.Linear(4, 4))
+    m2 = torch.nn.Sequential(torch.nn.Linear(4, 4), torch.nn.Linear(4, 4))
+    helper = CheckpointHelper()
+    path = str(tmp_path / 'ck.pth.tar')
+    it = CountVar(7)
+    helper.save(path, m1, last_iter=it)
+    # strict full load restores everything
+    helper.load(path, m2, strict=True)
+    for a, b in zip(m1.parameters(), m2.parameters()):
+        torch.testing.assert_close(a, b)
+    # masked load touches only layer 0
+    with torch.no_grad():
+        for p in m1.parameters():
+            p.add_(1.0)
+    helper.save(path, m1, last_iter=it)
+    before_l0 = m2[0].weight.detach().clone()
+    # mask EXCLUDES matching keys (reference semantics): skip layer 0
+    helper.load(path, m2, strict=False, state_dict_mask=['0.'])
+    torch.testing.assert_close(m2[0].weight, before_l0)     # masked out
+    torch.testing.assert_close(m2[1].weight, m1[1].weight)  # loaded
+
+
+def test_countvar_roundtrip(tmp_path):
+    helper = CheckpointHelper()
+    m = torch.nn.Linear(2, 2)
+    it = CountVar(41)
+    path = str(tmp_path / 'c.pth.tar')
+    helper.save(path, m, last_iter=it)
+    it2 = CountVar(0)
+    helper.load(path, torch.nn.Linear(2, 2), strict=True, last_iter=it2)
+    assert it2.val == 41
+
+
+# ---------------------------------------------------------------------- logs
+
+def test_variable_record_tables():
+    from distar_amd.utils.log import VariableRecord
+    rec = VariableRecord(length=4)
+    rec.register_var('loss')
+    rec.register_var('reward')
+    for i in range(6):
+        rec.update_var({'loss': float(i), 'reward': 2.0})
+    text = rec.get_vars_text()
+    assert 'loss' in text and 'reward' in text
+    # windowed average over the last 4 updates: (2+3+4+5)/4
+    assert abs(rec.var_dict['loss'].avg - 3.5) < 1e-6
