@@ -188,3 +188,79 @@ def test_trueskill_updates_on_results(league):
     while league.elo.game_count < 5 and time.time() < deadline:
         time.sleep(0.05)
     assert league.trueskill.mu['MP0'] > league.trueskill.mu['EP0']
+
+
+def test_adaptive_evolutionary_reset_band(league):
+    """AE exploiter resets to the hardest historical snapshot in the 20-50%
+    win-rate band; teacher path when none qualify (reference player.py:640-760
+    TStarBot-X policy)."""
+    from distar_amd.league.player import AdaptiveEvolutionaryExploiterPlayer
+    league.add_active_player(
+        ckpt_path='ae.pth', pipeline='default', frac_id=1,
+        z_path='3map.json', z_prob=0., teacher_id='sl',
+        teacher_ckpt='sl_teacher.pth', one_phase_step=1000,
+        chosen_weight=1.0, player_id='AE0')
+    ae = league.active_players['AE0']
+    assert isinstance(ae, AdaptiveEvolutionaryExploiterPlayer)
+    assert ae.is_reset()            # always resets
+    # empty historical pool -> teacher checkpoint
+    assert ae.reset_checkpoint(league.active_players, {}, 'none') == 'sl_teacher.pth'
+    # snapshot the mains, then shape win rates: H1 easy (0.9), H2 in-band (0.3)
+    mp = league.active_players['MP0']
+    h1, h2 = mp.snapshot(), mp.snapshot()
+    hist = {h1.player_id: h1, h2.player_id: h2}
+    def stat(w):
+        return {'winrate': w, 'game_steps': 100, 'game_iters': 10,
+                'game_duration': 60}
+    ae.payoff._min_win_rate_games = 0          # skip the 0.5 warm-up gate
+    ae.payoff.update(h1.player_id, stat(1.0))  # easy snapshot
+    for w in [0, 0, 0, 0, 0, 0, 0, 1, 1, 1]:   # ~0.3 in-band
+        ae.payoff.update(h2.player_id, stat(float(w)))
+    wr = {pid: ae.payoff.pfsp_winrate_info_dict.get(pid, 0.5) for pid in hist}
+    target = ae.reset_checkpoint(league.active_players, hist, 'none')
+    in_band = [pid for pid, w in wr.items() if 0.2 <= w <= 0.5]
+    if in_band:
+        assert target in {hist[p].checkpoint_path for p in in_band}
+    else:
+        assert target == 'sl_teacher.pth'
+
+
+def test_expert_exploiter_z_style_rotation(league):
+    """EEP rotates among hand-picked style Zs on reset (reference
+    player.py:425-525)."""
+    from distar_amd.league.player import ExpertExploiterPlayer
+    league.add_active_player(
+        ckpt_path='eep.pth', pipeline='default', frac_id=1,
+        z_path=['mutalisk.json', 'worker_rush.json', '3map.json'],
+        z_prob=0., teacher_id='sl', teacher_ckpt='sl.pth',
+        one_phase_step=1000, chosen_weight=1.0, player_id='EEP0')
+    eep = league.active_players['EEP0']
+    assert isinstance(eep, ExpertExploiterPlayer)
+    assert eep.z_path in eep.z_paths
+    seen = set()
+    for _ in range(40):
+        assert eep.is_reset()       # resets after every snapshot, rotating Z
+        seen.add(eep.z_path)
+    assert len(seen) >= 2           # rotation actually samples styles
+    # snapshot names carry the style tag; reset target = newest MP snapshot
+    hp = eep.snapshot()
+    assert hp.player_id.startswith('EEP0H')
+    mp = league.active_players['MP0']
+    h1, h2 = mp.snapshot(), mp.snapshot()
+    hist = {h1.player_id: h1, h2.player_id: h2}
+    assert eep.reset_checkpoint(league.active_players, hist, 'none') == \
+        h2.checkpoint_path
+
+
+def test_phase_gate_snapshot_protocol(league):
+    """total_agent_step crossing one_phase_step makes is_trained_enough
+    true -> league snapshots into the historical pool (reference
+    league.py:259-297)."""
+    mp = league.active_players['MP0']
+    assert not mp.is_trained_enough(league.historical_players,
+                                    league.active_players)
+    mp.total_agent_step = mp.one_phase_step + 1
+    n_hist = len(league.historical_players)
+    league.deal_with_learner_send_train_info(
+        {'player_id': 'MP0', 'train_steps': 0, 'checkpoint_path': 'mp0.pth'})
+    assert len(league.historical_players) >= n_hist
