@@ -153,3 +153,41 @@ def test_rl_learner_debug_endpoints(tmp_path, monkeypatch):
     assert not learner._reset_value_flag
     srv.stop()
     learner._dataloader.close()
+
+
+@pytest.mark.timeout(300)
+def test_remote_sl_dataloader_via_adapter(tmp_path, monkeypatch):
+    """'remote' SL source: a replay-actor fleet pushes decoded step lists
+    through the Adapter (token 'replay'); dataloader workers pull and fill
+    the shared batch (reference sl_dataloader.py remote mode +
+    replay_actor.py push)."""
+    monkeypatch.chdir(tmp_path)
+    from distar_amd.data.adapter import Adapter
+    from distar_amd.data.coordinator import Coordinator
+    from distar_amd.data.sl_dataloader import SLDataloader
+    from distar_amd.lib.consts import fake_step_data
+    torch.manual_seed(0)
+    coord = Coordinator().run()
+    loader = None
+    try:
+        producer = Adapter(coordinator_port=coord.port)
+        for _ in range(3):
+            steps = [fake_step_data(train=True, entity_num=32, randomize=True)
+                     for _ in range(4)]
+            producer.push(steps, token='replay', fs_type='nppickle')
+        cfg = Config({
+            'learner': {'use_cuda': False, 'use_distributed': False,
+                        'data': {'source': 'remote', 'train_data_file': 'none',
+                                 'batch_size': 2, 'trajectory_length': 3,
+                                 'num_workers': 2, 'epochs': 1}},
+            'communication': {'coordinator_ip': '127.0.0.1',
+                              'coordinator_port': coord.port}})
+        loader = SLDataloader(cfg)
+        batch = next(loader)
+        assert batch['traj_lens'] == [3, 3]
+        assert all(batch['new_episodes'])
+        assert batch['entity_num'].shape[0] == 6
+    finally:
+        if loader is not None:
+            loader.close()
+        coord.close()
