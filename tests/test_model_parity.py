@@ -118,3 +118,27 @@ def test_selected_units_sampling_semantics(model):
         assert len(body) == len(set(body)), 'duplicate selection'
         if n < su.shape[1]:
             assert sel[-1] == int(en[b]) or len(sel) == MAX_SELECTED_UNITS_NUM
+
+
+def test_action_type_head_race_mask_play_mode():
+    """In play mode the 327-way logits are masked to the agent race's legal
+    actions (reference action_type_head.py:53-55 + stat.py ACTION_RACE_MASK)."""
+    from distar_amd.lib.stat import ACTION_RACE_MASK
+    from distar_amd.models.alphastar.heads import ActionTypeHead
+    from distar_amd.utils.config import Config
+    from distar_amd.models.alphastar.model import alphastar_model_default_config
+    torch.manual_seed(0)
+    cfg = Config({'common': {'type': 'play'}})
+    from distar_amd.utils.config import deep_merge_dicts
+    cfg = deep_merge_dicts(alphastar_model_default_config, cfg)
+    head = ActionTypeHead(cfg)
+    assert head.use_mask
+    for race in ('zerg', 'terran', 'protoss'):
+        head.race = race
+        logits, action, _ = head(torch.randn(4, 384), torch.randn(4, 448))
+        mask = ACTION_RACE_MASK[race]
+        assert (logits[:, ~mask] <= -1e8).all()
+        assert mask[action].all()          # sampled actions are race-legal
+    # train mode leaves logits unmasked
+    head2 = ActionTypeHead(alphastar_model_default_config)
+    assert not head2.use_mask
