@@ -191,6 +191,8 @@ class Actor:
             side = {'player_id': agent.player_id, 'opponent_id': opp,
                     'winloss': float(rewards.get(i, 0))}
             side.update(agent.get_stat_data())
+            if hasattr(agent, 'get_unit_num_info'):
+                side.update(agent.get_unit_num_info())
             result[str(i)] = side
         return result
 
@@ -201,12 +203,18 @@ class Actor:
         thread sharing the job's model weights — inference is no-grad and
         per-agent recurrent state lives on the Agent)."""
         env = self._make_env()
+        crashes = 0
         while not self._end and (episode_num < 0 or self.episodes_done < episode_num):
             try:
                 self._run_episode(env, agents)
                 self.episodes_done += 1
+                crashes = 0
             except Exception:  # noqa: BLE001 - actors are cattle
                 traceback.print_exc()
+                crashes += 1
+                if crashes >= 20:   # systematic failure, not a flaky episode
+                    self._logger.info('20 consecutive episode crashes, stopping worker')
+                    break
                 try:
                     env.close()
                 except Exception:  # noqa: BLE001
@@ -237,13 +245,19 @@ class Actor:
             (1 + 0.3 * (2 * random.random() - 1))
         if env_num <= 1:
             env = self._make_env()
+            crashes = 0
             while not self._end and \
                     (episode_num < 0 or self.episodes_done < episode_num):
                 try:
                     self._run_episode(env)
                     self.episodes_done += 1
+                    crashes = 0
                 except Exception:  # noqa: BLE001 - actors are cattle
                     traceback.print_exc()
+                    crashes += 1
+                    if crashes >= 20:
+                        self._logger.info('20 consecutive episode crashes, stopping')
+                        break
                     try:
                         env.close()
                     except Exception:  # noqa: BLE001

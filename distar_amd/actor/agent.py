@@ -486,7 +486,7 @@ class Agent:
 
     # ------------------------------------------------------------- telemetry
     def get_unit_num_info(self):
-        return {'unit_num': {}}
+        return {'unit_num': self._stat_api.unit_num}
 
     def get_stat_data(self):
         data = self._stat_api.get_stat_data()

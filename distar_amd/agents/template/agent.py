@@ -26,5 +26,8 @@ class Agent:
     def collect_data(self, next_obs, reward, done, idx):
         return None
 
+    def get_unit_num_info(self):
+        return {'unit_num': {}}
+
     def get_stat_data(self):
         return {'race_id': self.race, 'z_type': 0}

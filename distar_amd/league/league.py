@@ -408,6 +408,8 @@ class League:
                 with self._lock:
                     player.dist_stat.update(frac_id, side)
                     player.cum_stat.update(frac_id, side)
+                    player.unit_num_stat.update(frac_id, side.get('side_id', 0),
+                                                side)
                 if player.total_game_count % self.cfg.print_freq == 0:
                     for opp, info in player.payoff.stat_info_dict.items():
                         for k, v in info.items():

@@ -25,6 +25,16 @@ for _race, _mask in ACTION_RACE_MASK.items():
     assert _mask.shape[0] == NUM_ACTIONS
 
 
+def _unit_name_of(idx):
+    """'Train_Drone_quick' -> 'Drone', 'Build_SpawningPool_pt' ->
+    'SpawningPool', 'Morph_Lair_quick' -> 'Lair'."""
+    name = ACTIONS[idx]['name'] if 0 <= idx < NUM_ACTIONS else ''
+    parts = name.split('_')
+    if len(parts) >= 2 and parts[0] in ('Train', 'Build', 'Morph', 'TrainWarp'):
+        return parts[1]
+    return None
+
+
 class Stat:
     """Per-episode action/build statistics (reference `lib/stat.py:6-60`)."""
 
@@ -36,6 +46,8 @@ class Stat:
         self._action_success_count = defaultdict(int)
         self._action_count = defaultdict(int)
         self._build_count = defaultdict(int)
+        self._unit_num = defaultdict(int)
+        self._unit_num['max_unit_num'] = 0
 
     def update(self, last_action_type, action_result):
         if last_action_type is None:
@@ -48,9 +60,25 @@ class Stat:
             goal = ACTIONS[idx]['goal'] if 0 <= idx < NUM_ACTIONS else 'other'
             if goal in ('build', 'unit'):
                 self._build_count[idx] += 1
+                name = _unit_name_of(idx)
+                if name:
+                    self._unit_num[name] += 1
+                    self._unit_num['max_unit_num'] = max(
+                        self._unit_num[name], self._unit_num['max_unit_num'])
+
+    @property
+    def unit_num(self):
+        """Per-unit-name build counts + running max (reference
+        `lib/stat.py:8-12,47-52` builds this from a func_id->unit-name dict;
+        here the name is parsed from the action name)."""
+        return dict(self._unit_num)
 
     def get_stat_data(self):
         data = {}
+        mx = max(self._unit_num['max_unit_num'], 1)
+        for k, v in self._unit_num.items():
+            if k != 'max_unit_num':
+                data['units/' + k] = v / mx
         for idx, count in self._action_count.items():
             name = ACTIONS[idx]['name']
             data[f'action/{name}'] = count

@@ -136,3 +136,22 @@ def test_actor_multi_env_workers():
     assert len(results) >= 3
     for r in results:
         assert '0' in r and '1' in r
+
+
+def test_stat_unit_num_tracking():
+    """Build/train successes accumulate per-unit-name counts (reference
+    lib/stat.py:8-12,47-52); normalized units/* appear in stat data."""
+    from distar_amd.lib.actions import ACTIONS, NUM_ACTIONS
+    from distar_amd.lib.stat import Stat
+    stat = Stat('zerg')
+    train_idx = next(i for i in range(NUM_ACTIONS)
+                     if ACTIONS[i]['name'].startswith('Train_')
+                     and ACTIONS[i]['goal'] in ('build', 'unit'))
+    name = ACTIONS[train_idx]['name'].split('_')[1]
+    for _ in range(3):
+        stat.update(train_idx, 1)       # success
+    stat.update(train_idx, 2)           # failure: not counted
+    assert stat.unit_num[name] == 3
+    assert stat.unit_num['max_unit_num'] == 3
+    data = stat.get_stat_data()
+    assert data[f'units/{name}'] == 1.0
