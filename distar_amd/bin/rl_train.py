@@ -74,6 +74,10 @@ def learner_run(cfg, args):
         comm = LearnerComm(cfg)
         comm.register_learner(learner)
         learner._setup_comm_hooks(comm)
+    if getattr(learner, '_rank', 0) == 0:
+        # live-control endpoints (reference rl_train.py:19-51 serves a Flask
+        # debug app on a picked port); port logged to the learner log
+        learner.start_debug_server()
     learner.run(max_iterations=args.max_iterations)
 
 
