@@ -30,3 +30,43 @@ class Adam(torch.optim.Adam):
             params = [p for group in self.param_groups for p in group['params']]
             self._clip.apply(params)
         return super().step(closure)
+
+
+class GradualWarmupScheduler(torch.optim.lr_scheduler.LRScheduler):
+    """Linear warm-up to base_lr * multiplier over ``total_epoch`` steps, then
+    hand off to ``after_scheduler`` (functional parity with the reference's
+    `ctools/torch_utils/lr_scheduler_util.py`; the learners use torch's
+    SequentialLR(LinearLR, ...) composition by default, this class is the
+    standalone utility)."""
+
+    def __init__(self, optimizer, multiplier, total_epoch, after_scheduler=None):
+        if multiplier < 1.0:
+            raise ValueError('multiplier must be >= 1.0')
+        self.multiplier = multiplier
+        self.total_epoch = total_epoch
+        self.after_scheduler = after_scheduler
+        self.finished = False
+        super().__init__(optimizer)
+
+    def get_lr(self):
+        if self.last_epoch > self.total_epoch:
+            if self.after_scheduler is not None:
+                if not self.finished:
+                    self.after_scheduler.base_lrs = [
+                        b * self.multiplier for b in self.base_lrs]
+                    self.finished = True
+                return self.after_scheduler.get_last_lr()
+            return [b * self.multiplier for b in self.base_lrs]
+        if self.multiplier == 1.0:
+            return [b * float(self.last_epoch) / self.total_epoch
+                    for b in self.base_lrs]
+        return [b * ((self.multiplier - 1.) * self.last_epoch / self.total_epoch + 1.)
+                for b in self.base_lrs]
+
+    def step(self, epoch=None):
+        if self.finished and self.after_scheduler is not None:
+            self.after_scheduler.step(
+                None if epoch is None else epoch - self.total_epoch)
+            self._last_lr = self.after_scheduler.get_last_lr()
+        else:
+            super().step(epoch)

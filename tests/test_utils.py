@@ -143,3 +143,41 @@ def test_variable_record_tables():
     assert 'loss' in text and 'reward' in text
     # windowed average over the last 4 updates: (2+3+4+5)/4
     assert abs(rec.var_dict['loss'].avg - 3.5) < 1e-6
+
+
+# ----------------------------------------------------------- misc / schedule
+
+def test_misc_helpers():
+    from distar_amd.utils.misc import (default_get, dicts_to_lists,
+                                       error_wrapper, get_tensor_data,
+                                       list_split, lists_to_dicts, squeeze)
+    lod = [{'a': 1, 'b': 2}, {'a': 3, 'b': 4}]
+    dol = lists_to_dicts(lod)
+    assert dol == {'a': [1, 3], 'b': [2, 4]}
+    assert dicts_to_lists(dol) == lod
+    assert squeeze((5,)) == 5 and squeeze([1, 2]) == [1, 2]
+    assert default_get({}, 'x', default_value=9) == 9
+    assert default_get({'x': 1}, 'x') == 1
+    assert list_split(list(range(5)), 2) == [[0, 1], [2, 3], [4]]
+    assert error_wrapper(lambda: 1 / 0, -1, 'warn')() == -1
+    t = torch.randn(3, requires_grad=True)
+    out = get_tensor_data({'t': t, 'l': [t * 2]})
+    assert not out['t'].requires_grad and not out['l'][0].requires_grad
+
+
+def test_gradual_warmup_scheduler():
+    from distar_amd.utils.optimizer import GradualWarmupScheduler
+    from torch.optim.lr_scheduler import StepLR
+    p = torch.nn.Parameter(torch.zeros(1))
+    opt = torch.optim.SGD([p], lr=0.1)
+    after = StepLR(opt, step_size=5, gamma=0.1)
+    sched = GradualWarmupScheduler(opt, multiplier=1.0, total_epoch=4,
+                                   after_scheduler=after)
+    lrs = []
+    for _ in range(12):
+        opt.step()
+        sched.step()
+        lrs.append(opt.param_groups[0]['lr'])
+    assert lrs[0] < lrs[1] < lrs[2] < lrs[3]       # warming up
+    assert abs(lrs[3] - 0.1) < 1e-9                # reaches base lr
+    assert min(lrs[4:]) < 0.1                      # StepLR decays after
