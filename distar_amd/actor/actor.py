@@ -109,19 +109,24 @@ class Actor:
             return SC2Env(self._whole_cfg)
         return MockSC2Env(self._whole_cfg)
 
-    def _start_batch_inference(self):
+    def _start_batch_inference(self, agent_groups=None):
         """Shared-slab batched inference (reference actor.py:268-299): one
-        device model serves every env worker; agents switch to writing their
-        obs into the slab and polling the signal."""
+        device model serves every (env worker x player) slot; agents switch
+        to writing their obs into the slab and polling the signal."""
         import threading
         from .batch_inference import BatchInferenceServer
         device = 'cuda' if (self._cfg.use_cuda and torch.cuda.is_available()) \
             else 'cpu'
-        agent0 = self._agents[0]
+        agent_groups = agent_groups or [self._agents]
+        n_slots = sum(len(g) for g in agent_groups)
+        agent0 = agent_groups[0][0]
         server = BatchInferenceServer(agent0.model.to(device),
-                                      env_num=len(self._agents), device=device)
-        for env_id, agent in enumerate(self._agents):
-            agent.attach_batch_inference(server, env_id)
+                                      env_num=n_slots, device=device)
+        slot = 0
+        for group in agent_groups:
+            for agent in group:
+                agent.attach_batch_inference(server, slot)
+                slot += 1
         self._batch_server = server
         self._batch_thread = threading.Thread(target=server.run, daemon=True)
         self._batch_thread.start()
@@ -237,10 +242,12 @@ class Actor:
     def run(self):
         import threading
         self._setup_job()
-        if self._cfg.get('gpu_batch_inference', False):
-            self._start_batch_inference()
         env_num = self._cfg.env_num
         episode_num = self._cfg.episode_num
+        agent_groups = [self._agents] + \
+            [self._clone_agents(i) for i in range(1, env_num)]
+        if self._cfg.get('gpu_batch_inference', False):
+            self._start_batch_inference(agent_groups)
         job_deadline = time.time() + self._cfg.actor_ask_for_job_interval * \
             (1 + 0.3 * (2 * random.random() - 1))
         if env_num <= 1:
@@ -270,9 +277,9 @@ class Actor:
             return self.results
         workers = []
         for env_id in range(env_num):
-            agents = self._clone_agents(env_id) if env_id else self._agents
             t = threading.Thread(target=self._env_loop,
-                                 args=(agents, episode_num), daemon=True)
+                                 args=(agent_groups[env_id], episode_num),
+                                 daemon=True)
             t.start()
             workers.append(t)
         while any(t.is_alive() for t in workers):

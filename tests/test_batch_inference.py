@@ -57,3 +57,22 @@ def test_actor_batched_inference_mode():
     actor._batch_server.stop()
     assert len(results) == 1
     assert all(agent._batch_server is not None for agent in actor._agents)
+
+
+@pytest.mark.timeout(900)
+def test_actor_batch_inference_multi_env():
+    """env_num x players share one slab server: every (worker, player) slot
+    gets its own signal lane (reference actor.py:268-299 with env_num>1)."""
+    torch.manual_seed(0)
+    from distar_amd.actor.actor import Actor
+    cfg = Config({'actor': {'episode_num': 2, 'env_num': 2, 'env_type': 'mock',
+                            'traj_len': 4, 'gpu_batch_inference': True,
+                            'use_cuda': False},
+                  'env': {'player_num': 2, 'max_episode_steps': 4},
+                  'common': {'experiment_name': 'test_actor_slab_multi',
+                             'type': 'train'}})
+    actor = Actor(cfg)
+    results = actor.run()
+    assert len(results) >= 2
+    assert actor._batch_server.env_num == 4      # 2 envs x 2 players
+    actor._batch_server.stop()
