@@ -272,6 +272,11 @@ class Actor:
                     env = self._make_env()
                 if time.time() > job_deadline:
                     self._setup_job()
+                    if self._cfg.get('gpu_batch_inference', False):
+                        # new job -> new agents/models: restart the slab server
+                        self._batch_server.stop()
+                        self._batch_thread.join(timeout=5)
+                        self._start_batch_inference()
                     job_deadline = time.time() + self._cfg.actor_ask_for_job_interval
             env.close()
             return self.results
