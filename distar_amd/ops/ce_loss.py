@@ -1,0 +1,47 @@
+"""Fused masked cross-entropy (K13, experimental).
+
+Autograd wrapper over `ops/hip/ce_loss.hip` for the SL location-head loss
+(reference `sl_training/sl_loss.py` uses `F.cross_entropy` on an
+(N, 24320) fp32 tensor).  Forward returns per-row losses like
+``F.cross_entropy(..., reduction='none') * mask``; callers reduce.
+
+EXPERIMENTAL: GPU numerics validation is scheduled for round 2 — the HIP
+path is only taken with ``DISTAR_AMD_FUSED_CE=1`` (and on CUDA); otherwise
+this is the eager composition.
+"""
+import os
+
+import torch
+
+
+class _FusedMaskedCE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, labels, mask):
+        from .hip_ext import ops
+        loss, lse = ops.masked_ce_fwd(logits, labels, mask)
+        ctx.save_for_backward(logits, labels, lse,
+                              mask if mask is not None else torch.empty(0))
+        return loss
+
+    @staticmethod
+    def backward(ctx, gout):
+        from .hip_ext import ops
+        logits, labels, lse, mask = ctx.saved_tensors
+        mask = mask if mask.numel() else None
+        dlogits = ops.masked_ce_bwd(logits, labels, mask, lse,
+                                    gout.contiguous())
+        return dlogits, None, None
+
+
+def masked_cross_entropy(logits, labels, mask=None):
+    """Per-row CE ``(logsumexp(logits_i) - logits_i[label_i]) * mask_i``.
+
+    logits: (N, C) float; labels: (N,) long; mask: (N,) float or None.
+    """
+    use_hip = (logits.is_cuda and logits.dtype == torch.float32
+               and os.environ.get('DISTAR_AMD_FUSED_CE') == '1')
+    if use_hip:
+        return _FusedMaskedCE.apply(logits.contiguous(), labels.contiguous(),
+                                    None if mask is None else mask.contiguous())
+    loss = torch.nn.functional.cross_entropy(logits, labels, reduction='none')
+    return loss if mask is None else loss * mask

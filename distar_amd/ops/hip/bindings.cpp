@@ -196,6 +196,12 @@ std::vector<torch::Tensor> lnlstm_backward(
   return {digates, dhgates_raw, dh0, dc0, dlnh_w, dlnh_b, dlnc_w, dlnc_b};
 }
 
+extern "C" __global__ void masked_ce_fwd_kernel(
+    const float*, const long*, const float*, float*, float*, int, int);
+extern "C" __global__ void masked_ce_bwd_kernel(
+    const float*, const long*, const float*, const float*, const float*,
+    float*, int, int);
+
 extern "C" __global__ void su_sample_kernel(
     const float*, const __hip_bfloat16*, const unsigned char*,
     const unsigned char*, const int*, const float*,
@@ -279,7 +285,49 @@ torch::Tensor entity_embed(torch::Tensor int_fields, torch::Tensor float_fields,
 
 }  // namespace
 
+std::vector<torch::Tensor> masked_ce_fwd(torch::Tensor logits,
+                                         torch::Tensor labels,
+                                         c10::optional<torch::Tensor> mask) {
+  check_2d(logits, "logits");
+  TORCH_CHECK(labels.scalar_type() == torch::kLong, "labels must be int64");
+  int64_t N = logits.size(0), C = logits.size(1);
+  auto loss = torch::empty({N}, logits.options());
+  auto lse = torch::empty({N}, logits.options());
+  const float* mptr = nullptr;
+  if (mask.has_value()) {
+    TORCH_CHECK(mask->scalar_type() == torch::kFloat32, "mask must be fp32");
+    mptr = mask->data_ptr<float>();
+  }
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(masked_ce_fwd_kernel, dim3((unsigned)N), dim3(256), 0,
+                     stream.stream(),
+                     logits.data_ptr<float>(), labels.data_ptr<long>(), mptr,
+                     loss.data_ptr<float>(), lse.data_ptr<float>(),
+                     (int)N, (int)C);
+  return {loss, lse};
+}
+
+torch::Tensor masked_ce_bwd(torch::Tensor logits, torch::Tensor labels,
+                            c10::optional<torch::Tensor> mask,
+                            torch::Tensor lse, torch::Tensor gout) {
+  check_2d(logits, "logits");
+  int64_t N = logits.size(0), C = logits.size(1);
+  auto dlogits = torch::empty_like(logits);
+  const float* mptr = nullptr;
+  if (mask.has_value()) mptr = mask->data_ptr<float>();
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(masked_ce_bwd_kernel, dim3((unsigned)N), dim3(256), 0,
+                     stream.stream(),
+                     logits.data_ptr<float>(), labels.data_ptr<long>(), mptr,
+                     lse.data_ptr<float>(), gout.data_ptr<float>(),
+                     dlogits.data_ptr<float>(), (int)N, (int)C);
+  return dlogits;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("masked_ce_fwd", &masked_ce_fwd,
+        "fused masked CE forward (per-row loss + lse)");
+  m.def("masked_ce_bwd", &masked_ce_bwd, "fused masked CE backward");
   m.def("su_sample", &su_sample, "selected-units sampling loop (one kernel)");
   m.def("entity_embed", &entity_embed,
         "fused 36-field entity embedding -> (R, 997) bf16");
