@@ -177,3 +177,31 @@ def test_transform_obs_upgrades_one_hot():
     up = out['scalar_info']['upgrades']
     assert int(up.sum()) == 2
     assert up[1] == 1 and up[3] == 1
+
+
+def test_filter_actions_dedups_spam():
+    """Spammed identical train/morph/research commands inside the window
+    collapse to the LAST occurrence; distinct tags/abilities and spaced
+    repeats survive (reference replay_decoder.py:70-213)."""
+    from distar_amd.data.replay_decoder import FilterActions
+    from distar_amd.lib.actions import ACTIONS
+    f = FilterActions()
+    train_gab = next(a['general_ability_id'] for a in ACTIONS
+                     if a['goal'] == 'unit' and a['general_ability_id'])
+    other_gab = next(a['general_ability_id'] for a in ACTIONS
+                     if a['goal'] == 'build' and a['general_ability_id']
+                     and 'Morph' not in a['name'])
+    tags = (10, 11)
+    seq = [
+        (0, train_gab, tags, 'a0'),
+        (1, train_gab, tags, 'a1'),      # spam: within window, same tags
+        (2, train_gab, tags, 'a2'),      # spam again -> keep only a2
+        (3, train_gab, (99,), 'b0'),     # different tags: kept
+        (4, other_gab, tags, 'c0'),      # non-filtered goal: kept
+        (20, train_gab, tags, 'd0'),     # far outside window: kept
+    ]
+    out = f.run(seq)
+    kept = [x[3] for x in out]
+    assert kept == ['a2', 'b0', 'c0', 'd0']
+    # the surviving spam entry carries the LAST loop (freshest delay timing)
+    assert out[0][0] == 2
