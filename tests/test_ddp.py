@@ -76,3 +76,85 @@ def test_distmodule_grads_are_averaged():
             expected[n] = expected.get(n, 0) + p.grad / 2
     for n, g in expected.items():
         torch.testing.assert_close(results[0][n], g)
+
+
+def _cross_rank_worker(rank, world, init_file, q):
+    dist.init_process_group('gloo', init_method=f'file://{init_file}',
+                            rank=rank, world_size=world)
+    from distar_amd.lib.fake_data import fake_sl_batch_fast
+    from distar_amd.losses.sl_loss import SupervisedLoss
+    from distar_amd.models.alphastar.model import Model
+    from distar_amd.utils.config import Config
+    torch.manual_seed(5)                  # same weights on both ranks
+    model = Model(Config({}), temperature=1.0)
+    torch.manual_seed(50 + rank)          # different data per rank
+    data = fake_sl_batch_fast(batch_size=2, traj_len=2, seed=50 + rank)
+    hidden = [(torch.zeros(2, 384), torch.zeros(2, 384)) for _ in range(3)]
+    logits, infer, _ = model.sl_train(**data, hidden_state=hidden)
+    loss = SupervisedLoss(Config({'learner': {'cross_rank_loss': True}}))
+    ld = loss.compute_loss(logits, data['action_info'], data['action_mask'],
+                           data['selected_units_num'], data['entity_num'],
+                           infer)
+    q.put((rank, float(loss.total_batch_size), float(ld['total_loss'])))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_sl_cross_rank_loss_renormalizes():
+    """cross_rank_loss allreduces the global batch size and renormalizes
+    each rank's loss by its share (reference sl_loss.py:100-104,127-135)."""
+    with tempfile.TemporaryDirectory() as d:
+        init_file = os.path.join(d, 'init')
+        ctx = mp.get_context('spawn')
+        q = ctx.Queue()
+        procs = [ctx.Process(target=_cross_rank_worker,
+                             args=(r, 2, init_file, q)) for r in range(2)]
+        for p in procs:
+            p.start()
+        out = {}
+        for _ in range(2):
+            rank, total_bs, loss = q.get(timeout=240)
+            out[rank] = (total_bs, loss)
+        for p in procs:
+            p.join(timeout=60)
+    # (B*T)=4 rows per rank -> global 8 on both ranks
+    assert out[0][0] == 8.0 and out[1][0] == 8.0
+    assert out[0][1] > 0 and out[1][1] > 0
+
+
+def _log_reduce_worker(rank, world, init_file, q):
+    dist.init_process_group('gloo', init_method=f'file://{init_file}',
+                            rank=rank, world_size=world)
+    from distar_amd.learner.hooks import LogReduceHook
+    from types import SimpleNamespace
+    engine = SimpleNamespace(log_buffer={'loss': float(rank), 'acc': 1.0 + rank,
+                                         'name': 'text-untouched'})
+    LogReduceHook(position='after_iter')(engine)
+    q.put((rank, dict(engine.log_buffer)))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_log_reduce_hook_averages_buffer():
+    """LogReduceHook allreduce-averages every scalar in the log buffer as one
+    flat tensor (reference learner_hook.py:271-325)."""
+    with tempfile.TemporaryDirectory() as d:
+        init_file = os.path.join(d, 'init')
+        ctx = mp.get_context('spawn')
+        q = ctx.Queue()
+        procs = [ctx.Process(target=_log_reduce_worker,
+                             args=(r, 2, init_file, q)) for r in range(2)]
+        for p in procs:
+            p.start()
+        out = {}
+        for _ in range(2):
+            rank, buf = q.get(timeout=240)
+            out[rank] = buf
+        for p in procs:
+            p.join(timeout=60)
+    for rank in (0, 1):
+        assert out[rank]['loss'] == 0.5          # mean(0, 1)
+        assert out[rank]['acc'] == 1.5           # mean(1, 2)
+        assert out[rank]['name'] == 'text-untouched'
