@@ -145,8 +145,23 @@ class SelectedUnitsHead(nn.Module):
         stdv = 1. / math.sqrt(self.end_embedding.size(1))
         self.end_embedding.data.uniform_(-stdv, stdv)
         self.extra_units = self.whole_cfg.get('agent', {}).get('extra_units', False)
-        assert self.whole_cfg.model.entity_reduce_type == 'selected_units_num', \
-            'this build implements the default reduce type (reference default)'
+        # 'entity_num'/'constant' are dead paths upstream (the reference's
+        # _get_key_mask calls a self.embed_fc that is never constructed,
+        # action_arg_head.py:131-136 vs :98-99); the working set is:
+        self.reduce_type = self.whole_cfg.model.entity_reduce_type
+        assert self.reduce_type in ('selected_units_num', 'attention_pool',
+                                    'attention_pool_add_num'), self.reduce_type
+        if self.reduce_type == 'attention_pool':
+            from ..nn.transformer import AttentionPool
+            self.attention_pool = AttentionPool(
+                key_dim=self.cfg.key_dim, head_num=2,
+                output_dim=self.cfg.input_dim)
+        elif self.reduce_type == 'attention_pool_add_num':
+            from ..nn.transformer import AttentionPool
+            self.attention_pool = AttentionPool(
+                key_dim=self.cfg.key_dim, head_num=2,
+                output_dim=self.cfg.input_dim,
+                max_num=MAX_SELECTED_UNITS_NUM + 1)
 
     def _get_key_mask(self, entity_embedding, entity_num):
         """Keys (with the learned end-embedding spliced in at index
@@ -157,7 +172,7 @@ class SelectedUnitsHead(nn.Module):
         flag = torch.ones(bs, n + 1, 1, dtype=torch.bool, device=key.device)
         flag[torch.arange(bs, device=key.device), entity_num] = 0
         key = key * flag + self.end_embedding.squeeze(0) * (~flag)
-        key_embeddings = key     # reduce_type == 'selected_units_num'
+        key_embeddings = key     # same for selected_units_num + attention_*
         new_entity_num = entity_num + 1      # end slot is a valid position
         mask = sequence_mask(new_entity_num, max_len=n + 1)
         return key, mask, key_embeddings
@@ -191,22 +206,47 @@ class SelectedUnitsHead(nn.Module):
         # ae at step s uses labels < s: shift right by one
         sum_s = torch.cat([cum_sum.new_zeros(bs, 1, self.key_dim), cum_sum[:, :-1]], dim=1)
         cnt_s = torch.cat([cum_cnt.new_zeros(bs, 1), cum_cnt[:, :-1]], dim=1)
-        mean_s = sum_s.clone()
-        # reference: divide only rows with selected_units_num != 0; guard
-        # count==0 (keeps the raw sum, avoiding the reference's 0/0 edge)
-        div_rows = (selected_units_num != 0).unsqueeze(1) & (cnt_s > 0)
-        mean_s = torch.where(div_rows.unsqueeze(-1), sum_s / cnt_s.clamp(min=1).unsqueeze(-1), sum_s)
-        ae_delta = self.embed_fc2(self.embed_fc1(mean_s))          # B, S, input_dim
-        ae_delta[:, 0] = 0.                                        # step 0: raw base ae
+        if self.reduce_type == 'selected_units_num':
+            mean_s = sum_s.clone()
+            # reference: divide only rows with selected_units_num != 0; guard
+            # count==0 (keeps the raw sum, avoiding the reference's 0/0 edge)
+            div_rows = (selected_units_num != 0).unsqueeze(1) & (cnt_s > 0)
+            mean_s = torch.where(div_rows.unsqueeze(-1), sum_s / cnt_s.clamp(min=1).unsqueeze(-1), sum_s)
+            ae_delta = self.embed_fc2(self.embed_fc1(mean_s))      # B, S, input_dim
+            ae_delta[:, 0] = 0.                                    # step 0: raw base ae
+        else:
+            # attention-pool variants (reference action_arg_head.py:201-208):
+            # ae_s = base + pool(keys, mask=one-hot of labels < s).  The pool
+            # is nonlinear in the mask, so it runs per step over cumulative
+            # one-hots (S small); LSTM and logits stay batched.
+            gated = onehot.float() * include.float().unsqueeze(-1)
+            cum_oh = gated.cumsum(dim=1).clamp(max=1.0)            # B, S, N+1
+            deltas = [key_embeddings.new_zeros(bs, self.cfg.input_dim)]
+            for st in range(1, seq_len):
+                m = cum_oh[:, st - 1].unsqueeze(-1)
+                if self.reduce_type == 'attention_pool':
+                    deltas.append(self.attention_pool(key_embeddings, mask=m))
+                else:
+                    deltas.append(self.attention_pool(
+                        key_embeddings, num=m.sum(dim=(1, 2)), mask=m))
+            ae_delta = torch.stack(deltas, dim=1)                  # B, S, input_dim
         # prev-selected mask for the logits (any label j < s disables index L[b,j])
         prev_cnt = torch.cat([cum.new_zeros(bs, 1, cum.shape[2]), cum[:, :-1]], dim=1)
         prev_selected = prev_cnt > 0                               # B, S, N+1
         # final ae after the full unroll (consumed by the target-unit head)
-        fin_div = (selected_units_num != 0) & (cum_cnt[:, -1] > 0)
-        fin_mean = torch.where(fin_div.unsqueeze(-1),
-                               cum_sum[:, -1] / cum_cnt[:, -1].clamp(min=1).unsqueeze(-1),
-                               cum_sum[:, -1])
-        final_delta = self.embed_fc2(self.embed_fc1(fin_mean))
+        if self.reduce_type == 'selected_units_num':
+            fin_div = (selected_units_num != 0) & (cum_cnt[:, -1] > 0)
+            fin_mean = torch.where(fin_div.unsqueeze(-1),
+                                   cum_sum[:, -1] / cum_cnt[:, -1].clamp(min=1).unsqueeze(-1),
+                                   cum_sum[:, -1])
+            final_delta = self.embed_fc2(self.embed_fc1(fin_mean))
+        else:
+            mfin = cum_oh[:, -1].unsqueeze(-1)
+            if self.reduce_type == 'attention_pool':
+                final_delta = self.attention_pool(key_embeddings, mask=mfin)
+            else:
+                final_delta = self.attention_pool(
+                    key_embeddings, num=mfin.sum(dim=(1, 2)), mask=mfin)
         return ae_delta, prev_selected, final_delta
 
     def _query_train(self, key, entity_num, autoregressive_embedding, logits_mask,
@@ -270,6 +310,7 @@ class SelectedUnitsHead(nn.Module):
         """Data-dependent sampling loop (reference `action_arg_head.py:262-313`)."""
         import os
         if autoregressive_embedding.is_cuda and \
+                self.reduce_type == 'selected_units_num' and \
                 os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1':
             return self._query_sample_hip(key, entity_num, autoregressive_embedding,
                                           logits_mask, key_embeddings, su_mask,
@@ -290,6 +331,7 @@ class SelectedUnitsHead(nn.Module):
         result: Optional[Tensor] = None
         sel_sum = ae.new_zeros(bs, self.key_dim)
         sel_cnt = ae.new_zeros(bs)
+        sel_oh = ae.new_zeros(bs, key.shape[1])
         step_logits = None
         for i in range(self.max_select_num):
             if i == 1:
@@ -316,11 +358,23 @@ class SelectedUnitsHead(nn.Module):
             results_list.append(result)
             logits_list.append(step_logits)
             picked = (~end_flag).float()
-            sel_sum = sel_sum + key_embeddings[arange, result] * picked.unsqueeze(1)
-            sel_cnt = sel_cnt + picked
-            mean = torch.where((sel_cnt > 0).unsqueeze(1), sel_sum / sel_cnt.clamp(min=1).unsqueeze(1),
-                               sel_sum)
-            ae = autoregressive_embedding + self.embed_fc2(self.embed_fc1(mean))
+            if self.reduce_type == 'selected_units_num':
+                sel_sum = sel_sum + key_embeddings[arange, result] * picked.unsqueeze(1)
+                sel_cnt = sel_cnt + picked
+                mean = torch.where((sel_cnt > 0).unsqueeze(1),
+                                   sel_sum / sel_cnt.clamp(min=1).unsqueeze(1),
+                                   sel_sum)
+                ae = autoregressive_embedding + self.embed_fc2(self.embed_fc1(mean))
+            else:
+                sel_oh = sel_oh.clone()
+                sel_oh[arange[~end_flag], result[~end_flag]] = 1.
+                m = sel_oh.unsqueeze(-1)
+                if self.reduce_type == 'attention_pool':
+                    ae = autoregressive_embedding + \
+                        self.attention_pool(key_embeddings, mask=m)
+                else:
+                    ae = autoregressive_embedding + self.attention_pool(
+                        key_embeddings, num=sel_oh.sum(dim=1), mask=m)
             if bool(end_flag.all()):
                 break
         extra_units = torch.zeros(bs, MAX_ENTITY_NUM + 1, device=device)

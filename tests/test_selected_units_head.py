@@ -9,7 +9,7 @@ from distar_amd.utils.config import Config
 
 
 def reference_loop(head, key, entity_num, ae_base, logits_mask, key_embeddings,
-                   selected_units_num, selected_units):
+                   selected_units_num, selected_units, reduce_type='selected_units_num'):
     """Literal transcription of the reference's train branch semantics."""
     bs = ae_base.shape[0]
     device = ae_base.device
@@ -35,11 +35,18 @@ def reference_loop(head, key, entity_num, ae_base, logits_mask, key_embeddings,
         new_one_hot = selected_units_one_hot.clone()
         end_flag[selected_units[:, i] == entity_num] = True
         new_one_hot[torch.arange(bs)[~end_flag], selected_units[:, i][~end_flag], :] = 1
-        emb = (key_embeddings * new_one_hot).sum(dim=1)
-        rows = selected_units_num != 0
-        emb[rows] = emb[rows] / new_one_hot.sum(dim=1)[rows]
-        emb = head.embed_fc2(head.embed_fc1(emb))
-        ae = ae_base + emb
+        if reduce_type == 'selected_units_num':
+            emb = (key_embeddings * new_one_hot).sum(dim=1)
+            rows = selected_units_num != 0
+            emb[rows] = emb[rows] / new_one_hot.sum(dim=1)[rows]
+            emb = head.embed_fc2(head.embed_fc1(emb))
+            ae = ae_base + emb
+        elif reduce_type == 'attention_pool':
+            ae = ae_base + head.attention_pool(key_embeddings, mask=new_one_hot)
+        else:   # attention_pool_add_num
+            ae = ae_base + head.attention_pool(
+                key_embeddings, num=new_one_hot.sum(dim=1).squeeze(dim=1),
+                mask=new_one_hot)
         selected_units_one_hot = new_one_hot.clone()
     queries = torch.cat(queries, dim=0).unsqueeze(dim=2)
     logits = (queries * key.unsqueeze(0)).sum(dim=3)
@@ -96,3 +103,57 @@ def test_train_path_gradients_flow():
     loss.backward()
     assert ae_base.grad is not None and torch.isfinite(ae_base.grad).all()
     assert entity_embedding.grad is not None and torch.isfinite(entity_embedding.grad).all()
+
+
+def test_train_path_matches_reference_loop_attention_variants():
+    """Closed-form train path under the attention-pool reduce variants
+    (reference action_arg_head.py:112-116,201-208)."""
+    for reduce_type in ('attention_pool', 'attention_pool_add_num'):
+        torch.manual_seed(0)
+        model = Model(Config({'common': {'type': 'train'},
+                              'model': {'entity_reduce_type': reduce_type}}))
+        head = model.policy.selected_units_head
+        B, N = 3, 24
+        entity_embedding = torch.randn(B, N, 256)
+        entity_num = torch.tensor([24, 10, 17])
+        ae_base = torch.randn(B, 1024)
+        su_num = torch.tensor([5, 1, 3])
+        su = torch.zeros(B, 5, dtype=torch.long)
+        su[0] = torch.tensor([3, 9, 0, 11, 24])
+        su[1, 0] = 10
+        su[2, :3] = torch.tensor([1, 16, 17])
+        with torch.no_grad():
+            key, mask, key_embeddings = head._get_key_mask(entity_embedding, entity_num)
+            ref_logits, ref_ae = reference_loop(
+                head, key, entity_num, ae_base, mask, key_embeddings, su_num, su,
+                reduce_type=reduce_type)
+            logits, _, final_ae, _, _ = head._query_train(
+                key, entity_num, ae_base, mask, key_embeddings, su_num, su)
+        assert torch.isfinite(logits).all(), reduce_type
+        finite = torch.isfinite(ref_logits)
+        torch.testing.assert_close(logits[finite], ref_logits[finite],
+                                   rtol=1e-4, atol=1e-4)
+        finite_ae = torch.isfinite(ref_ae)
+        torch.testing.assert_close(final_ae[finite_ae], ref_ae[finite_ae],
+                                   rtol=1e-4, atol=1e-4)
+
+
+def test_sample_path_attention_variants_run():
+    """Eager sampling loop under the attention variants produces valid
+    shapes and respects the end-token contract."""
+    for reduce_type in ('attention_pool', 'attention_pool_add_num'):
+        torch.manual_seed(1)
+        model = Model(Config({'common': {'type': 'train'},
+                              'model': {'entity_reduce_type': reduce_type}}))
+        head = model.policy.selected_units_head
+        B, N = 2, 16
+        entity_embedding = torch.randn(B, N, 256)
+        entity_num = torch.tensor([16, 9])
+        ae = torch.randn(B, 1024)
+        su_mask = torch.ones(B, dtype=torch.bool)
+        with torch.no_grad():
+            key, mask, key_embeddings = head._get_key_mask(entity_embedding, entity_num)
+            logits, results, out_ae, num, extra = head._query_sample(
+                key, entity_num, ae, mask, key_embeddings, su_mask)
+        assert results.shape[0] == B and out_ae.shape == (B, 1024)
+        assert (num >= 1).all() and (num <= 64).all()
