@@ -298,11 +298,17 @@ class SelectedUnitsHead(nn.Module):
         ae = autoregressive_embedding + self.embed_fc2(self.embed_fc1(mean))
         extra_units = torch.zeros(bs, MAX_ENTITY_NUM + 1, device=device)
         if self.extra_units:
+            # reference action_arg_head.py:307-309: extras come from the LAST
+            # executed step's logits, and only for rows whose selection was
+            # truncated at the cap (rows that picked the end token get none)
             last = (num - 1).clamp(min=0)
             last_logits = logits[arange, last]
             end_logit = last_logits[arange, entity_num]
+            ended = (results.gather(1, last.unsqueeze(1)).squeeze(1) ==
+                     entity_num) | (num == 0)
             extra_units[:, :last_logits.shape[1]] = \
-                (last_logits > end_logit.unsqueeze(1)).float()
+                ((last_logits > end_logit.unsqueeze(1)) &
+                 ~ended.unsqueeze(1)).float()
         return logits, results, ae, num, extra_units
 
     def _query_sample(self, key, entity_num, autoregressive_embedding, logits_mask,

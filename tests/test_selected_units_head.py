@@ -157,3 +157,28 @@ def test_sample_path_attention_variants_run():
                 key, entity_num, ae, mask, key_embeddings, su_mask)
         assert results.shape[0] == B and out_ae.shape == (B, 1024)
         assert (num >= 1).all() and (num <= 64).all()
+
+
+def test_extra_units_last_step_semantics():
+    """extra_units: last-step logits beating the end logit, only for rows
+    truncated at the 64-cap (reference action_arg_head.py:307-309)."""
+    torch.manual_seed(3)
+    model = Model(Config({'common': {'type': 'train'},
+                          'agent': {'extra_units': True}}))
+    head = model.policy.selected_units_head
+    assert head.extra_units
+    B, N = 2, 12
+    entity_embedding = torch.randn(B, N, 256)
+    entity_num = torch.tensor([12, 12])
+    ae = torch.randn(B, 1024)
+    su_mask = torch.ones(B, dtype=torch.bool)
+    # uniforms driven to 0 -> always pick argmax-of-cdf index 0.. just run it
+    with torch.no_grad():
+        key, mask, key_embeddings = head._get_key_mask(entity_embedding, entity_num)
+        logits, results, out_ae, num, extra = head._query_sample(
+            key, entity_num, ae, mask, key_embeddings, su_mask)
+    arange = torch.arange(B)
+    last = (num - 1).clamp(min=0)
+    ended = results.gather(1, last.unsqueeze(1)).squeeze(1) == entity_num
+    # rows that ended with the end token must have zero extras
+    assert (extra[ended] == 0).all()
