@@ -100,6 +100,20 @@ class Actor:
                 agent.teacher_model.eval()
                 self._ckpt_helper.load(tpath, agent.teacher_model, strict=False,
                                        logger_prints=self._logger.info)
+            # DAPO: successive (lagged-self) model per main player (reference
+            # actor_comm.py:127-147 reconstructs the league's saved path)
+            spaths = job.get('successive_model_paths', [])
+            spath = spaths[i] if i < len(spaths) else 'none'
+            if self._whole_cfg.get('learner', {}).get('use_dapo', False) and \
+                    spath not in ('none', None) and \
+                    getattr(agent, 'HAS_MODEL', False) and \
+                    not self._cfg.fake_model and os.path.isfile(str(spath)):
+                from ..models.alphastar.model import Model as _Model
+                agent.successive_model = _Model(self._whole_cfg)
+                agent.successive_model.eval()
+                self._ckpt_helper.load(spath, agent.successive_model,
+                                       strict=False,
+                                       logger_prints=self._logger.info)
             self._agents.append(agent)
         return job
 
@@ -236,6 +250,7 @@ class Actor:
             if getattr(agent, 'HAS_MODEL', False):
                 clone.model = agent.model
                 clone.teacher_model = agent.teacher_model
+                clone.successive_model = agent.successive_model
             clones.append(clone)
         return clones
 

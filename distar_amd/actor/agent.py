@@ -60,6 +60,7 @@ class Agent:
         self.model = Model(self._whole_cfg or {'common': {'type': 'train'}})
         self.model.eval()
         self.teacher_model = None
+        self.successive_model = None    # DAPO (reference agent.py:506-515)
         self.player_id = 'MP0'
         self.race = 'zerg'
         self._num_layers = self.model.cfg.encoder.core_lstm.num_layers
@@ -84,6 +85,8 @@ class Agent:
                               for _ in range(self._num_layers)]
         self._teacher_hidden_state = [(z(self._hidden_size), z(self._hidden_size))
                                       for _ in range(self._num_layers)]
+        self._successive_hidden_state = [(z(self._hidden_size), z(self._hidden_size))
+                                         for _ in range(self._num_layers)]
         self._hidden_state_backup = self._hidden_state
         self._last_action_type = torch.tensor(0, dtype=torch.long)
         self._last_delay = torch.tensor(0, dtype=torch.long)
@@ -352,6 +355,17 @@ class Agent:
         teacher_output = self.decollate_output(teacher_output)
         self._teacher_hidden_state = teacher_output['hidden_state']
 
+        successive_output = None
+        if self.successive_model is not None:
+            succ_input = default_collate_with_dim([teacher_obs])
+            succ_input['hidden_state'] = [(h.unsqueeze(0), c.unsqueeze(0))
+                                          for h, c in self._successive_hidden_state]
+            with torch.no_grad():
+                successive_output = self.successive_model.compute_teacher_logit(
+                    **succ_input)
+            successive_output = self.decollate_output(successive_output)
+            self._successive_hidden_state = successive_output['hidden_state']
+
         action_info = copy.deepcopy(self._output['action_info'])
         at = int(action_info['action_type'])
         mask = {
@@ -383,6 +397,8 @@ class Agent:
             'step': torch.tensor(float(self._game_step)),
             'mask': mask,
         }
+        if successive_output is not None:
+            step_data['successive_logit'] = successive_output['logit']
         if self._use_value_feature and 'value_feature' in agent_obs:
             step_data['value_feature'] = dict(agent_obs['value_feature'])
             step_data['value_feature'].update(self.get_behavior_z())

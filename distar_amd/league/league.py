@@ -257,12 +257,15 @@ class League:
     def _job_from_players(self, players, branch):
         successive = [p.player_id if isinstance(p, MainPlayer) else 'none'
                       for p in players]
+        successive_paths = [p.successive_model_path if isinstance(p, MainPlayer)
+                            else 'none' for p in players]
         job_info = {
             'player_ids': [p.player_id for p in players],
             'side_ids': list(range(len(players))),
             'pipelines': [p.pipeline for p in players],
             'checkpoint_paths': [p.checkpoint_path for p in players],
             'successive_ids': successive,
+            'successive_model_paths': successive_paths,
             'z_path': [p.z_path for p in players],
             'z_prob': [p.z_prob for p in players],
             'teacher_player_ids': [p.teacher_id for p in players],

@@ -122,6 +122,13 @@ def padding_entity_info(traj_data, max_entity_num):
             traj_data['teacher_logit']['selected_units'], (0, pad, 0, su_pad), 'constant', -1e9)
         traj_data['teacher_logit']['target_unit'] = torch.nn.functional.pad(
             traj_data['teacher_logit']['target_unit'], (0, pad), 'constant', -1e9)
+        if 'successive_logit' in traj_data:     # DAPO lagged-self logits
+            sl = traj_data['successive_logit']
+            s_su_pad = MAX_SELECTED_UNITS_NUM - sl['selected_units'].shape[0]
+            sl['selected_units'] = torch.nn.functional.pad(
+                sl['selected_units'], (0, pad, 0, s_su_pad), 'constant', -1e9)
+            sl['target_unit'] = torch.nn.functional.pad(
+                sl['target_unit'], (0, pad), 'constant', -1e9)
         traj_data['mask']['selected_units_logits_mask'] = sequence_mask(
             traj_data['entity_num'].unsqueeze(0) + 1, max_len=max_entity_num + 1).squeeze(0)
         traj_data['mask']['target_units_logits_mask'] = sequence_mask(

@@ -178,6 +178,6 @@ def dapo_loss(target_policy_log_probs_dict, successive_policy_logits_dict, mask,
             kl = kl * mask['actions_mask'][head_type]
         kl = (kl * flag).mean()
         total = total + kl * head_weights_dict[head_type]
-        info['battle/' + head_type] = kl.item()
-    info['battle/total'] = total.item()
+        info['dapo/' + head_type] = kl.item()
+    info['dapo/total'] = total.item()
     return total, info

@@ -155,3 +155,50 @@ def test_stat_unit_num_tracking():
     assert stat.unit_num['max_unit_num'] == 3
     data = stat.get_stat_data()
     assert data[f'units/{name}'] == 1.0
+
+
+@pytest.mark.timeout(600)
+def test_agent_dapo_successive_logits_flow():
+    """With DAPO on and a successive model attached, every rollout step
+    carries successive_logit and the RL loss consumes it (reference
+    agent.py:506-515,566 + rl_loss.py dapo term)."""
+    from distar_amd.losses.rl_loss import ReinforcementLoss
+    torch.manual_seed(0)
+    cfg = Config({'common': {'type': 'train'},
+                  'actor': {'traj_len': 2, 'job_type': 'train'},
+                  'learner': {'use_dapo': True},
+                  'env': {'player_num': 2, 'max_episode_steps': 100000},
+                  'agent': {}})
+    env = MockSC2Env(cfg, entity_num_range=(24, 40), seed=3)
+    agent = Agent(cfg, env_id=0)
+    agent.player_id = 'MP0'
+    agent.successive_model = Model(cfg)
+    agent.successive_model.eval()
+    obs = env.reset()
+    agent.reset(obs=obs.get(0))
+    trajs, last_obs, done, steps = [], obs, False, 0
+    while not done and steps < 6 and not trajs:
+        actions = {0: agent.step(last_obs[0])[0]}
+        obs, rewards, done, infos = env.step(actions)
+        out = agent.collect_data(obs.get(0), rewards.get(0, 0), done, 0)
+        if out is not None:
+            trajs.append(out)
+        last_obs = {**last_obs, **obs}
+        steps += 1
+    assert trajs
+    assert all('successive_logit' in td for td in trajs[0][:-1])
+    batch = rl_collate([trajs[0]])
+    batch.pop('model_last_iter', None)
+    batch.pop('aux_type', None)
+    model = Model(Config({'common': {'type': 'train'},
+                          'model': {'enable_baselines': ['winloss']}}),
+                  use_value_network=True)
+    with torch.no_grad():
+        out = model.rl_learner_forward(**batch)
+        out['successive_logit'] = batch['successive_logit']
+        loss = ReinforcementLoss(Config({'learner': {'use_dapo': True},
+                                         'model': {'enable_baselines': ['winloss']}}).learner,
+                                 'MP0')
+        ld = loss.compute_loss(out)
+    assert torch.isfinite(ld['total_loss'])
+    assert 'dapo/total' in ld or any('dapo' in k for k in ld)

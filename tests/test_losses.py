@@ -104,7 +104,7 @@ def test_rl_loss_with_dapo():
     loss = ReinforcementLoss(Config({'use_dapo': True}), 'MP0')
     assert loss.use_dapo
     ld = loss.compute_loss(out)
-    assert 'battle/total' in ld and torch.isfinite(ld['total_loss'])
+    assert 'dapo/total' in ld and torch.isfinite(ld['total_loss'])
     ld['total_loss'].backward()
     # non-main players silently disable dapo
     loss2 = ReinforcementLoss(Config({'use_dapo': True}), 'EP0')
