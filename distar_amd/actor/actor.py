@@ -124,27 +124,42 @@ class Actor:
         return MockSC2Env(self._whole_cfg)
 
     def _start_batch_inference(self, agent_groups=None):
-        """Shared-slab batched inference (reference actor.py:268-299): one
-        device model serves every (env worker x player) slot; agents switch
-        to writing their obs into the slab and polling the signal."""
+        """Shared-slab batched inference (reference actor.py:268-299): ONE
+        server per PLAYER (each serving that player's model — and its teacher
+        slab on train jobs — across all env workers' slots); agents switch to
+        writing their obs into their server's slab and polling the signal."""
         import threading
         from .batch_inference import BatchInferenceServer
         device = 'cuda' if (self._cfg.use_cuda and torch.cuda.is_available()) \
             else 'cpu'
         agent_groups = agent_groups or [self._agents]
-        n_slots = sum(len(g) for g in agent_groups)
-        agent0 = agent_groups[0][0]
-        server = BatchInferenceServer(agent0.model.to(device),
-                                      env_num=n_slots, device=device)
-        slot = 0
-        for group in agent_groups:
-            for agent in group:
-                agent.attach_batch_inference(server, slot)
-                slot += 1
-        self._batch_server = server
-        self._batch_thread = threading.Thread(target=server.run, daemon=True)
-        self._batch_thread.start()
-        return server
+        env_n = len(agent_groups)
+        self._batch_servers, self._batch_threads = [], []
+        for i, agent0 in enumerate(self._agents):
+            if not getattr(agent0, 'HAS_MODEL', False):
+                self._batch_servers.append(None)
+                continue
+            teacher = agent0.teacher_model \
+                if 'train' in self._job_type and agent0.teacher_model is not None \
+                else None
+            server = BatchInferenceServer(
+                agent0.model.to(device), env_num=env_n, device=device,
+                teacher_model=teacher.to(device) if teacher is not None else None)
+            for env_id, group in enumerate(agent_groups):
+                group[i].attach_batch_inference(server, env_id)
+            t = threading.Thread(target=server.run, daemon=True)
+            t.start()
+            self._batch_servers.append(server)
+            self._batch_threads.append(t)
+        self._batch_server = next((x for x in self._batch_servers if x), None)
+        return self._batch_servers
+
+    def _stop_batch_inference(self):
+        for srv in getattr(self, '_batch_servers', []):
+            if srv is not None:
+                srv.stop()
+        for t in getattr(self, '_batch_threads', []):
+            t.join(timeout=5)
 
     def _update_models(self):
         if self._comm is None:
@@ -288,12 +303,12 @@ class Actor:
                 if time.time() > job_deadline:
                     self._setup_job()
                     if self._cfg.get('gpu_batch_inference', False):
-                        # new job -> new agents/models: restart the slab server
-                        self._batch_server.stop()
-                        self._batch_thread.join(timeout=5)
+                        # new job -> new agents/models: restart the slab servers
+                        self._stop_batch_inference()
                         self._start_batch_inference()
                     job_deadline = time.time() + self._cfg.actor_ask_for_job_interval
             env.close()
+            self._stop_batch_inference()
             return self.results
         workers = []
         for env_id in range(env_num):
@@ -306,7 +321,9 @@ class Actor:
             for t in workers:
                 t.join(timeout=0.5)
             self._update_models()
+        self._stop_batch_inference()
         return self.results
 
     def close(self):
         self._end = True
+        self._stop_batch_inference()

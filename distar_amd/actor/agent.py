@@ -280,6 +280,36 @@ class Agent:
             result['action_logp']['selected_units'][:max(su, 1)]
         return result
 
+    def _batched_teacher_logit(self, teacher_obs):
+        """Teacher KL through the server's teacher slab (reference
+        agent.py:715-739): write obs + recurrent state at our slot, signal,
+        poll, read back the trimmed logits."""
+        import time as _time
+        from .batch_inference import copy_input_data
+        server = self._batch_server
+        idx = self._env_id
+        payload = dict(teacher_obs)
+        payload['hidden_state'] = self._teacher_hidden_state
+        copy_input_data(server.teacher_input, payload, data_idx=idx)
+        server.teacher_signals[idx] += 1
+        while int(server.teacher_signals[idx]) != 0:
+            _time.sleep(0.001)
+        out = server.teacher_output
+        result = {
+            'logit': {k: v[idx].clone() for k, v in out['logit'].items()},
+            'entity_num': out['entity_num'][idx].clone(),
+            'selected_units_num': out['selected_units_num'][idx].clone(),
+            'hidden_state': [(out['hidden_state'][l][0][idx].clone(),
+                              out['hidden_state'][l][1][idx].clone())
+                             for l in range(len(out['hidden_state']))],
+        }
+        en = int(teacher_obs['entity_num'])
+        su = int(teacher_obs['selected_units_num'])
+        result['logit']['selected_units'] = \
+            result['logit']['selected_units'][:max(su, 1), :en + 1]
+        result['logit']['target_unit'] = result['logit']['target_unit'][:en]
+        return result
+
     def step(self, observation):
         if 'eval' in self._job_type and self._iter_count > 0:
             self._update_fake_reward(int(self._last_action_type),
@@ -346,13 +376,17 @@ class Agent:
             'selected_units_num': self._output['selected_units_num'],
             'action_info': self._output['action_info'],
         }
-        teacher_input = default_collate_with_dim([teacher_obs])
-        teacher_input['hidden_state'] = [(h.unsqueeze(0), c.unsqueeze(0))
-                                         for h, c in self._teacher_hidden_state]
-        teacher = self.teacher_model or self.model
-        with torch.no_grad():
-            teacher_output = teacher.compute_teacher_logit(**teacher_input)
-        teacher_output = self.decollate_output(teacher_output)
+        server = getattr(self, '_batch_server', None)
+        if server is not None and server.teacher_model is not None:
+            teacher_output = self._batched_teacher_logit(teacher_obs)
+        else:
+            teacher_input = default_collate_with_dim([teacher_obs])
+            teacher_input['hidden_state'] = [(h.unsqueeze(0), c.unsqueeze(0))
+                                             for h, c in self._teacher_hidden_state]
+            teacher = self.teacher_model or self.model
+            with torch.no_grad():
+                teacher_output = teacher.compute_teacher_logit(**teacher_input)
+            teacher_output = self.decollate_output(teacher_output)
         self._teacher_hidden_state = teacher_output['hidden_state']
 
         successive_output = None

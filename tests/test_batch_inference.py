@@ -54,7 +54,6 @@ def test_actor_batched_inference_mode():
                              'type': 'train'}})
     actor = Actor(cfg)
     results = actor.run()
-    actor._batch_server.stop()
     assert len(results) == 1
     assert all(agent._batch_server is not None for agent in actor._agents)
 
@@ -74,5 +73,48 @@ def test_actor_batch_inference_multi_env():
     actor = Actor(cfg)
     results = actor.run()
     assert len(results) >= 2
-    assert actor._batch_server.env_num == 4      # 2 envs x 2 players
-    actor._batch_server.stop()
+    # one server per player, each with one slot per env worker
+    assert len([s for s in actor._batch_servers if s is not None]) == 2
+    assert all(s.env_num == 2 for s in actor._batch_servers if s is not None)
+
+
+@pytest.mark.timeout(900)
+def test_teacher_slab_collect_data():
+    """collect_data routes the teacher forward through the server's teacher
+    slab when one is attached (reference agent.py:715-739)."""
+    import threading
+    from distar_amd.actor.agent import Agent
+    from distar_amd.envs.mock_env import MockSC2Env
+    torch.manual_seed(0)
+    cfg = Config({'common': {'type': 'train'},
+                  'actor': {'traj_len': 2, 'job_type': 'train'},
+                  'env': {'player_num': 1, 'max_episode_steps': 100000},
+                  'agent': {}})
+    env = MockSC2Env(cfg, entity_num_range=(24, 40), seed=2)
+    agent = Agent(cfg, env_id=0)
+    agent.player_id = 'MP0'
+    teacher = Model(cfg)
+    teacher.eval()
+    server = BatchInferenceServer(agent.model, env_num=1, device='cpu',
+                                  teacher_model=teacher)
+    agent.attach_batch_inference(server, 0)
+    thread = threading.Thread(target=server.run, daemon=True)
+    thread.start()
+    try:
+        obs = env.reset()
+        agent.reset(obs=obs.get(0))
+        traj, last_obs, done, steps = None, obs, False, 0
+        while not done and steps < 5 and traj is None:
+            actions = {0: agent.step(last_obs[0])[0]}
+            obs, rewards, done, infos = env.step(actions)
+            traj = agent.collect_data(obs.get(0), rewards.get(0, 0), done, 0)
+            last_obs = {**last_obs, **obs}
+            steps += 1
+        assert traj is not None
+        tl = traj[0]['teacher_logit']
+        assert set(tl) >= {'action_type', 'delay', 'queued', 'selected_units',
+                           'target_unit', 'target_location'}
+        assert torch.isfinite(tl['action_type']).all()
+    finally:
+        server.stop()
+        thread.join(timeout=10)
