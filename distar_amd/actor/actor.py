@@ -64,9 +64,12 @@ class Actor:
             job = self._comm.ask_for_job(
                 self._job_type, player_id=self._cfg.get('job_player_id'))
         else:
+            # standalone jobs (play/eval or comm-less training) take model
+            # checkpoints from actor.model{0,1}_path (bin/play.py contract)
             job = {'player_ids': ['MP0', 'MP1'],
                    'pipelines': ['default', 'default'],
-                   'checkpoint_paths': ['none', 'none'],
+                   'checkpoint_paths': [self._cfg.get('model0_path', 'none'),
+                                        self._cfg.get('model1_path', 'none')],
                    'teacher_checkpoint_paths': ['none', 'none'],
                    'z_path': ['3map.json', '3map.json'],
                    'send_data_players': ['MP0'],

@@ -96,3 +96,31 @@ def test_bench_distributed_cpu_gloo_rl(tmp_path):
     res = _json.loads(json_lines[0])
     assert res['n_gpus'] == 2 and res['config']['mode'] == 'rl'
     assert res['config']['value_feature'] is True
+
+
+@pytest.mark.timeout(600)
+def test_play_loads_model_checkpoints(tmp_path, monkeypatch):
+    """play/eval without a league: actor loads actor.model{0,1}_path into the
+    agents (reference play.py model_paths contract)."""
+    monkeypatch.chdir(tmp_path)
+    from distar_amd.actor.actor import Actor
+    from distar_amd.models import Model
+    from distar_amd.utils.checkpoint import CheckpointHelper
+    from distar_amd.utils.config import Config
+    torch.manual_seed(0)
+    m = Model(Config({}))
+    ckpt = str(tmp_path / 'custom.pth')
+    CheckpointHelper().save(ckpt, m)
+    cfg = Config({'actor': {'episode_num': 1, 'env_type': 'mock',
+                            'traj_len': 4, 'job_type': 'eval_test',
+                            'model0_path': ckpt},
+                  'env': {'player_num': 2, 'max_episode_steps': 3},
+                  'common': {'experiment_name': 'test_play_load',
+                             'type': 'play'}})
+    actor = Actor(cfg)
+    actor._setup_job()
+    a0 = actor._agents[0]
+    sd = m.state_dict()
+    got = a0.model.state_dict()
+    k = next(iter(sd))
+    torch.testing.assert_close(got[k], sd[k])
