@@ -38,6 +38,60 @@ def create_league_server(league, host='0.0.0.0', port=None):
         player.reset_flag = True
         return {'ok': True}
 
+    def _players_of(body):
+        pid = body.get('player_id')
+        if pid:
+            return [league.all_players[pid]]
+        return list(league.active_players.values())
+
+    def show_dist_stat(body):
+        return {p.player_id: p.dist_stat.stat_info_dict
+                for p in _players_of(body) if hasattr(p, 'dist_stat')}
+
+    def show_cum_stat(body):
+        return {p.player_id: p.cum_stat.stat_info_dict
+                for p in _players_of(body) if hasattr(p, 'cum_stat')}
+
+    def show_unit_num_stat(body):
+        return {p.player_id: p.unit_num_stat.stat_info_dict
+                for p in _players_of(body) if hasattr(p, 'unit_num_stat')}
+
+    def show_trueskill(body):
+        return {'mu': dict(league.trueskill.mu),
+                'sigma': dict(league.trueskill.sigma)}
+
+    def show_config(body):
+        def plain(x):
+            if isinstance(x, dict):
+                return {k: plain(v) for k, v in x.items()}
+            if isinstance(x, (list, tuple)):
+                return [plain(v) for v in x]
+            return x if isinstance(x, (int, float, bool, str, type(None))) \
+                else str(x)
+        return plain(dict(league.cfg))
+
+    def update_config(body):
+        from ..utils.config import Config, deep_merge_dicts
+        league.cfg = deep_merge_dicts(league.cfg,
+                                      Config(body.get('overrides', {})))
+        return {'ok': True}
+
+    def remove_hist_player(body):
+        removed = league.historical_players.pop(body['player_id'], None)
+        return {'ok': removed is not None}
+
+    def display_player(body):
+        out = {}
+        for p in _players_of(body):
+            out[p.player_id] = {
+                'checkpoint_path': p.checkpoint_path,
+                'pipeline': p.pipeline, 'frac_id': p.frac_id,
+                'z_path': p.z_path, 'z_prob': p.z_prob,
+                'total_agent_step': getattr(p, 'total_agent_step', 0),
+                'total_game_count': getattr(p, 'total_game_count', 0),
+            }
+        return out
+
     server = JsonHttpServer({
         '/league/register_learner': league.deal_with_register_learner,
         '/league/learner_send_train_info': league.deal_with_learner_send_train_info,
@@ -50,5 +104,13 @@ def create_league_server(league, host='0.0.0.0', port=None):
         '/league/snapshot_player': snapshot_player,
         '/league/reset_player': reset_player,
         '/league/save_resume': lambda body: {'path': league.save_resume()},
+        '/league/show_dist_stat': show_dist_stat,
+        '/league/show_cum_stat': show_cum_stat,
+        '/league/show_unit_num_stat': show_unit_num_stat,
+        '/league/show_trueskill': show_trueskill,
+        '/league/show_config': show_config,
+        '/league/update_config': update_config,
+        '/league/remove_hist_player': remove_hist_player,
+        '/league/display_player': display_player,
     }, host=host, port=port)
     return server

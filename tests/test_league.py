@@ -264,3 +264,33 @@ def test_phase_gate_snapshot_protocol(league):
     league.deal_with_learner_send_train_info(
         {'player_id': 'MP0', 'train_steps': 0, 'checkpoint_path': 'mp0.pth'})
     assert len(league.historical_players) >= n_hist
+
+
+def test_league_debug_api_extended(league):
+    """Debug/ops endpoints (reference league_api.py:56-305 subset)."""
+    import requests
+    from distar_amd.league.api import create_league_server
+    api = create_league_server(league, host='127.0.0.1').start()
+    url = f'http://127.0.0.1:{api.port}'
+    try:
+        r = requests.post(f'{url}/league/show_dist_stat', json={})
+        assert 'MP0' in r.json()
+        r = requests.post(f'{url}/league/show_trueskill', json={})
+        assert 'mu' in r.json()
+        r = requests.post(f'{url}/league/show_config', json={})
+        assert 'branch_probs' in r.json()
+        r = requests.post(f'{url}/league/display_player',
+                          json={'player_id': 'MP0'})
+        assert r.json()['MP0']['pipeline'] == 'default'
+        r = requests.post(f'{url}/league/update_config',
+                          json={'overrides': {'print_freq': 7}})
+        assert r.json()['ok'] and league.cfg.print_freq == 7
+        # snapshot then remove a historical player over the API
+        mp = league.active_players['MP0']
+        hp = mp.snapshot()
+        league.set_hist_player(hp)
+        r = requests.post(f'{url}/league/remove_hist_player',
+                          json={'player_id': hp.player_id})
+        assert r.json()['ok'] and hp.player_id not in league.historical_players
+    finally:
+        api.stop()
