@@ -118,3 +118,41 @@ def test_rl_dataloader_pulls_and_collates():
         if loader is not None:
             loader.close()
         coord.close()
+
+
+@pytest.mark.timeout(300)
+def test_model_publication_refreshes_actor_weights(tmp_path, monkeypatch):
+    """Learner publishes a policy state_dict; the actor's pull_model applies
+    it to the agent (reference learner_comm.py:53-99 + actor_comm.py:172-196).
+    Value-network keys are stripped from the wire payload."""
+    monkeypatch.chdir(tmp_path)
+    from types import SimpleNamespace
+    from distar_amd.actor.comm import ActorComm, LearnerComm
+    from distar_amd.models import Model
+    from distar_amd.utils.config import Config
+    from distar_amd.utils.checkpoint import CountVar
+    torch.manual_seed(0)
+    coord = Coordinator().run()
+    try:
+        cfg = Config({'learner': {'player_id': 'MP0'},
+                      'communication': {'coordinator_ip': '127.0.0.1',
+                                        'coordinator_port': coord.port}})
+        lcomm = LearnerComm(cfg)
+        learner_model = Model(Config({}), use_value_network=True)
+        fake_learner = SimpleNamespace(model=learner_model,
+                                       last_iter=CountVar(123))
+        lcomm.send_model(fake_learner)
+        acomm = ActorComm(cfg)
+        payload = acomm.pull_model('MP0', timeout=30)
+        assert payload is not None and payload['model_last_iter'] == 123
+        assert not any(k.startswith('value_networks') for k in payload['model'])
+        actor_model = Model(Config({}))
+        with torch.no_grad():
+            for p in actor_model.parameters():
+                p.zero_()
+        actor_model.load_state_dict(payload['model'], strict=False)
+        k = next(k for k in payload['model'])
+        torch.testing.assert_close(actor_model.state_dict()[k],
+                                   learner_model.state_dict()[k])
+    finally:
+        coord.close()
