@@ -81,7 +81,7 @@ def test_rl_league_loop(tmp_path, monkeypatch):
         coord.close()
 
 
-@pytest.mark.timeout(900)
+@pytest.mark.timeout(1800)
 def test_three_player_league_loop(tmp_path, monkeypatch):
     """Config-5 league shape: MP0 (main) + ME0 (main exploiter) + EP0
     (exploiter) sharing one coordinator+league; each player's learner runs 1
@@ -159,7 +159,7 @@ def test_three_player_league_loop(tmp_path, monkeypatch):
                     for pid, ln in zip(ids, learners)]
         for t in lthreads:
             t.start()
-        deadline = time.time() + 600
+        deadline = time.time() + 1200       # generous: loaded CI hosts
         while len(done) < n and time.time() < deadline:
             time.sleep(1)
         assert done == {pid: 1 for pid in ids}, f'learners finished: {done}'
