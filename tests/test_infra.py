@@ -1,0 +1,69 @@
+"""Learner-hook registry, timers, scalar logging (reference
+`ctools/worker/learner/learner_hook.py`, `ctools/utils/time_helper.py`,
+`ctools/utils/log_helper.py`)."""
+import json
+import os
+import time
+from types import SimpleNamespace
+
+import torch
+
+from distar_amd.learner.hooks import (Hook, add_learner_hook,
+                                      build_learner_hook_by_cfg)
+from distar_amd.utils.log import ScalarLogger
+from distar_amd.utils.timing import EasyTimer
+
+
+class _Probe(Hook):
+    calls = []
+
+    def __init__(self, tag, **kwargs):
+        super().__init__(tag, **kwargs)
+        self.tag = tag
+
+    def __call__(self, engine):
+        _Probe.calls.append(self.tag)
+
+
+def test_hooks_run_in_priority_order():
+    """Lower priority number runs first within a position (reference
+    learner_hook.py registry semantics)."""
+    _Probe.calls = []
+    hooks = {p: [] for p in ('before_run', 'before_iter', 'after_iter',
+                             'after_run')}
+    add_learner_hook(hooks, _Probe('late', priority=90, position='after_iter'))
+    add_learner_hook(hooks, _Probe('early', priority=10, position='after_iter'))
+    add_learner_hook(hooks, _Probe('mid', priority=50, position='after_iter'))
+    engine = SimpleNamespace()
+    for h in hooks['after_iter']:
+        h(engine)
+    assert _Probe.calls == ['early', 'mid', 'late']
+
+
+def test_build_hooks_from_cfg():
+    from distar_amd.utils.config import Config
+    cfg = Config({'after_iter': {'log_show': {'ext_args': {'freq': 5}}},
+                  'before_run': {'load_ckpt': {'ext_args': {}}}})
+    hooks = build_learner_hook_by_cfg(cfg)
+    assert any(h.name == 'log_show' for h in hooks['after_iter'])
+    assert any(h.name == 'load_ckpt' for h in hooks['before_run'])
+
+
+def test_easy_timer_cpu():
+    t = EasyTimer(cuda=False)
+    with t:
+        time.sleep(0.05)
+    assert 0.04 < t.value < 1.0
+
+
+def test_scalar_logger_jsonl(tmp_path):
+    logger = ScalarLogger(str(tmp_path), name='test')
+    logger.register_var('loss')
+    logger.add_scalar('loss', 1.5, global_step=3)
+    logger.add_scalar('loss', torch.tensor(2.5), global_step=4)
+    logger.flush()
+    lines = [json.loads(l) for l in
+             open(os.path.join(tmp_path, 'test.jsonl'))]
+    assert lines[0] == {**lines[0], 'step': 3, 'key': 'loss', 'value': 1.5}
+    assert lines[1]['value'] == 2.5
+    logger.close()
