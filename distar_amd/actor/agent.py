@@ -19,7 +19,7 @@ import copy
 import json
 import os
 import random
-from collections import defaultdict
+from collections import defaultdict, deque
 from functools import partial
 
 import torch
@@ -67,7 +67,7 @@ class Agent:
         self._hidden_size = self.model.cfg.encoder.core_lstm.hidden_size
         self._stat_api = Stat('zerg')
         self._model_last_iter = 0
-        self._data_buffer = []
+        self._data_buffer = deque(maxlen=self._traj_len)
         self._push_count = 0
         self._iter_count = 0
 
@@ -78,7 +78,7 @@ class Agent:
         self.race = race
         self._stat_api = Stat(race)
         self._iter_count = 0
-        self._data_buffer = []
+        self._data_buffer = deque(maxlen=self._traj_len)
         self._push_count = 0
         z = torch.zeros
         self._hidden_state = [(z(self._hidden_size), z(self._hidden_size))
@@ -460,8 +460,11 @@ class Agent:
             if self._use_value_feature and 'value_feature' in last_obs:
                 last_step['value_feature'] = dict(last_obs['value_feature'])
                 last_step['value_feature'].update(self.get_behavior_z())
+            # sliding window (reference agent.py:173,571-604): the deque is
+            # NOT cleared between sends, so an episode-end window re-reaches
+            # back to a full traj_len steps instead of going out short (the
+            # learner collate requires uniform T)
             data = list(self._data_buffer) + [last_step]
-            self._data_buffer = []
             self._push_count = 0
             return data
         return None

@@ -89,11 +89,27 @@ class RLDataLoader:
                 data = new + data + new          # double re-injection
             if self._stop:
                 return
+            # collate needs uniform T: batch same-length trajectories (a
+            # trajectory from an episode shorter than traj_len is shorter)
+            from collections import Counter
+            counts = Counter(len(d) for d in data)
+            want, n_want = counts.most_common(1)[0]
+            if n_want < self.batch_size:
+                # drop one minority-length trajectory and pull more
+                for i, d in enumerate(data):
+                    if len(d) != want:
+                        data.pop(i)
+                        break
+                continue
+            picked, rest = [], []
+            for d in data:
+                (picked if len(d) == want and len(picked) < self.batch_size
+                 else rest).append(d)
             try:
-                batch = self.collate_fn(data[:self.batch_size])
+                batch = self.collate_fn(picked)
             except Exception as e:  # noqa: BLE001 - skip malformed trajectories
                 print(f'[RLDataLoader] collate failed: {e!r}')
-                data = data[self.batch_size:]
+                data = rest
                 continue
             if self._stager is not None:
                 batch = self._stager.stage(batch)
@@ -103,7 +119,7 @@ class RLDataLoader:
                     break
                 except queue.Full:
                     continue
-            data = data[self.batch_size:]
+            data = rest
             random.shuffle(data)
 
     def __iter__(self):

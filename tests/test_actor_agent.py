@@ -202,3 +202,31 @@ def test_agent_dapo_successive_logits_flow():
         ld = loss.compute_loss(out)
     assert torch.isfinite(ld['total_loss'])
     assert 'dapo/total' in ld or any('dapo' in k for k in ld)
+
+
+def test_episode_end_window_is_full_length():
+    """Episode-end trajectories slide back to a full traj_len window
+    (reference agent.py:173 deque(maxlen) semantics) so the learner collate
+    always sees uniform T."""
+    torch.manual_seed(0)
+    cfg = Config({'common': {'type': 'train'},
+                  'actor': {'traj_len': 4, 'job_type': 'train'},
+                  'env': {'player_num': 1, 'max_episode_steps': 2000},
+                  'agent': {}})
+    env = MockSC2Env(cfg, entity_num_range=(24, 32), seed=5)
+    agent = Agent(cfg, env_id=0)
+    agent.player_id = 'MP0'
+    obs = env.reset()
+    agent.reset(obs=obs.get(0))
+    lengths, last_obs, done = [], obs, False
+    while not done:
+        actions = {0: agent.step(last_obs[0])[0]}
+        obs, rewards, done, infos = env.step(actions)
+        out = agent.collect_data(obs.get(0), rewards.get(0, 0), done, 0)
+        if out is not None:
+            lengths.append(len(out))
+        last_obs = {**last_obs, **obs}
+    assert len(lengths) >= 2, f'need multiple windows, got {lengths}'
+    # every emitted window has exactly traj_len steps + bootstrap frame,
+    # including the episode-end one (slides back over earlier steps)
+    assert all(l == 5 for l in lengths), lengths
