@@ -145,6 +145,9 @@ class SelectedUnitsHead(nn.Module):
         stdv = 1. / math.sqrt(self.end_embedding.size(1))
         self.end_embedding.data.uniform_(-stdv, stdv)
         self.extra_units = self.whole_cfg.get('agent', {}).get('extra_units', False)
+        # free-running rollout during train for the IoU metric (reference
+        # action_arg_head.py:173,218-259)
+        self.test_iou = self.cfg.get('test_iou', False)
         # 'entity_num'/'constant' are dead paths upstream (the reference's
         # _get_key_mask calls a self.embed_fc that is never constructed,
         # action_arg_head.py:131-136 vs :98-99); the working set is:
@@ -269,7 +272,59 @@ class SelectedUnitsHead(nn.Module):
         mask[arange, 0, entity_num] = False
         logits = logits.masked_fill(~mask, -1e9)
         final_ae = autoregressive_embedding + final_delta
-        return logits, None, final_ae, selected_units_num, None
+        results = None
+        if self.test_iou:
+            results = self._iou_rollout(key, entity_num, autoregressive_embedding,
+                                        logits_mask, key_embeddings, seq_len)
+        return logits, results, final_ae, selected_units_num, results
+
+    def _iou_rollout(self, key, entity_num, autoregressive_embedding,
+                     logits_mask, key_embeddings, seq_len):
+        """Free-running greedy-sampled selections capped at the teacher
+        sequence length (reference action_arg_head.py:218-259); used only for
+        the SL IoU metric, no gradients."""
+        with torch.no_grad():
+            bs = autoregressive_embedding.shape[0]
+            device = autoregressive_embedding.device
+            ae = autoregressive_embedding
+            end_flag = torch.zeros(bs, dtype=torch.bool, device=device)
+            mask = logits_mask.clone()
+            arange = torch.arange(bs, device=device)
+            mask[arange, entity_num] = False
+            state = [(ae.new_zeros(bs, self.cfg.hidden_dim),
+                      ae.new_zeros(bs, self.cfg.hidden_dim))
+                     for _ in range(self.num_layers)]
+            sel_oh = ae.new_zeros(bs, key.shape[1], 1)
+            results = []
+            result = None
+            for i in range(seq_len):
+                if i == 1:
+                    mask[arange, entity_num] = True
+                if result is not None:
+                    mask[arange, result] = False
+                lstm_input = self.query_fc2(self.query_fc1(ae)).unsqueeze(0)
+                lstm_output, state = self.lstm(lstm_input, state)
+                step_logits = (lstm_output.permute(1, 0, 2) * key).sum(dim=2)
+                step_logits = step_logits.masked_fill(~mask, -1e9)
+                result = torch.multinomial(F.softmax(step_logits, dim=-1), 1)[:, 0]
+                end_flag[result == entity_num] = True
+                results.append(result)
+                sel_oh = sel_oh.clone()
+                sel_oh[arange[~end_flag], result[~end_flag]] = 1
+                if self.reduce_type == 'selected_units_num':
+                    emb = (key_embeddings * sel_oh).sum(dim=1)
+                    cnt = sel_oh.sum(dim=1)
+                    nz = (cnt > 0).squeeze(-1)
+                    emb[nz] = emb[nz] / cnt[nz]
+                    ae = autoregressive_embedding + self.embed_fc2(self.embed_fc1(emb))
+                elif self.reduce_type == 'attention_pool':
+                    ae = autoregressive_embedding + \
+                        self.attention_pool(key_embeddings, mask=sel_oh)
+                else:
+                    ae = autoregressive_embedding + self.attention_pool(
+                        key_embeddings, num=sel_oh.sum(dim=1).squeeze(-1),
+                        mask=sel_oh)
+            return torch.stack(results, dim=0).transpose(1, 0).contiguous()
 
     def _query_sample_hip(self, key, entity_num, autoregressive_embedding,
                           logits_mask, key_embeddings, su_mask, uniforms=None):

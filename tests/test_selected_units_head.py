@@ -182,3 +182,27 @@ def test_extra_units_last_step_semantics():
     ended = results.gather(1, last.unsqueeze(1)).squeeze(1) == entity_num
     # rows that ended with the end token must have zero extras
     assert (extra[ended] == 0).all()
+
+
+def test_test_iou_free_running_rollout():
+    """test_iou: the train path additionally emits free-running selections
+    (no-grad) whose IoU vs labels the SL loss reports (reference
+    action_arg_head.py:173,218-259 + sl_loss.py IoU)."""
+    from distar_amd.lib.fake_data import fake_sl_batch_fast
+    from distar_amd.losses import SupervisedLoss
+    torch.manual_seed(0)
+    m = Model(Config({'model': {'policy': {'head': {
+        'selected_units_head': {'test_iou': True}}}}}))
+    data = fake_sl_batch_fast(batch_size=2, traj_len=2)
+    hidden = [(torch.zeros(2, 384), torch.zeros(2, 384)) for _ in range(3)]
+    logits, infer, _ = m.sl_train(**data, hidden_state=hidden)
+    su = infer['selected_units']
+    assert su is not None and su.shape[0] == 4          # (B*T) rows
+    ld = SupervisedLoss(Config({'learner': {}})).compute_loss(
+        logits, data['action_info'], data['action_mask'],
+        data['selected_units_num'], data['entity_num'], infer)
+    assert 0.0 <= float(ld['selected_units_iou']) <= 1.0
+    # default config: no free-running rollout, IoU metric reads 0
+    m2 = Model(Config({}))
+    logits2, infer2, _ = m2.sl_train(**data, hidden_state=hidden)
+    assert infer2['selected_units'] is None
