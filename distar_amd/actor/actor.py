@@ -90,7 +90,20 @@ class Actor:
             if ckpt not in ('none', None) and not self._cfg.fake_model and \
                     getattr(agent, 'HAS_MODEL', False):
                 try:
-                    self._ckpt_helper.load(ckpt, agent.model, strict=False,
+                    raw = torch.load(str(ckpt), map_location='cpu',
+                                     weights_only=False)
+                    if isinstance(raw, dict) and 'map_name' in raw:
+                        # play checkpoints carry their strategy context
+                        # (reference actor.py:65-73)
+                        job.setdefault('env_info', {})['map_name'] = raw['map_name']
+                        agent._cfg = dict(agent._cfg)
+                        if 'z_path' in raw:
+                            agent._cfg['z_path'] = raw['z_path']
+                        if 'fake_reward_prob' in raw:
+                            agent._cfg['fake_reward_prob'] = raw['fake_reward_prob']
+                        agent.z_idx = raw.get('z_idx')
+                    self._ckpt_helper.load(raw, agent.model, strict=False,
+                                           need_torch_load=False,
                                            logger_prints=self._logger.info)
                 except FileNotFoundError:
                     self._logger.info(f'checkpoint missing: {ckpt}, random init')

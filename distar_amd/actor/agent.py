@@ -61,6 +61,7 @@ class Agent:
         self.model.eval()
         self.teacher_model = None
         self.successive_model = None    # DAPO (reference agent.py:506-515)
+        self.z_idx = None               # play-ckpt curated Z subset (agent.py:119,202-204)
         self.player_id = 'MP0'
         self.race = 'zerg'
         self._num_layers = self.model.cfg.encoder.core_lstm.num_layers
@@ -125,11 +126,22 @@ class Agent:
         map_entry = z_data.get(map_name) or next(iter(z_data.values()), {})
         race_entry = map_entry.get(mix_race) or map_entry.get(race) or \
             next(iter(map_entry.values()), {})
+        z_type = None
         if race_entry:
             born = self._born_location_key(obs, race_entry)
             zs = race_entry.get(born) or next(iter(race_entry.values()))
-            entry = random.choice(zs)
-        z_type = None
+            if self.z_idx is not None:
+                # curated subset shipped inside a play checkpoint (reference
+                # agent.py:202-204): pick (index, z_type) pairs
+                curated = self.z_idx.get(map_name, {}).get(mix_race, {}) \
+                    .get(str(born)) if isinstance(self.z_idx, dict) else None
+                if curated:
+                    idx, z_type = random.choice(curated)
+                    entry = zs[idx]
+                else:
+                    entry = random.choice(zs)
+            else:
+                entry = random.choice(zs)
         if entry is not None:
             if len(entry) == 5:
                 bo, cum, bo_loc, self._target_z_loop, z_type = entry

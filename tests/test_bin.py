@@ -124,3 +124,31 @@ def test_play_loads_model_checkpoints(tmp_path, monkeypatch):
     got = a0.model.state_dict()
     k = next(iter(sd))
     torch.testing.assert_close(got[k], sd[k])
+
+
+@pytest.mark.timeout(600)
+def test_play_checkpoint_strategy_context(tmp_path, monkeypatch):
+    """Play checkpoints carrying map_name/z_path/fake_reward_prob/z_idx apply
+    to the loaded agent (reference actor.py:65-73, agent.py:202-204)."""
+    monkeypatch.chdir(tmp_path)
+    from distar_amd.actor.actor import Actor
+    from distar_amd.models import Model
+    from distar_amd.utils.config import Config
+    torch.manual_seed(0)
+    m = Model(Config({}))
+    ckpt = str(tmp_path / 'play.pth')
+    torch.save({'model': m.state_dict(), 'map_name': 'NewRepugnancy',
+                'z_path': '7map_filter_spine.json', 'fake_reward_prob': 0.25,
+                'z_idx': None}, ckpt)
+    cfg = Config({'actor': {'episode_num': 1, 'env_type': 'mock',
+                            'traj_len': 4, 'job_type': 'eval_test',
+                            'model0_path': ckpt},
+                  'env': {'player_num': 2, 'max_episode_steps': 3},
+                  'common': {'experiment_name': 'test_play_ctx',
+                             'type': 'play'}})
+    actor = Actor(cfg)
+    job = actor._setup_job()
+    a0 = actor._agents[0]
+    assert job['env_info']['map_name'] == 'NewRepugnancy'
+    assert a0._cfg['z_path'] == '7map_filter_spine.json'
+    assert a0._cfg['fake_reward_prob'] == 0.25
