@@ -74,6 +74,17 @@ class BaseLearner:
         self._checkpoint_helper = CheckpointHelper(self._rank)
         self._logger, self._scalar_logger, self._record = build_logger(
             self._whole_cfg, name=self._name, rank=self._rank)
+        if self._rank == 0:
+            # back up the fully-merged config for reproducibility (reference
+            # bin/rl_train.py:27-42 copies configs into the experiment dir)
+            from ..utils.config import save_config
+            cfg_dir = os.path.join(self._exp_dir, 'config_backup')
+            os.makedirs(cfg_dir, exist_ok=True)
+            try:
+                save_config(self._whole_cfg,
+                            os.path.join(cfg_dir, f'{self._name}_whole_config.yaml'))
+            except Exception as e:  # noqa: BLE001 - never block training on this
+                self.info(f'config backup failed: {e!r}')
         self._setup_model()
         if self._use_cuda:
             self._model = self._model.cuda()

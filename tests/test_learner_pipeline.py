@@ -191,3 +191,21 @@ def test_remote_sl_dataloader_via_adapter(tmp_path, monkeypatch):
         if loader is not None:
             loader.close()
         coord.close()
+
+
+def test_config_backup_written(tmp_path, monkeypatch):
+    """Learners back up the fully-merged config into the experiment dir
+    (reference bin/rl_train.py:27-42)."""
+    monkeypatch.chdir(tmp_path)
+    import yaml
+    from distar_amd.learner.sl_learner import SLLearner
+    cfg = Config({'learner': {'job_type': 'fake', 'use_cuda': False,
+                              'data': {'batch_size': 2, 'trajectory_length': 3}},
+                  'common': {'experiment_name': 'test_cfg_backup',
+                             'type': 'train'}})
+    learner = SLLearner(cfg)
+    path = os.path.join(learner._exp_dir, 'config_backup',
+                        'SLLearner_whole_config.yaml')
+    assert os.path.isfile(path)
+    loaded = yaml.safe_load(open(path))
+    assert loaded['learner']['data']['batch_size'] == 2
