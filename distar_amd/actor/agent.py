@@ -32,6 +32,7 @@ from ..lib.stat import Stat, cum_dict
 from ..models.alphastar.model import Model
 from ..utils.data import default_collate_with_dim, to_device
 from ..utils.metric import hamming_distance, l2_distance, levenshtein_distance
+from ..utils.timing import sw
 
 Z_DIR = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
                      'assets', 'z_files')
@@ -322,6 +323,7 @@ class Agent:
         result['logit']['target_unit'] = result['logit']['target_unit'][:en]
         return result
 
+    @sw.decorate('agent_step')
     def step(self, observation):
         if 'eval' in self._job_type and self._iter_count > 0:
             self._update_fake_reward(int(self._last_action_type),

@@ -27,6 +27,7 @@ from .actions import (ACTIONS, ABILITY_TO_QUEUE_ACTION, BEGINNING_ORDER_ACTIONS,
 from .consts import (ACTION_INFO, BEGINNING_ORDER_LENGTH, EFFECT_LEN,
                      ENTITY_INFO, MAX_ENTITY_NUM, MAX_SELECTED_UNITS_NUM,
                      SPATIAL_INFO, SPATIAL_SIZE, UPGRADE_LENGTH)
+from ..utils.timing import sw
 from .static_data import (ADDON_REORDER_ARRAY, BUFFS_REORDER_ARRAY,
                           NUM_UNIT_TYPES, NUM_UPGRADES,
                           UNIT_TYPES_REORDER_ARRAY, UPGRADES_REORDER_ARRAY)
@@ -126,6 +127,7 @@ class Features:
     away_born_location = property(lambda self: self._away_born_location)
 
     # ------------------------------------------------------------------- obs
+    @sw.decorate('transform_obs')
     def transform_obs(self, obs, padding_spatial=False, opponent_obs=None):
         spatial_info = defaultdict(list)
         scalar_info = {}

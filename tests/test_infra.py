@@ -67,3 +67,27 @@ def test_scalar_logger_jsonl(tmp_path):
     assert lines[0] == {**lines[0], 'step': 3, 'key': 'loss', 'value': 1.5}
     assert lines[1]['value'] == 2.5
     logger.close()
+
+
+def test_stopwatch_hierarchy_and_report():
+    from distar_amd.utils.timing import Stopwatch
+    s = Stopwatch(enabled=True)
+    for _ in range(3):
+        with s('outer'):
+            time.sleep(0.01)
+            with s('inner'):
+                time.sleep(0.01)
+    report = str(s)
+    assert 'outer' in report and 'outer.inner' in report
+    assert s._times['outer'][0] == 3 and s._times['outer.inner'][0] == 3
+    assert s._times['outer'][1] >= s._times['outer.inner'][1]
+    # disabled stopwatch records nothing
+    s2 = Stopwatch(enabled=False)
+    with s2('x'):
+        pass
+    assert not s2._times
+
+    @s.decorate('deco')
+    def f():
+        return 41 + 1
+    assert f() == 42 and 'deco' in s._times
