@@ -85,3 +85,27 @@ def get_tensor_data(data: Any) -> Any:
     if isinstance(data, Sequence) and not isinstance(data, str):
         return type(data)(get_tensor_data(v) for v in data)
     return data
+
+
+def seed_everything(seed, deterministic=False):
+    """Seed python/numpy/torch (+ cuda) in one call; with ``deterministic``
+    also flips torch into deterministic-algorithms mode (the
+    reference has no deterministic test mode; this is the rebuild's).
+
+    Returns the seed so callers can log it.
+    """
+    import os
+    import random as _random
+
+    import numpy as _np
+    _random.seed(seed)
+    _np.random.seed(seed % (2 ** 32))
+    torch.manual_seed(seed)
+    if torch.cuda.is_available():
+        torch.cuda.manual_seed_all(seed)
+    if deterministic:
+        os.environ.setdefault('CUBLAS_WORKSPACE_CONFIG', ':4096:8')
+        torch.use_deterministic_algorithms(True, warn_only=True)
+        torch.backends.cudnn.deterministic = True     # MIOpen on ROCm
+        torch.backends.cudnn.benchmark = False
+    return seed

@@ -91,3 +91,26 @@ def test_stopwatch_hierarchy_and_report():
     def f():
         return 41 + 1
     assert f() == 42 and 'deco' in s._times
+
+
+def test_seed_everything_reproducible_training():
+    """Two seeded SL steps produce bit-identical losses (deterministic mode,
+    SURVEY §5.2 rebuild requirement)."""
+    from distar_amd.lib.fake_data import fake_sl_batch_fast
+    from distar_amd.losses import SupervisedLoss
+    from distar_amd.models import Model
+    from distar_amd.utils.config import Config
+    from distar_amd.utils.misc import seed_everything
+
+    def one_loss():
+        seed_everything(1234)
+        m = Model(Config({}))
+        data = fake_sl_batch_fast(batch_size=2, traj_len=2)
+        hidden = [(torch.zeros(2, 384), torch.zeros(2, 384)) for _ in range(3)]
+        logits, infer, _ = m.sl_train(**data, hidden_state=hidden)
+        ld = SupervisedLoss(Config({'learner': {}})).compute_loss(
+            logits, data['action_info'], data['action_mask'],
+            data['selected_units_num'], data['entity_num'], infer)
+        return float(ld['total_loss'])
+
+    assert one_loss() == one_loss()
