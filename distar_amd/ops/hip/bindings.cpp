@@ -202,6 +202,16 @@ extern "C" __global__ void masked_ce_bwd_kernel(
     const float*, const long*, const float*, const float*, const float*,
     float*, int, int);
 
+extern "C" __global__ void entropy_fwd_kernel(
+    const float*, float*, float*, int, int);
+extern "C" __global__ void entropy_bwd_kernel(
+    const float*, const float*, const float*, const float*, float*, int, int);
+extern "C" __global__ void kl_fwd_kernel(
+    const float*, const float*, float*, float*, float*, int, int);
+extern "C" __global__ void kl_bwd_kernel(
+    const float*, const float*, const float*, const float*, const float*,
+    float*, int, int);
+
 extern "C" __global__ void su_sample_kernel(
     const float*, const __hip_bfloat16*, const unsigned char*,
     const unsigned char*, const int*, const float*,
@@ -324,7 +334,69 @@ torch::Tensor masked_ce_bwd(torch::Tensor logits, torch::Tensor labels,
   return dlogits;
 }
 
+std::vector<torch::Tensor> entropy_fwd(torch::Tensor logits) {
+  check_2d(logits, "logits");
+  int64_t N = logits.size(0), C = logits.size(1);
+  auto ent = torch::empty({N}, logits.options());
+  auto lse = torch::empty({N}, logits.options());
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(entropy_fwd_kernel, dim3((unsigned)N), dim3(256), 0,
+                     stream.stream(), logits.data_ptr<float>(),
+                     ent.data_ptr<float>(), lse.data_ptr<float>(),
+                     (int)N, (int)C);
+  return {ent, lse};
+}
+
+torch::Tensor entropy_bwd(torch::Tensor logits, torch::Tensor lse,
+                          torch::Tensor ent, torch::Tensor gout) {
+  check_2d(logits, "logits");
+  int64_t N = logits.size(0), C = logits.size(1);
+  auto d = torch::empty_like(logits);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(entropy_bwd_kernel, dim3((unsigned)N), dim3(256), 0,
+                     stream.stream(), logits.data_ptr<float>(),
+                     lse.data_ptr<float>(), ent.data_ptr<float>(),
+                     gout.data_ptr<float>(), d.data_ptr<float>(),
+                     (int)N, (int)C);
+  return d;
+}
+
+std::vector<torch::Tensor> kl_fwd(torch::Tensor t_logits, torch::Tensor s_logits) {
+  check_2d(t_logits, "t_logits");
+  check_2d(s_logits, "s_logits");
+  int64_t N = t_logits.size(0), C = t_logits.size(1);
+  auto kl = torch::empty({N}, t_logits.options());
+  auto t_lse = torch::empty({N}, t_logits.options());
+  auto s_lse = torch::empty({N}, t_logits.options());
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(kl_fwd_kernel, dim3((unsigned)N), dim3(256), 0,
+                     stream.stream(), t_logits.data_ptr<float>(),
+                     s_logits.data_ptr<float>(), kl.data_ptr<float>(),
+                     t_lse.data_ptr<float>(), s_lse.data_ptr<float>(),
+                     (int)N, (int)C);
+  return {kl, t_lse, s_lse};
+}
+
+torch::Tensor kl_bwd(torch::Tensor t_logits, torch::Tensor s_logits,
+                     torch::Tensor t_lse, torch::Tensor s_lse,
+                     torch::Tensor gout) {
+  check_2d(s_logits, "s_logits");
+  int64_t N = s_logits.size(0), C = s_logits.size(1);
+  auto d = torch::empty_like(s_logits);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(kl_bwd_kernel, dim3((unsigned)N), dim3(256), 0,
+                     stream.stream(), t_logits.data_ptr<float>(),
+                     s_logits.data_ptr<float>(), t_lse.data_ptr<float>(),
+                     s_lse.data_ptr<float>(), gout.data_ptr<float>(),
+                     d.data_ptr<float>(), (int)N, (int)C);
+  return d;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("entropy_fwd", &entropy_fwd, "fused rowwise entropy forward");
+  m.def("entropy_bwd", &entropy_bwd, "fused rowwise entropy backward");
+  m.def("kl_fwd", &kl_fwd, "fused rowwise KL(teacher||student) forward");
+  m.def("kl_bwd", &kl_bwd, "fused rowwise KL backward (student grads)");
   m.def("masked_ce_fwd", &masked_ce_fwd,
         "fused masked CE forward (per-row loss + lse)");
   m.def("masked_ce_bwd", &masked_ce_bwd, "fused masked CE backward");

@@ -30,7 +30,9 @@ __device__ inline float block_reduce_max(float v, float* lds) {
     lds[0] = v;
   }
   __syncthreads();
-  return lds[0];
+  v = lds[0];
+  __syncthreads();    // lds is reused by the next reduction
+  return v;
 }
 
 __device__ inline float block_reduce_sum(float v, float* lds) {
@@ -44,7 +46,9 @@ __device__ inline float block_reduce_sum(float v, float* lds) {
     lds[0] = v;
   }
   __syncthreads();
-  return lds[0];
+  v = lds[0];
+  __syncthreads();    // lds is reused by the next reduction
+  return v;
 }
 
 // loss_i = (logsumexp_i - logits[i, label_i]) * mask_i ; lse saved for bwd.

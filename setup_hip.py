@@ -26,6 +26,7 @@ SOURCES = [
     os.path.join(HIP_DIR, 'entity_embed.hip'),
     os.path.join(HIP_DIR, 'su_sample.hip'),
     os.path.join(HIP_DIR, 'ce_loss.hip'),
+    os.path.join(HIP_DIR, 'rl_rowwise.hip'),
     os.path.join(HIP_DIR, 'bindings.cpp'),
 ]
 

@@ -307,3 +307,29 @@ def test_fused_masked_ce_matches_eager():
                                    rtol=1e-4, atol=1e-6)
     finally:
         os.environ.pop('DISTAR_AMD_FUSED_CE', None)
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(os.environ.get('DISTAR_AMD_EXPERIMENTAL') != '1',
+                    reason='experimental kernel, round-2 validation pending')
+def test_fused_rowwise_entropy_kl_match_eager():
+    """K13 fused entropy/KL vs eager fp32 (forward + backward)."""
+    from distar_amd.ops.rl_rowwise import rowwise_entropy, rowwise_kl
+    torch.manual_seed(0)
+    N, C = 64, 24320
+    t = torch.randn(N, C, device='cuda') * 2
+    s_ref = (torch.randn(N, C, device='cuda') * 2).requires_grad_(True)
+    ent_ref = rowwise_entropy(s_ref)            # eager (env off)
+    kl_ref = rowwise_kl(t, s_ref)
+    (ent_ref.sum() + kl_ref.sum()).backward()
+    os.environ['DISTAR_AMD_FUSED_RL_ROWWISE'] = '1'
+    try:
+        s = s_ref.detach().clone().requires_grad_(True)
+        ent = rowwise_entropy(s)
+        kl = rowwise_kl(t, s)
+        torch.testing.assert_close(ent, ent_ref, rtol=1e-5, atol=1e-5)
+        torch.testing.assert_close(kl, kl_ref, rtol=1e-5, atol=1e-5)
+        (ent.sum() + kl.sum()).backward()
+        torch.testing.assert_close(s.grad, s_ref.grad, rtol=1e-4, atol=1e-6)
+    finally:
+        os.environ.pop('DISTAR_AMD_FUSED_RL_ROWWISE', None)
