@@ -80,31 +80,34 @@ class RLDataLoader:
                                  sleep_time=0.5, size=self.buffer_size,
                                  worker_num=self.worker_num)
         data = data + data[:self.batch_size // 2 + 1]
+        # collate needs uniform T, but episodes shorter than traj_len emit
+        # shorter trajectories: bucket by length and emit whichever bucket
+        # fills first — nothing is discarded (a capped buffer evicts oldest)
+        max_buffered = max(4 * self.buffer_size, 4 * self.batch_size)
         while not self._stop:
-            while len(data) < self.batch_size and not self._stop:
+            while not any(len(v) >= self.batch_size
+                          for v in self._length_buckets(data).values()) \
+                    and not self._stop:
                 new = self.adapter.pull(self.player_id + 'traj', fs_type='nppickle',
                                         sleep_time=0.2,
-                                        size=self.buffer_size - len(data),
+                                        size=max(self.batch_size, 2),
                                         worker_num=self.worker_num, timeout=5)
                 data = new + data + new          # double re-injection
+                if len(data) > max_buffered:
+                    data = data[:max_buffered]
             if self._stop:
                 return
-            # collate needs uniform T: batch same-length trajectories (a
-            # trajectory from an episode shorter than traj_len is shorter)
-            from collections import Counter
-            counts = Counter(len(d) for d in data)
-            want, n_want = counts.most_common(1)[0]
-            if n_want < self.batch_size:
-                # drop one minority-length trajectory and pull more
-                for i, d in enumerate(data):
-                    if len(d) != want:
-                        data.pop(i)
-                        break
-                continue
-            picked, rest = [], []
+            buckets = self._length_buckets(data)
+            want = max((k for k, v in buckets.items()
+                        if len(v) >= self.batch_size),
+                       key=lambda k: len(buckets[k]))
+            picked, rest, taken = [], [], 0
             for d in data:
-                (picked if len(d) == want and len(picked) < self.batch_size
-                 else rest).append(d)
+                if len(d) == want and taken < self.batch_size:
+                    picked.append(d)
+                    taken += 1
+                else:
+                    rest.append(d)
             try:
                 batch = self.collate_fn(picked)
             except Exception as e:  # noqa: BLE001 - skip malformed trajectories
@@ -121,6 +124,13 @@ class RLDataLoader:
                     continue
             data = rest
             random.shuffle(data)
+
+    @staticmethod
+    def _length_buckets(data):
+        buckets = {}
+        for d in data:
+            buckets.setdefault(len(d), []).append(d)
+        return buckets
 
     def __iter__(self):
         return self
