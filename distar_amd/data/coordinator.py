@@ -37,10 +37,15 @@ class Coordinator:
         return {'ok': True}
 
     def _request_datum(self, body):
+        # LIFO: hand out the NEWEST datum.  Producers evict their OLDEST
+        # payloads when a bounded queue overflows, so oldest-first metadata
+        # lets a lagging consumer chase only already-evicted ids forever
+        # (head-of-line livelock); newest-first always resolves, and for RL
+        # trajectories fresher data also means lower staleness.
         with self._lock:
             q = self._queues.get(body['token'])
             if q:
-                return q.popleft()
+                return q.pop()
         return {'ip': None}
 
     def _queue_length(self, body):

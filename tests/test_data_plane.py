@@ -156,3 +156,23 @@ def test_model_publication_refreshes_actor_weights(tmp_path, monkeypatch):
                                    learner_model.state_dict()[k])
     finally:
         coord.close()
+
+
+@pytest.mark.timeout(120)
+def test_slow_consumer_still_gets_fresh_data():
+    """A consumer far behind a bounded producer must still make progress:
+    newest-first metadata hand-out means the fetched payload is always
+    live (regression test for the oldest-first eviction livelock)."""
+    coord = Coordinator().run()
+    try:
+        producer = Adapter(coordinator_port=coord.port, maxlen=4)
+        consumer = Adapter(coordinator_port=coord.port)
+        # producer races 50 pushes ahead of a 4-deep payload buffer
+        for i in range(50):
+            producer.push({'i': torch.tensor([i])}, token='fresh')
+        out = consumer.pull('fresh', size=2, sleep_time=0.05, timeout=60)
+        vals = [int(d['i'][0]) for d in out]
+        assert len(vals) == 2
+        assert all(v >= 46 for v in vals), vals    # only the live tail
+    finally:
+        coord.close()
