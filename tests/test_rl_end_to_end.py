@@ -162,6 +162,20 @@ def test_three_player_league_loop(tmp_path, monkeypatch):
         deadline = time.time() + 1200       # generous: loaded CI hosts
         while len(done) < n and time.time() < deadline:
             time.sleep(1)
+        if len(done) < n:
+            # diagnostics: where is the stuck learner, what do the queues say
+            import faulthandler
+            import sys
+            for pid, ln in zip(ids, learners):
+                if pid not in done:
+                    q = ln._dataloader._batch_queue.qsize() \
+                        if hasattr(ln._dataloader, '_batch_queue') else '?'
+                    print(f'[diag] {pid}: batch_queue={q} '
+                          f'adapter_len={ln._dataloader.adapter.length(pid + "traj")}',
+                          flush=True)
+            print(f'[diag] coordinator queues: '
+                  f'{ {t: len(qq) for t, qq in coord._queues.items()} }', flush=True)
+            faulthandler.dump_traceback(file=sys.stdout)
         assert done == {pid: 1 for pid in ids}, f'learners finished: {done}'
         # league saw game results from the actor fleet
         deadline = time.time() + 180
