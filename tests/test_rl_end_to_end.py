@@ -96,10 +96,14 @@ def test_three_player_league_loop(tmp_path, monkeypatch):
         'common': {'experiment_name': 'e2e3'},
         'league': {
             'save_resume_freq': 10000,
-            'branch_probs': {           # force data-producing branches
-                'MainPlayer': {'sp': 1.0},
-                'ExploiterPlayer': {'pfsp': 1.0},
-                'MainExploiterPlayer': {'vs_main': 1.0}},
+            # force data-producing branches.  NOTE: deep_merge keeps sibling
+            # keys from the defaults, so 'eval' must be zeroed explicitly —
+            # an actor that draws an eval job sends no training data and
+            # holds the job for actor_ask_for_job_interval.
+            'branch_probs': {
+                'MainPlayer': {'sp': 1.0, 'pfsp': 0.0, 'eval': 0.0},
+                'ExploiterPlayer': {'pfsp': 1.0, 'eval': 0.0},
+                'MainExploiterPlayer': {'vs_main': 1.0, 'eval': 0.0}},
             'active_players': {
                 'player_id': ids, 'checkpoint_path': ['none'] * n,
                 'pipeline': ['default'] * n, 'frac_id': [1] * n,
