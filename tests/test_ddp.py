@@ -24,9 +24,12 @@ def _worker(rank, world, init_file, q):
     y = dm.module[2](torch.relu(dm.module[0](x))).sum()
     y.backward()
     dm.sync_gradients()
-    grads = {n: p.grad.clone() for n, p in model.named_parameters()
+    # send plain numpy copies: mp.Queue ships torch tensors via fd-passed
+    # shared memory, which breaks if the child exits before the parent
+    # materializes them (ConnectionResetError in recvfds)
+    grads = {n: p.grad.clone().numpy() for n, p in model.named_parameters()
              if p.grad is not None}
-    q.put((rank, {n: g for n, g in grads.items()}))
+    q.put((rank, grads))
     dist.barrier()
     dist.destroy_process_group()
 
@@ -63,6 +66,8 @@ def test_distmodule_grads_are_averaged():
                 raise
             _time.sleep(5)
     # both ranks end with identical (averaged) gradients
+    results = {r: {n: torch.from_numpy(g) for n, g in d.items()}
+               for r, d in results.items()}
     assert set(results[0].keys()) == set(results[1].keys())
     for n in results[0]:
         torch.testing.assert_close(results[0][n], results[1][n])
