@@ -13,6 +13,8 @@ import itertools
 import os
 import queue
 import random
+
+import numpy as np
 import threading
 import time
 from collections import defaultdict
@@ -74,6 +76,12 @@ class League:
         self._scalars = ScalarLogger(os.path.join(exp_dir, 'log'), name='league')
         self._stat_decay = self.cfg.stat_decay
         self._stat_warm_up_size = self.cfg.stat_warm_up_size
+        seed = self.cfg.get('seed')
+        if seed is not None:
+            # league job dispatch (branch/opponent/map draws) uses the global
+            # RNG; seeding here makes league schedules reproducible
+            random.seed(seed)
+            np.random.seed(seed % (2 ** 32))
         self._payoff_min_win_rate_games = self.cfg.payoff_min_win_rate_games
         self.elo = ELORating()
         self.trueskill = TrueSkill()
