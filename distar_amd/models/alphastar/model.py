@@ -38,9 +38,9 @@ def detach_grad(data):
 
 
 class Model(nn.Module):
-    def __init__(self, cfg={}, use_value_network=False, temperature=None):
+    def __init__(self, cfg=None, use_value_network=False, temperature=None):
         super().__init__()
-        self.whole_cfg = deep_merge_dicts(alphastar_model_default_config, cfg)
+        self.whole_cfg = deep_merge_dicts(alphastar_model_default_config, cfg or {})
         if temperature is not None:
             self.whole_cfg.model.temperature = temperature
         self.cfg = self.whole_cfg.model
