@@ -10,7 +10,6 @@ league-ordered checkpoint reset + broadcast).
 import os
 import time
 
-import torch
 
 from ..data.adapter import Adapter
 from ..parallel.dist import broadcast, get_rank, get_world_size, is_initialized

@@ -8,7 +8,6 @@ synthesizing observations.
 """
 import itertools
 
-import torch
 
 from ..lib.fake_data import fake_rl_learner_data, fake_sl_batch
 from ..utils.data import to_device

@@ -10,7 +10,6 @@ outcome.  Used by the actor's synthetic-rollout mode and the CPU tests.
 """
 import random
 
-import torch
 
 from ..lib.consts import fake_step_data
 from .map_info import get_map_size

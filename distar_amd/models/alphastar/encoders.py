@@ -22,7 +22,6 @@ from torch import Tensor
 from ..nn.blocks import (fc_block, conv2d_block, build_activation, ResBlock,
                          sequence_mask, one_hot_embedding, binary_embedding)
 from ..nn.transformer import Transformer, AttentionPool
-from ..nn.lnlstm import script_lnlstm
 from ...ops.scatter import scatter_connection, spatial_effect_plane
 from ...lib.consts import MAX_ENTITY_NUM
 
