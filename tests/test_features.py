@@ -205,3 +205,35 @@ def test_filter_actions_dedups_spam():
     assert kept == ['a2', 'b0', 'c0', 'd0']
     # the surviving spam entry carries the LAST loop (freshest delay timing)
     assert out[0][0] == 2
+
+
+def test_reverse_raw_action_families():
+    """Target-unit, quick (no-target), camera-move and special remaps
+    (unload->3664, cancel-slot->3671, frivolous dropped) decode per the
+    reference (features.py:853-952)."""
+    from distar_amd.lib.actions import ACTIONS
+    feat, raw_ob, units = make_features()
+    tags = [u.tag for u in units]
+
+    # raw_cmd_unit: attack a specific unit
+    act = D.raw_action(ability_id=3674, unit_tags=[100], target_unit_tag=102)
+    action_ret, mask, su_num, lsu, ltu, invalid = feat.reverse_raw_action(act, tags)
+    assert not invalid
+    assert int(action_ret['target_unit']) == tags.index(102)
+    assert ACTIONS[int(action_ret['action_type'])]['func_id'] == 3   # Attack_unit
+
+    # raw_cmd (quick): stop
+    act = D.raw_action(ability_id=3665, unit_tags=[100])
+    action_ret, *_, invalid = feat.reverse_raw_action(act, tags)
+    assert not invalid and action_ret['action_type'] is not None
+    assert ACTIONS[int(action_ret['action_type'])]['name'].startswith('Stop')
+
+    # frivolous abilities (6/7) are dropped -> masked default + invalid
+    act = D.raw_action(ability_id=6, unit_tags=[100])
+    action_ret, mask, _, _, _, invalid = feat.reverse_raw_action(act, tags)
+    assert invalid and not bool(mask['action_type'])
+
+    # target-unit tag missing from the obs -> invalid flagged
+    act = D.raw_action(ability_id=3674, unit_tags=[100], target_unit_tag=999999)
+    *_, invalid = feat.reverse_raw_action(act, tags)
+    assert invalid
