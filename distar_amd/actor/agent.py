@@ -502,7 +502,25 @@ class Agent:
         bo_reward = torch.zeros((), dtype=torch.float)
         cum_reward = torch.zeros((), dtype=torch.float)
         battle_reward = torch.zeros((), dtype=torch.float)
-        if next_obs is None or not self._exceed_flag:
+        if next_obs is None:
+            return bo_reward, cum_reward, battle_reward
+        # battle pseudo-reward: own score delta minus opponent score delta
+        # (reference agent.py:623-626), computed even past the Z loop bound
+        raw = next_obs.get('raw_obs') if isinstance(next_obs, dict) else None
+        if raw is not None:
+            from ..lib.features import compute_battle_score
+            try:
+                battle_score = compute_battle_score(raw)
+                opp = next_obs.get('opponent_obs')
+                opp_score = compute_battle_score(opp) if opp is not None \
+                    else self._game_info['opponent_battle_score']
+                battle_reward = torch.tensor(
+                    (battle_score - self._game_info['battle_score']) -
+                    (opp_score - self._game_info['opponent_battle_score']),
+                    dtype=torch.float) / self._cfg['battle_norm']
+            except AttributeError:      # mock obs without score protos
+                pass
+        if not self._exceed_flag:
             return bo_reward, cum_reward, battle_reward
         if action_type in BEGINNING_ORDER_ACTIONS and \
                 (next_obs.get('action_result') or [1])[0] == 1:

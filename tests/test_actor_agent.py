@@ -230,3 +230,26 @@ def test_episode_end_window_is_full_length():
     # every emitted window has exactly traj_len steps + bootstrap frame,
     # including the episode-end one (slides back over earlier steps)
     assert all(l == 5 for l in lengths), lengths
+
+
+def test_battle_pseudo_reward_deltas():
+    """Battle reward = delta(own score) - delta(opponent score), /battle_norm
+    (reference agent.py:623-626)."""
+    import dummy_obs as D
+    torch.manual_seed(0)
+    cfg = Config({'common': {'type': 'train'}, 'actor': {'traj_len': 4},
+                  'env': {'player_num': 2}, 'agent': {}})
+    agent = Agent(cfg, env_id=0)
+    agent.player_id = 'MP0'
+    agent.reset()
+    agent._game_info['battle_score'] = 100.
+    agent._game_info['opponent_battle_score'] = 50.
+    nxt = {'raw_obs': D.raw_observation([D.unit(tag=1)]),
+           'opponent_obs': D.raw_observation([D.unit(tag=2)], player_id=2),
+           'action_result': [1]}
+    from distar_amd.lib.features import compute_battle_score
+    own = compute_battle_score(nxt['raw_obs'])
+    opp = compute_battle_score(nxt['opponent_obs'])
+    _, _, battle = agent._update_fake_reward(0, torch.tensor(0), nxt)
+    expected = ((own - 100.) - (opp - 50.)) / 30.
+    assert abs(float(battle) - expected) < 1e-6
