@@ -529,6 +529,10 @@ class Agent:
                 if self._bo_zergling_count > 8:
                     return bo_reward, cum_reward, battle_reward
             order_index = BEGINNING_ORDER_ACTIONS.index(action_type)
+            if order_index == 39 and 39 not in self._target_building_order:
+                # spine crawler outside the target style: ignored (reference
+                # agent.py:637-638)
+                return bo_reward, cum_reward, battle_reward
             if len(self._behaviour_building_order) < len(self._target_building_order):
                 self._behaviour_building_order.append(order_index)
                 self._behaviour_bo_location.append(
