@@ -26,7 +26,9 @@ import torch
 
 from ..lib.actions import (ACTIONS, BEGINNING_ORDER_ACTIONS,
                            CUMULATIVE_STAT_ACTIONS,
-                           NUM_CUMULATIVE_STAT_ACTIONS)
+                           NUM_CUMULATIVE_STAT_ACTIONS, QUEUE_ACTIONS,
+                           UNIT_ABILITY_TO_ACTION, UNIT_TO_CUM,
+                           UPGRADE_TO_CUM)
 from ..lib.consts import (BEGINNING_ORDER_LENGTH, MAX_DELAY, SPATIAL_SIZE)
 from ..lib.stat import Stat, cum_dict
 from ..models.alphastar.model import Model
@@ -106,6 +108,19 @@ class Agent:
         self._total_bo_reward = torch.zeros((), dtype=torch.float)
         self._total_cum_reward = torch.zeros((), dtype=torch.float)
         self._game_info = {'battle_score': 0, 'opponent_battle_score': 0}
+        # own first-base raw position for the observation-mode cum skip
+        # (reference agent.py:176-183)
+        self._born_location_xy = None
+        raw = (obs or {}).get('raw_obs') if isinstance(obs, dict) else None
+        if raw is not None:
+            try:
+                locs = [[u.pos.x, u.pos.y]
+                        for u in raw.observation.raw_data.units
+                        if u.unit_type in (59, 18, 86)]
+                if locs:
+                    self._born_location_xy = locs[0]
+            except AttributeError:
+                pass
         self._load_z(map_name, race, opponent_race, obs)
         self._old_bo_reward = -levenshtein_distance(
             torch.as_tensor(self._behaviour_building_order, dtype=torch.long),
@@ -538,22 +553,72 @@ class Agent:
                 self._behaviour_bo_location.append(
                     int(location) if ACTIONS[action_type]['target_location'] else 0)
                 if self.use_bo_reward:
+                    if self._cfg['clip_bo']:    # compare against the target
+                        tz = self._target_building_order[:len(self._behaviour_building_order)]
+                        tz_lo = self._target_bo_location[:len(self._behaviour_building_order)]
+                    else:
+                        tz, tz_lo = self._target_building_order, self._target_bo_location
                     new_bo_dist = -levenshtein_distance(
                         torch.as_tensor(self._behaviour_building_order, dtype=torch.int),
-                        self._target_building_order.int(),
+                        tz.int(),
                         torch.as_tensor(self._behaviour_bo_location, dtype=torch.int),
-                        self._target_bo_location.int(),
+                        tz_lo.int(),
                         partial(l2_distance, spatial_x=SPATIAL_SIZE[1])) / self._bo_norm
                     bo_reward = new_bo_dist - self._old_bo_reward
                     self._old_bo_reward = new_bo_dist
         cum_flag = False
-        if self._cfg['cum_type'] == 'action':
+        cum_type = self._cfg['cum_type']
+        if cum_type == 'observation':
+            # scan completed own units/upgrades (reference agent.py:663-677);
+            # the first base at the born location is not a built structure
+            cum_flag = True
+            try:
+                for u in next_obs['raw_obs'].observation.raw_data.units:
+                    if u.alliance == 1 and u.unit_type in (59, 18, 86) and \
+                            self._born_location_xy is not None and \
+                            u.pos.x == self._born_location_xy[0] and \
+                            u.pos.y == self._born_location_xy[1]:
+                        continue
+                    if u.alliance == 1 and u.build_progress == 1 and \
+                            UNIT_TO_CUM[u.unit_type] != -1:
+                        self._behaviour_cumulative_stat[UNIT_TO_CUM[u.unit_type]] = 1
+                for uid in next_obs['raw_obs'].observation.raw_data.player.upgrade_ids:
+                    if UPGRADE_TO_CUM[uid] != -1:
+                        self._behaviour_cumulative_stat[UPGRADE_TO_CUM[uid]] = 1
+            except (AttributeError, KeyError, TypeError):
+                cum_flag = False        # mock obs without raw protos
+        elif cum_type == 'action':
+            action_name = ACTIONS[action_type]['name']
+            if action_name in ('Cancel_quick', 'Cancel_Last_quick') and \
+                    self._output is not None:
+                # cancelling an in-progress train/build refunds its cum slot
+                # (reference agent.py:682-696)
+                try:
+                    unit_index = int(self._output['action_info']['selected_units'][0])
+                    order_len = int(self._observation['entity_info']['order_length'][unit_index])
+                    action_index = None
+                    if order_len == 1:
+                        action_index = UNIT_ABILITY_TO_ACTION.get(
+                            int(self._observation['entity_info']['order_id_0'][unit_index]))
+                    elif order_len > 1:
+                        qi = int(self._observation['entity_info']
+                                 [f'order_id_{order_len - 1}'][unit_index]) - 1
+                        if 0 <= qi < len(QUEUE_ACTIONS):
+                            action_index = QUEUE_ACTIONS[qi]
+                    if action_index in CUMULATIVE_STAT_ACTIONS:
+                        cum_flag = True
+                        ci = CUMULATIVE_STAT_ACTIONS.index(action_index)
+                        self._behaviour_cumulative_stat[ci] = \
+                            max(0, self._behaviour_cumulative_stat[ci] - 1)
+                except (KeyError, IndexError, TypeError):
+                    pass
             if action_type in CUMULATIVE_STAT_ACTIONS:
                 cum_flag = True
                 ci = CUMULATIVE_STAT_ACTIONS.index(action_type)
                 self._behaviour_cumulative_stat[ci] += 1
         if self.use_cum_reward and cum_flag and \
-                (next_obs.get('action_result') or [1])[0] == 1:
+                (cum_type == 'observation' or
+                 (next_obs.get('action_result') or [1])[0] == 1):
             new_cum = -hamming_distance(
                 torch.as_tensor(self._behaviour_cumulative_stat, dtype=torch.bool),
                 self._target_cumulative_stat.bool()) / self._cum_norm

@@ -253,3 +253,55 @@ def test_battle_pseudo_reward_deltas():
     _, _, battle = agent._update_fake_reward(0, torch.tensor(0), nxt)
     expected = ((own - 100.) - (opp - 50.)) / 30.
     assert abs(float(battle) - expected) < 1e-6
+
+
+def test_cum_reward_observation_mode():
+    """cum_type='observation': completed own units/upgrades scan into the
+    behaviour cumulative stat (reference agent.py:663-677)."""
+    import dummy_obs as D
+    from distar_amd.lib.actions import UNIT_TO_CUM
+    torch.manual_seed(0)
+    cfg = Config({'common': {'type': 'train'},
+                  'actor': {'traj_len': 4},
+                  'agent': {'cum_type': 'observation'},
+                  'env': {'player_num': 2}})
+    agent = Agent(cfg, env_id=0)
+    agent.player_id = 'MP0'
+    agent.reset()
+    agent.use_cum_reward = True
+    # pick a unit type with a cumulative slot
+    ut = next(u for u in list(UNIT_TO_CUM.keys()) if UNIT_TO_CUM[u] != -1)
+    nxt = {'raw_obs': D.raw_observation(
+        [D.unit(tag=1, unit_type=ut, alliance=1)]),
+        'action_result': [1]}
+    _, cum_reward, _ = agent._update_fake_reward(0, torch.tensor(0), nxt)
+    assert agent._behaviour_cumulative_stat[UNIT_TO_CUM[ut]] == 1
+
+
+def test_cum_reward_cancel_refund():
+    """Cancel_quick refunds the cancelled unit's cumulative slot
+    (reference agent.py:682-696)."""
+    from distar_amd.lib.actions import (ACTIONS, CUMULATIVE_STAT_ACTIONS,
+                                        UNIT_ABILITY_TO_ACTION)
+    torch.manual_seed(0)
+    cfg = Config({'common': {'type': 'train'}, 'actor': {'traj_len': 4},
+                  'env': {'player_num': 2}, 'agent': {}})
+    agent = Agent(cfg, env_id=0)
+    agent.player_id = 'MP0'
+    agent.reset()
+    agent.use_cum_reward = True
+    cancel_at = next(i for i, a in enumerate(ACTIONS)
+                     if a['name'] == 'Cancel_quick')
+    # choose an order ability that maps to a cumulative action
+    ability_idx, mapped = next(
+        (k, v) for k, v in UNIT_ABILITY_TO_ACTION.items()
+        if v in CUMULATIVE_STAT_ACTIONS)
+    ci = CUMULATIVE_STAT_ACTIONS.index(mapped)
+    agent._behaviour_cumulative_stat[ci] = 1
+    agent._output = {'action_info': {'selected_units': torch.tensor([0])}}
+    agent._observation = {'entity_info': {
+        'order_length': torch.tensor([1]),
+        'order_id_0': torch.tensor([ability_idx])}}
+    nxt = {'action_result': [1]}
+    agent._update_fake_reward(cancel_at, torch.tensor(0), nxt)
+    assert agent._behaviour_cumulative_stat[ci] == 0
