@@ -390,6 +390,11 @@ class Agent:
             action_info['target_unit_tag'] if ACTIONS[at]['target_unit'] else None
         loc = int(output['action_info']['target_location'])
         x, y = loc % SPATIAL_SIZE[1], loc // SPATIAL_SIZE[1]
+        # model space -> game world coordinates: transform_obs maps
+        # y_world -> map_size.y - y, so invert on the way out (reference
+        # agent.py:389-391); the mock env has no real coordinate frame
+        if hasattr(self, '_feature'):
+            y = max(self._feature.map_size.y - y, 0)
         action_info['location'] = (x, y)
         return [action_info]
 

@@ -237,3 +237,30 @@ def test_reverse_raw_action_families():
     act = D.raw_action(ability_id=3674, unit_tags=[100], target_unit_tag=999999)
     *_, invalid = feat.reverse_raw_action(act, tags)
     assert invalid
+
+
+def test_agent_emits_world_coordinates():
+    """_post_process inverts the y axis back to game-world coordinates
+    (reference agent.py:389-391): transform_obs in / action out roundtrip."""
+    import dummy_obs as D
+    from distar_amd.actor.agent import Agent
+    from distar_amd.lib.actions import ACTIONS
+    torch.manual_seed(0)
+    cfg = Config({'common': {'type': 'train'}, 'actor': {'traj_len': 4},
+                  'env': {'player_num': 2}, 'agent': {}})
+    agent = Agent(cfg, env_id=0)
+    agent.player_id = 'MP0'
+    gi = D.game_info()
+    # exactly one own hatchery (type 86) = the born base; drones otherwise
+    units = [D.unit(tag=100, unit_type=86)] + \
+        [D.unit(tag=101 + i, unit_type=104) for i in range(2)]
+    raw_ob = D.raw_observation(units)
+    obs = {'game_info_proto': gi, 'raw_obs': raw_ob, 'game_info': gi}
+    agent.reset(obs=obs)
+    action = agent.step(obs)[0]
+    x, y_world = action['location']
+    # the model's flat location index for this action
+    loc = int(agent._output['action_info']['target_location'])
+    y_model = loc // 160
+    assert y_world == max(agent._feature.map_size.y - y_model, 0)
+    assert x == loc % 160
