@@ -340,7 +340,8 @@ class Agent:
 
     @sw.decorate('agent_step')
     def step(self, observation):
-        if 'eval' in self._job_type and self._iter_count > 0:
+        if 'eval' in self._job_type and self._iter_count > 0 and \
+                not self._whole_cfg.get('env', {}).get('realtime', False):
             self._update_fake_reward(int(self._last_action_type),
                                      self._last_location, observation)
         model_input = self._pre_process(observation)
