@@ -53,6 +53,10 @@ class RLLearner(BaseLearner):
         self._update_config_flag = False
         self._config_update = None
         self._debug_server = None
+        # optional per-parameter grad/weight norm streams (reference
+        # rl_learner.py:35-47,118-130); JSONL keyed grad/<param> etc.
+        self._save_grad = self._whole_cfg.learner.get('save_grad', False)
+        self._save_grad_freq = self._whole_cfg.learner.get('save_log_freq', 100)
 
     def _setup_loss(self):
         self._loss = ReinforcementLoss(self._whole_cfg.learner, self._player_id)
@@ -163,7 +167,26 @@ class RLLearner(BaseLearner):
             loss.backward()
             if self._use_distributed:
                 self._model.sync_gradients()
+            log_params = self._save_grad and self._rank == 0 and \
+                self.last_iter.val % self._save_grad_freq == 0
+            if log_params:
+                with torch.no_grad():
+                    for k, p in self._model.named_parameters():
+                        if p.grad is not None:
+                            self._scalar_logger.add_scalar(
+                                f'grad/{k}', p.grad.norm().item(),
+                                global_step=self.last_iter.val)
+                            self._scalar_logger.add_scalar(
+                                f'param/{k}', p.data.norm().item(),
+                                global_step=self.last_iter.val)
             gradient = self._grad_clip.apply(self._model.parameters())
+            if log_params:
+                with torch.no_grad():
+                    for k, p in self._model.named_parameters():
+                        if p.grad is not None:
+                            self._scalar_logger.add_scalar(
+                                f'clip_grad/{k}', p.grad.norm().item(),
+                                global_step=self.last_iter.val)
             self._optimizer.step()
         self._log_buffer['gradient'] = gradient
         self._log_buffer['backward_time'] = self._timer.value

@@ -209,3 +209,31 @@ def test_config_backup_written(tmp_path, monkeypatch):
     assert os.path.isfile(path)
     loaded = yaml.safe_load(open(path))
     assert loaded['learner']['data']['batch_size'] == 2
+
+
+@pytest.mark.timeout(600)
+def test_rl_learner_param_norm_streams(tmp_path, monkeypatch):
+    """save_grad logs per-parameter grad/weight/clipped norms every
+    save_log_freq iterations (reference rl_learner.py:118-130)."""
+    monkeypatch.chdir(tmp_path)
+    import json
+    from distar_amd.learner.rl_learner import RLLearner
+    torch.manual_seed(0)
+    cfg = Config({'learner': {'player_id': 'MP0', 'job_type': 'fake',
+                              'use_cuda': False, 'use_amp': False,
+                              'save_grad': True, 'save_log_freq': 1,
+                              'data': {'batch_size': 2, 'trajectory_length': 3},
+                              'hook': {'after_iter': {
+                                  'log_show': {'ext_args': {'freq': 1000}}}}},
+                  'common': {'experiment_name': 'test_grad_stream',
+                             'type': 'train'},
+                  'model': {'enable_baselines': ['winloss']}})
+    learner = RLLearner(cfg)
+    learner.run(max_iterations=1)
+    learner._scalar_logger.flush()
+    files = glob.glob(os.path.join(learner._exp_dir, 'log', '*.jsonl'))
+    assert files
+    keys = {json.loads(l)['key'] for f in files for l in open(f)}
+    assert any(k.startswith('grad/') for k in keys)
+    assert any(k.startswith('clip_grad/') for k in keys)
+    assert any(k.startswith('param/') for k in keys)
