@@ -128,9 +128,14 @@ class BaseLearner:
             self.model.parameters(), lr=cfg.learning_rate,
             weight_decay=cfg.weight_decay, fused=self._use_cuda or None)
         milestones = list(cfg.lr_decay_milestones)
+        if not milestones and cfg.get('lr_decay_interval'):
+            # reference-config compatibility (base_learner.py:171-176): decay
+            # by lr_decay every lr_decay_interval iterations, 40 steps out
+            interval = int(cfg.lr_decay_interval)
+            milestones = list(range(interval, interval * 40, interval))
         decay = torch.optim.lr_scheduler.MultiStepLR(
             self._optimizer, milestones=milestones, gamma=cfg.lr_decay)
-        warmup_iters = int(cfg.warmup_iters)
+        warmup_iters = int(cfg.get('warm_up_steps', cfg.warmup_iters))
         if warmup_iters > 0:
             warmup = torch.optim.lr_scheduler.LinearLR(
                 self._optimizer, start_factor=1e-3, total_iters=warmup_iters)
