@@ -176,3 +176,28 @@ def test_slow_consumer_still_gets_fresh_data():
         assert all(v >= 46 for v in vals), vals    # only the live tail
     finally:
         coord.close()
+
+
+@pytest.mark.timeout(120)
+def test_coordinator_worker_offload():
+    """start_worker spawns a per-token broker with the identical API
+    (reference coordinator.py:20-59,156-165); adapters pointed at the worker
+    port roundtrip normally."""
+    from distar_amd.utils.http import post_json
+    coord = Coordinator().run()
+    worker_srv = None
+    try:
+        resp = post_json(f'http://127.0.0.1:{coord.port}/start_worker',
+                         {'token': 'hot'})
+        assert resp['port'] and resp['port'] != coord.port
+        # same worker is reused per token
+        again = post_json(f'http://127.0.0.1:{coord.port}/start_worker',
+                          {'token': 'hot'})
+        assert again['port'] == resp['port']
+        producer = Adapter(coordinator_port=resp['port'])
+        consumer = Adapter(coordinator_port=resp['port'])
+        producer.push({'v': torch.tensor([7.0])}, token='hot')
+        out = consumer.pull('hot', size=1, timeout=30)
+        assert float(out[0]['v'][0]) == 7.0
+    finally:
+        coord.close()
