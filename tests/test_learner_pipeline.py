@@ -237,3 +237,28 @@ def test_rl_learner_param_norm_streams(tmp_path, monkeypatch):
     assert any(k.startswith('grad/') for k in keys)
     assert any(k.startswith('clip_grad/') for k in keys)
     assert any(k.startswith('param/') for k in keys)
+
+
+def test_sl_hidden_state_lane_reset():
+    """reset_hidden_state zeroes exactly the new-episode lanes and detaches
+    the rest (reference sl_learner.py:31-36)."""
+    from distar_amd.learner.sl_learner import SLLearner
+    torch.manual_seed(0)
+    cfg = Config({'learner': {'job_type': 'fake', 'use_cuda': False,
+                              'data': {'batch_size': 3, 'trajectory_length': 2}},
+                  'common': {'experiment_name': 'test_lane_reset',
+                             'type': 'train'}})
+    learner = SLLearner(cfg)
+    for l in range(learner.num_layers):
+        h = torch.randn(3, learner.hidden_size, requires_grad=True) * 1
+        c = torch.randn(3, learner.hidden_size, requires_grad=True) * 1
+        learner.hidden_state[l] = (h + 0, c + 0)     # non-leaf, grad-tracked
+    before = [tuple(t.clone() for t in hc) for hc in learner.hidden_state]
+    learner.reset_hidden_state(torch.tensor([True, False, True]))
+    for l in range(learner.num_layers):
+        h, c = learner.hidden_state[l]
+        assert not h.requires_grad and not c.requires_grad    # detached
+        assert (h[0] == 0).all() and (h[2] == 0).all()
+        assert (c[0] == 0).all() and (c[2] == 0).all()
+        torch.testing.assert_close(h[1], before[l][0][1])     # lane 1 kept
+        torch.testing.assert_close(c[1], before[l][1][1])
