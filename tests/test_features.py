@@ -264,3 +264,28 @@ def test_agent_emits_world_coordinates():
     y_model = loc // 160
     assert y_world == max(agent._feature.map_size.y - y_model, 0)
     assert x == loc % 160
+
+
+def test_action_table_invariants():
+    """Derived action-table structures match the reference derivations
+    (reference lib/actions.py:355-426); guards the data assets."""
+    from distar_amd.lib.actions import (ACTIONS, NUM_ACTIONS, QUEUE_ACTIONS,
+                                        BEGINNING_ORDER_ACTIONS,
+                                        CUMULATIVE_STAT_ACTIONS,
+                                        SELECTED_UNITS_MASK, UNIT_TO_CUM,
+                                        UPGRADE_TO_CUM)
+    assert NUM_ACTIONS == 327 and len(ACTIONS) == 327
+    assert len(QUEUE_ACTIONS) == 109          # 'Train_'/'Research' actions
+    assert len(BEGINNING_ORDER_ACTIONS) == 174
+    assert len(CUMULATIVE_STAT_ACTIONS) == 167
+    assert int(SELECTED_UNITS_MASK.sum()) == 325
+    # every queue action really is a train/research action
+    assert all('Train_' in ACTIONS[i]['name'] or 'Research' in ACTIONS[i]['name']
+               for i in QUEUE_ACTIONS)
+    # cum LUTs map into valid slots
+    assert all(0 <= v < 167 for v in UNIT_TO_CUM.values() if v != -1)
+    assert all(0 <= v < 167 for v in UPGRADE_TO_CUM.values() if v != -1)
+    # action schema fields present on every row
+    for a in ACTIONS:
+        assert {'name', 'func_id', 'general_ability_id', 'goal', 'queued',
+                'selected_units', 'target_location', 'target_unit'} <= set(a)
