@@ -305,3 +305,31 @@ def test_cum_reward_cancel_refund():
     nxt = {'action_result': [1]}
     agent._update_fake_reward(cancel_at, torch.tensor(0), nxt)
     assert agent._behaviour_cumulative_stat[ci] == 0
+
+
+def test_agent_loads_real_z_assets():
+    """_load_z samples a real strategy-statistics entry from the shipped
+    asset files and arms the targets/flags (reference agent.py:176-243)."""
+    import random
+    torch.manual_seed(0)
+    random.seed(0)
+    cfg = Config({'common': {'type': 'train'},
+                  'actor': {'traj_len': 4},
+                  'agent': {'z_path': '3map.json', 'fake_reward_prob': 1.0},
+                  'env': {'player_num': 2}})
+    agent = Agent(cfg, env_id=0)
+    agent.player_id = 'MP0'
+    agent.reset(map_name='KingsCove', race='zerg', opponent_race='zerg')
+    assert agent._target_building_order.shape[0] >= 1
+    assert agent._target_cumulative_stat.shape[0] == 167
+    assert int(agent._target_cumulative_stat.sum()) > 0     # real Z content
+    assert agent._target_z_loop > 0
+    assert agent.use_bo_reward or agent.use_cum_reward
+    # style files with z_type annotations load too
+    cfg2 = Config({'common': {'type': 'train'}, 'actor': {'traj_len': 4},
+                   'agent': {'z_path': 'mutalisk.json'},
+                   'env': {'player_num': 2}})
+    agent2 = Agent(cfg2, env_id=0)
+    agent2.player_id = 'MP0'
+    agent2.reset(map_name='KingsCove', race='zerg', opponent_race='zerg')
+    assert agent2._target_building_order.shape[0] >= 1
