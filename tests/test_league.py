@@ -294,3 +294,33 @@ def test_league_debug_api_extended(league):
         assert r.json()['ok'] and hp.player_id not in league.historical_players
     finally:
         api.stop()
+
+
+def test_league_resume_restores_payoff_and_ratings(league):
+    """Resume restores deep state: payoff meters, ELO/TrueSkill ratings and
+    snapshot lineage (reference league.py:535-556)."""
+    mp = league.active_players['MP0']
+    hp = mp.snapshot()
+    league.set_hist_player(hp)
+    def stat(w):
+        return {'winrate': w, 'game_steps': 10, 'game_iters': 1,
+                'game_duration': 5}
+    for w in (1.0, 0.0, 1.0):
+        mp.payoff.update('ME0', stat(w))
+    league.elo.update('MP0', 'ME0', 1.0)
+    league.trueskill.update('MP0', 'ME0')
+    path = league.save_resume()
+    cfg = _league_cfg(None)
+    cfg.league.resume_path = path
+    lg2 = League(cfg)
+    try:
+        mp2 = lg2.active_players['MP0']
+        assert hp.player_id in lg2.historical_players
+        assert lg2.historical_players[hp.player_id].parent_id == 'MP0'
+        r1 = mp.payoff.stat_info_record['ME0']['winrate']
+        r2 = mp2.payoff.stat_info_record['ME0']['winrate']
+        assert r2.count == r1.count and abs(r2.val - r1.val) < 1e-9
+        assert lg2.elo.ratings['MP0'] == league.elo.ratings['MP0']
+        assert abs(lg2.trueskill.mu['MP0'] - league.trueskill.mu['MP0']) < 1e-9
+    finally:
+        lg2.close()
