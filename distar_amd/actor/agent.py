@@ -397,7 +397,34 @@ class Agent:
         if hasattr(self, '_feature'):
             y = max(self._feature.map_size.y - y, 0)
         action_info['location'] = (x, y)
+        if 'test' in self._job_type:
+            self._print_action(output['action_info'], (x, y),
+                               output.get('action_logp', {}))
         return [action_info]
+
+    def _print_action(self, action_info, location, logp):
+        """Human-readable action trace for eval/test runs (reference
+        agent.py:398-417)."""
+        at = int(action_info['action_type'])
+        name = ACTIONS[at]['name']
+        def p(head):
+            v = logp.get(head)
+            return f'{float(torch.exp(v)):.2f}' if v is not None else '?'
+        su = ''
+        if ACTIONS[at]['selected_units'] and 'selected_units' in action_info:
+            sel = action_info['selected_units']
+            lp = logp.get('selected_units')
+            for i, u in enumerate(sel[:-1].tolist()):
+                prob = f'({float(torch.exp(lp[i])):.2f})' if lp is not None else ''
+                su += f' {int(self._observation["entity_info"]["unit_type"][u])}{prob}'
+            if lp is not None and len(lp):
+                su += f' end({float(torch.exp(lp[-1])):.2f})'
+        tu = int(action_info['target_unit']) if ACTIONS[at]['target_unit'] else None
+        print(f'{self.player_id}, game_step:{self._game_step}, '
+              f'at:{name}({p("action_type")}), '
+              f'delay:{int(action_info["delay"])}({p("delay")}), su:{su}, '
+              f'tu:{tu}({p("target_unit")}), '
+              f'lo:{location}({p("target_location")})')
 
     # ----------------------------------------------------------- collect_data
     def collect_data(self, next_obs, reward, done, idx):
