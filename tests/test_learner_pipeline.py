@@ -262,3 +262,37 @@ def test_sl_hidden_state_lane_reset():
         assert (c[0] == 0).all() and (c[2] == 0).all()
         torch.testing.assert_close(h[1], before[l][0][1])     # lane 1 kept
         torch.testing.assert_close(c[1], before[l][1][1])
+
+
+@pytest.mark.timeout(600)
+def test_rl_value_pretrain_phase_switch(tmp_path, monkeypatch):
+    """value_pretrain_iters: critic-only updates for N iterations, then the
+    policy unfreezes (reference rl_learner.py:147-172)."""
+    monkeypatch.chdir(tmp_path)
+    from distar_amd.learner.rl_learner import RLLearner
+    torch.manual_seed(0)
+    cfg = Config({'learner': {'player_id': 'MP0', 'job_type': 'fake',
+                              'use_cuda': False, 'use_amp': False,
+                              'value_pretrain_iters': 1,
+                              'data': {'batch_size': 2, 'trajectory_length': 3},
+                              'hook': {'after_iter': {
+                                  'log_show': {'ext_args': {'freq': 1000}}}}},
+                  'common': {'experiment_name': 'test_vp_switch',
+                             'type': 'train'},
+                  'model': {'enable_baselines': ['winloss']}})
+    learner = RLLearner(cfg)
+    m = getattr(learner._model, 'module', learner._model)
+    pol_before = m.policy.action_type_head.action_fc.layer2[0].weight.detach().clone()
+    learner.run(max_iterations=1)
+    # iteration 1 ran under pretrain: policy untouched, loss was critic-only
+    torch.testing.assert_close(
+        m.policy.action_type_head.action_fc.layer2[0].weight, pol_before)
+    assert learner._remain_value_pretrain_iters == 0
+    learner.run(max_iterations=2)
+    assert learner._remain_value_pretrain_iters == -1
+    assert not learner._loss.only_update_value
+    assert not m.only_update_baseline
+    # policy now moves
+    assert not torch.equal(
+        m.policy.action_type_head.action_fc.layer2[0].weight, pol_before)
+    learner._dataloader.close()
