@@ -114,3 +114,20 @@ def test_seed_everything_reproducible_training():
         return float(ld['total_loss'])
 
     assert one_loss() == one_loss()
+
+
+def test_alphastar_var_record_grid():
+    """The RL metric grid renders registered {field}x{column} cells
+    (reference log_helper.py:689-751)."""
+    from distar_amd.utils.log import AlphaStarVarRecord
+    rec = AlphaStarVarRecord(length=4)
+    for key in ('winloss/reward', 'winloss/td', 'battle/value', 'kl/action_type'):
+        rec.register_var(key)
+        rec.update_var({key: 1.25})
+    text = rec.get_star_text()
+    assert 'winloss' in text and 'battle' in text and 'kl' in text
+    assert '1.25000' in text
+    # pretty_print survives nested dicts
+    from distar_amd.utils.log import pretty_print
+    out = pretty_print({'a': {'b': 1, 'c': 'x'}, 'd': 2.5}, direct_print=False)
+    assert 'b' in out and 'd' in out
