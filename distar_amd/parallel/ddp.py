@@ -32,7 +32,7 @@ from .dist import is_initialized
 
 
 class _Bucket:
-    __slots__ = ('params', 'flat', 'views', 'pending', 'work', 'launched')
+    __slots__ = ('params', 'flat', 'views', 'pending', 'work', 'launched', 'stale')
 
     def __init__(self, params: List[torch.nn.Parameter]):
         self.params = params
@@ -47,11 +47,13 @@ class _Bucket:
         self.pending = len(params)
         self.work = None
         self.launched = False
+        self.stale = False
 
     def reset(self):
         self.pending = len(self.params)
         self.work = None
         self.launched = False
+        self.stale = False
 
     def launch(self, process_group):
         for p, v in zip(self.params, self.views):
@@ -113,6 +115,13 @@ class DistModule(nn.Module):
     def _register_hooks(self):
         def make_hook(bucket):
             def hook(param):
+                if bucket.launched:
+                    # gradient accumulation: a backward after the bucket's
+                    # eager launch means the reduced flat buffer is partial —
+                    # mark stale so sync_gradients() relaunches with the
+                    # fully-accumulated grads.
+                    bucket.stale = True
+                    return
                 bucket.pending -= 1
                 if bucket.pending == 0:
                     bucket.launch(self.group)
@@ -156,6 +165,12 @@ class DistModule(nn.Module):
         world_size = dist.get_world_size(self.group)
         for b in self._buckets:
             if not b.launched:
+                b.launch(self.group)
+            elif b.stale:
+                # accumulation happened after the eager launch: drain the
+                # in-flight reduce, then relaunch with the accumulated grads.
+                if b.work is not None:
+                    b.work.wait()
                 b.launch(self.group)
         for b in self._buckets:
             b.finish(world_size)

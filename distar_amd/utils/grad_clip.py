@@ -48,21 +48,33 @@ class GradClip:
                     self.norm_type)
             return total.item()
         if self.clip_type == 'max_norm':
-            # clip against an EMA of the global grad norm
+            # clip against a bias-corrected EMA of the global grad norm —
+            # reference semantics (`grad_clip.py:50-74`): beta1=0.95 EMA
+            # started at 0, bias correction 1-beta1^step, warmup without
+            # clipping for begin_step steps, EMA always fed the raw norm.
+            beta1 = 0.95
             with torch.no_grad():
                 total = torch.norm(torch.stack(
                     [torch.norm(p.grad, self.norm_type) for p in params]),
                     self.norm_type).item()
                 if self._ema is None:
-                    self._ema = total
-                clip_to = self._ema * self.threshold
-                if self.step > self.begin_step and total > clip_to:
-                    scale = clip_to / (total + 1e-6)
-                    torch._foreach_mul_([p.grad for p in params], scale)
-                self._ema = 0.99 * self._ema + 0.01 * min(total, self._ema * self.ignore_threshold)
+                    self._ema = 0.0
+                if self.step > self.begin_step:
+                    bias_correction = 1 - beta1 ** self.step
+                    clip_coef = (self._ema / bias_correction) * self.threshold \
+                        / (total + 1e-6)
+                    if clip_coef < 1:
+                        torch._foreach_mul_([p.grad for p in params], clip_coef)
+                self._ema = beta1 * self._ema + (1 - beta1) * total
             return total
         if self.clip_type == 'momentum_norm':
-            # per-parameter EMA-normalized clip (reference SL default).
+            # per-parameter EMA-normalized clip (reference SL default,
+            # `grad_clip.py:75-109`).  Reference *intended* semantics: no clip
+            # on the first step (EMA seeded with the first observed norms),
+            # from step 2 scale each grad so its norm <= threshold * ema, then
+            # update ema with the POST-clip norm at 0.99/0.01.  (The reference
+            # code has an append-instead-of-assign bug that leaves the EMA
+            # None forever, i.e. it never clips — see PARITY.md.)
             # Fully vectorized: one fused norm kernel + tensorized EMA state,
             # ZERO host syncs (the reference loops with .item() per param).
             with torch.no_grad():
@@ -70,14 +82,15 @@ class GradClip:
                 norms = torch.stack(torch._foreach_norm(grads, self.norm_type))
                 ema = self._state.get('ema')
                 if ema is None or ema.shape != norms.shape:
-                    ema = norms.clone()
-                clip_to = ema * self.threshold
-                if self.step > self.begin_step:
-                    scale = (clip_to / (norms + 1e-6)).clamp(max=1.0)
+                    # first step: no clip, seed the EMA with the raw norms
+                    self._state['ema'] = norms.clone()
+                    post = norms
+                else:
+                    scale = (self.threshold * ema / (norms + 1e-6)).clamp(max=1.0)
                     torch._foreach_mul_(grads, list(scale.unbind()))
-                self._state['ema'] = 0.99 * ema + \
-                    0.01 * torch.minimum(norms, ema * self.ignore_threshold)
-                total = torch.norm(norms, 2)
+                    post = norms * scale
+                    self._state['ema'] = 0.99 * ema + 0.01 * post
+                total = torch.norm(post, self.norm_type)
             return total.item()
         if self.clip_type == 'clip_value':
             # Adam-like second-moment clamp
