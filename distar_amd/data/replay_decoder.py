@@ -13,14 +13,16 @@ Functional parity with the reference's
 Needs a StarCraft II install + s2clientprotocol (gated like envs/env.py);
 `FilterActions` and the windowing logic are pure and unit-tested on CPU.
 """
+import json
 import os
 
 import torch
 
 from ..envs.protocol import (SC2_PROTO_AVAILABLE, RemoteController,
-                             launch_game_process)
+                             launch_game_process, version_info)
 from ..lib.consts import MAX_DELAY
 from ..lib.features import Features
+from ..lib.mpq import MPQArchive
 
 RESTART_REPLAY_INTERVAL = 10
 PASS1_STEP = 50
@@ -57,12 +59,8 @@ class FilterActions:
 
 class ReplayDecoder:
     def __init__(self, cfg):
-        if not SC2_PROTO_AVAILABLE:
-            raise ImportError(
-                'ReplayDecoder needs s2clientprotocol + a StarCraft II install '
-                '(not shipped in this offline image). SL training here uses '
-                "the 'offline' dataloader source (pre-decoded step files) or "
-                "the 'remote' source fed by a replay-actor fleet.")
+        if not SC2_PROTO_AVAILABLE:      # pragma: no cover - protobuf absent
+            raise ImportError('ReplayDecoder needs google.protobuf')
         self._whole_cfg = cfg
         self._filter = FilterActions(cfg)
         self._proc = None
@@ -71,15 +69,19 @@ class ReplayDecoder:
         self._cur_version = None
 
     def _version_of(self, replay_path):
-        """SC2 version sniff from MPQ metadata (reference :361-380)."""
+        """SC2 version sniff from the replay's MPQ `replay.gamemetadata.json`
+        (reference :361-380).  Returns a '4.10.0'-style string the version
+        table (assets/sc2_versions.json) can route to a Base<build> binary,
+        or None when the metadata is unreadable/unknown."""
         try:
-            import mpyq
-            archive = mpyq.MPQArchive(replay_path)
-            header = archive.header['user_data_header']['content']
-            # versions are at fixed offsets of the user data header
-            import struct
-            parts = struct.unpack('>4I', header[60:76])
-            return '.'.join(map(str, parts[:3]))
+            meta = json.loads(
+                MPQArchive(replay_path).read_file('replay.gamemetadata.json'))
+            version = '.'.join(meta['GameVersion'].split('.')[:3])
+            base_build = int(meta['BaseBuild'][4:])
+            if version_info(version) is None:
+                # unknown point release: route by the base build directly
+                return base_build
+            return version
         except Exception:  # noqa: BLE001
             return None
 
@@ -88,7 +90,9 @@ class ReplayDecoder:
                 self._decode_count % RESTART_REPLAY_INTERVAL != 0:
             return
         self.close()
-        self._proc, port = launch_game_process(self._whole_cfg)
+        # the sniffed version selects the binary + -dataVersion
+        self._proc, port = launch_game_process(self._whole_cfg,
+                                               version=version)
         self._controller = RemoteController('127.0.0.1', port)
         self._cur_version = version
 
@@ -106,8 +110,12 @@ class ReplayDecoder:
         version = self._version_of(replay_path)
         self._ensure_sc2(version)
         ctrl = self._controller
-        # pass 1: harvest actions at 1x1
-        ctrl.start_replay(replay_path, player_idx + 1, resolution=1)
+        # pass 1: harvest actions at 1x1 minimap resolution
+        ctrl.start_replay(replay_path, player_idx + 1,
+                          minimap_resolution=(1, 1))
+        game_info = ctrl.game_info()
+        map_size = (game_info.start_raw.map_size.x,
+                    game_info.start_raw.map_size.y)
         raw_actions = []
         while True:
             obs = ctrl.observe()
@@ -122,9 +130,10 @@ class ReplayDecoder:
                 break
             ctrl.step(PASS1_STEP)
         raw_actions = self._filter.run(raw_actions)
-        # pass 2: step by inter-action delays, transform each action
-        ctrl.start_replay(replay_path, player_idx + 1)
-        game_info = None  # populated from the controller's game info request
+        # pass 2: re-open at map resolution, step by inter-action delays,
+        # transform each action (reference :279-335)
+        ctrl.start_replay(replay_path, player_idx + 1,
+                          minimap_resolution=map_size)
         feature = Features(ctrl.game_info(), ctrl.observe()['raw_obs'],
                            self._whole_cfg)
         traj_data = []

@@ -3,20 +3,22 @@
 `SC2Env` keeps the reference's public contract — ``reset() -> {agent: obs}``,
 ``step(actions)`` with per-agent skip-step scheduling and 0-3 random delay
 steps, win/loss outcome extraction, replay saving, and a full game restart
-every 10 episodes — implemented against a `RemoteController` protocol
-abstraction (envs/protocol.py).
+every 10 episodes — on top of the repo-native protocol stack
+(envs/protocol.py): one SC2 process + RemoteController per agent, game/base
+port sets reserved via portspicker (NOT the websocket listen ports —
+reference `envs/env.py:211-274`), Computer-player (bot) opponents with
+difficulty/ai_build, and realtime mode for play.
 
-The StarCraft II protobuf bindings (s2clientprotocol) and game binary are
-not shipped in this image; constructing `SC2Env` without them raises
-ImportError with instructions, and `MockSC2Env` (mock_env.py) provides the
-spec-identical no-binary environment used by tests and synthetic rollouts
-(the same split the reference makes with `pysc2/env/mock_sc2_env.py`).
+Requires a StarCraft II install (SC2PATH); the protobuf wire layer is
+self-contained (lib/sc2_protos.py).  `MockSC2Env` (mock_env.py) remains the
+spec-identical no-binary environment used by tests and synthetic rollouts.
 """
 import random
 import time
 
 from .map_info import get_map_size
-from .protocol import SC2_PROTO_AVAILABLE, RemoteController, launch_game_process
+from .portspicker import pick_unused_ports, return_ports
+from .protocol import RemoteController, launch_game_process
 
 RESTART_EPISODE_INTERVAL = 10
 DELAY_WEIGHTS = [1, 1, 1, 1]           # 0-3 extra latency steps
@@ -24,23 +26,26 @@ DELAY_WEIGHTS = [1, 1, 1, 1]           # 0-3 extra latency steps
 
 class SC2Env:
     def __init__(self, cfg, seed=None):
-        if not SC2_PROTO_AVAILABLE:
-            raise ImportError(
-                'SC2Env needs the s2clientprotocol protobuf bindings and a '
-                'StarCraft II install; neither ships in this offline image. '
-                'Use distar_amd.envs.MockSC2Env for spec-identical synthetic '
-                'episodes, or install s2clientprotocol + SC2 and point '
-                'SC2PATH at the install.')
         self._whole_cfg = cfg
         self._cfg = cfg.env
         self._agent_num = self._cfg.get('player_num', 2)
+        self._bots = list(self._cfg.get('bots', []))
+        # bot_difficulty shorthand: one built-in bot opponent
+        if not self._bots and self._cfg.get('bot_difficulty'):
+            self._bots = [{'race': self._cfg.get('bot_race', 'zerg'),
+                           'difficulty': self._cfg.get('bot_difficulty')}]
+        if self._bots:                   # bots replace absent agents
+            self._agent_num = max(1, self._agent_num - len(self._bots))
+        self._realtime = bool(self._cfg.get('realtime', False))
         self._map_name = self._cfg.get('map_name', 'KingsCove')
         self.map_size = get_map_size(self._map_name)
         self._rng = random.Random(seed)
+        self._random_seed = seed
         self._save_replay_episodes = self._cfg.get('save_replay_episodes', 0)
         self._replay_dir = self._cfg.get('replay_dir', '.')
         self._game_procs = []
         self._controllers = []
+        self._game_ports = []
         self._episode_count = 0
         self._episode_steps = 0
         self._next_obs_step = [0] * self._agent_num
@@ -49,20 +54,40 @@ class SC2Env:
     # ------------------------------------------------------------ lifecycle
     def _launch_game(self):
         self.close()
-        ports = []
         for _ in range(self._agent_num):
             proc, port = launch_game_process(self._whole_cfg)
             self._game_procs.append(proc)
-            ports.append(port)
             self._controllers.append(RemoteController('127.0.0.1', port))
-        self._create_join(ports)
+        self._create_join()
         self._launched = True
 
-    def _create_join(self, ports):
-        self._controllers[0].create_game(self._map_name, self._agent_num, ports)
+    def _player_setup(self):
+        players = [{'type': 'participant'} for _ in range(self._agent_num)]
+        for bot in self._bots:
+            players.append({'type': 'computer',
+                            'race': bot.get('race', 'random'),
+                            'difficulty': bot.get('difficulty', 'very_hard'),
+                            'ai_build': bot.get('ai_build')})
+        return players
+
+    def _create_join(self):
+        races = self._cfg.get('races', ['zerg'] * self._agent_num)
+        self._controllers[0].create_game(
+            map_name=self._map_name, players=self._player_setup(),
+            realtime=self._realtime, random_seed=self._random_seed)
+        ports = None
+        if self._agent_num > 1:
+            # 2 server ports + 2 per additional client, freshly reserved —
+            # the SC2 websocket listen ports are already bound and must not
+            # be reused here (reference `envs/env.py:258-266`)
+            flat = pick_unused_ports(2 * self._agent_num)
+            self._game_ports = flat
+            ports = {'server': (flat[0], flat[1]),
+                     'clients': [(flat[2 + 2 * i], flat[3 + 2 * i])
+                                 for i in range(self._agent_num - 1)]}
         for i, ctrl in enumerate(self._controllers):
-            ctrl.join_game(race=self._cfg.get('races', ['zerg'] * self._agent_num)[i],
-                           ports=ports)
+            ctrl.join_game(race=races[i], ports=ports,
+                           minimap_resolution=self.map_size)
 
     def reset(self):
         for attempt in range(10):
@@ -98,8 +123,9 @@ class SC2Env:
         target = min(self._next_obs_step)
         delay = self._rng.choices([0, 1, 2, 3], weights=DELAY_WEIGHTS)[0]
         step_count = max(target + delay - self._episode_steps, 1)
-        for ctrl in self._controllers:
-            ctrl.step(step_count)
+        if not self._realtime:
+            for ctrl in self._controllers:
+                ctrl.step(step_count)
         self._episode_steps += step_count
         obs, rewards, done = {}, {}, False
         for i, ctrl in enumerate(self._controllers):
@@ -135,6 +161,9 @@ class SC2Env:
                 proc.kill()
             except Exception:  # noqa: BLE001
                 pass
+        if self._game_ports:
+            return_ports(self._game_ports)
+            self._game_ports = []
         self._controllers = []
         self._game_procs = []
         self._launched = False

@@ -53,9 +53,17 @@ class LoadCkptHook(Hook):
             state_dict_mask=engine.whole_cfg.learner.get('state_dict_mask', []),
             strict=engine.whole_cfg.learner.get('load_strict', True),
             logger_prints=engine.info)
-        # fast-forward the LR scheduler to the restored iteration
-        for _ in range(engine.last_iter.val):
-            engine.lr_scheduler.step()
+        # fast-forward the LR scheduler to the restored iteration.  torch
+        # warns "scheduler.step() before optimizer.step()" here because no
+        # optimizer step has run yet this process — that is exactly the
+        # resume situation, so the warning is noise; the schedule lands on
+        # last_epoch == last_iter, same as if training had never stopped.
+        import warnings
+        with warnings.catch_warnings():
+            warnings.filterwarnings(
+                'ignore', message='.*lr_scheduler.step().*optimizer.step().*')
+            for _ in range(engine.last_iter.val):
+                engine.lr_scheduler.step()
         engine.info(f'loaded checkpoint {path} (iter {engine.last_iter.val})')
 
 

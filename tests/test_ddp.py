@@ -88,6 +88,70 @@ def test_distmodule_grads_are_averaged():
         torch.testing.assert_close(results[0][n], g)
 
 
+def _accum_worker(rank, world, init_file, q):
+    dist.init_process_group('gloo', init_method=f'file://{init_file}',
+                            rank=rank, world_size=world)
+    from distar_amd.parallel.ddp import DistModule
+    torch.manual_seed(7)
+    model = torch.nn.Sequential(torch.nn.Linear(16, 8), torch.nn.Linear(8, 4))
+    dm = DistModule(model, bucket_cap_mb=1)
+    torch.manual_seed(100 + rank)
+    # two backwards (gradient accumulation) before one sync: the eager bucket
+    # launch after backward #1 must be superseded by the accumulated grads
+    for _ in range(2):
+        x = torch.randn(8, 16)
+        dm.module(x).sum().backward()
+    dm.sync_gradients()
+    grads = {n: p.grad.clone().numpy() for n, p in model.named_parameters()}
+    q.put((rank, grads))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_distmodule_gradient_accumulation():
+    import time as _time
+    for attempt in range(3):
+        try:
+            with tempfile.TemporaryDirectory() as d:
+                init_file = os.path.join(d, 'init')
+                ctx = mp.get_context('spawn')
+                q = ctx.Queue()
+                procs = [ctx.Process(target=_accum_worker,
+                                     args=(r, 2, init_file, q))
+                         for r in range(2)]
+                for p in procs:
+                    p.start()
+                results = {}
+                for _ in range(2):
+                    rank, grads = q.get(timeout=90)
+                    results[rank] = grads
+                for p in procs:
+                    p.join(timeout=60)
+            break
+        except Exception:
+            if attempt == 2:
+                raise
+            _time.sleep(5)
+    results = {r: {n: torch.from_numpy(g) for n, g in d.items()}
+               for r, d in results.items()}
+    # expected: average over ranks of the two-backward accumulated grads
+    torch.manual_seed(7)
+    model = torch.nn.Sequential(torch.nn.Linear(16, 8), torch.nn.Linear(8, 4))
+    expected = {}
+    for rank in range(2):
+        torch.manual_seed(100 + rank)
+        model.zero_grad()
+        for _ in range(2):
+            x = torch.randn(8, 16)
+            model(x).sum().backward()
+        for n, p in model.named_parameters():
+            expected[n] = expected.get(n, 0) + p.grad / 2
+    for n, g in expected.items():
+        torch.testing.assert_close(results[0][n], g)
+        torch.testing.assert_close(results[1][n], g)
+
+
 def _cross_rank_worker(rank, world, init_file, q):
     dist.init_process_group('gloo', init_method=f'file://{init_file}',
                             rank=rank, world_size=world)
