@@ -5,12 +5,15 @@ TransformerLayer / Transformer) and `:37-68` (AttentionPool).  Module / key
 layout matches the reference checkpoints (`attention_pre.0.*`, `project.0.*`,
 `layernorm1/2.*`, `mlp.{i}.0.*`, `embedding.0.*`).
 
-MI355X note: attention math is expressed as batched matmuls on contiguous
-(B, heads, N, head_dim) blocks so hipBLASLt strided-batched GEMM (MFMA) takes
-all of it in bf16 autocast; the additive -1e9 mask convention is preserved
-because downstream heads rely on it.
+MI355X note: on GPU in bf16 with head_dim 128 (the entity transformer), the
+attention core runs the hand-written CDNA4 MFMA flash kernel
+(ops/hip/entity_attn.hip, K1): the key-padding mask travels as an integer
+entity count per row — no (B,1,N,N) mask tensor, no qkv permute copies.
+Other shapes (the 20-token build-order mini-transformer, CPU) use the eager
+composition with the reference's additive -1e9 mask convention.
 """
 import math
+import os
 from typing import Optional
 
 import torch
@@ -30,26 +33,42 @@ class Attention(nn.Module):
         self.project = fc_block(head_dim * head_num, output_dim)
 
     def forward(self, x, mask: Optional[torch.Tensor] = None):
+        """`mask`: (B, N) boolean key-validity PREFIX mask (sequence_mask of
+        the entity count) or None."""
         assert len(x.shape) == 3
         B, N = x.shape[:2]
-        qkv = self.attention_pre(x).view(B, N, 3, self.head_num, self.head_dim)
+        qkv = self.attention_pre(x)                      # (B, N, 3*H*D)
+        use_hip = (x.is_cuda and self.dropout is None and self.head_dim == 128
+                   and qkv.dtype == torch.bfloat16
+                   and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1')
+        if use_hip:
+            from ...ops.entity_attn import entity_attention
+            entity_num = mask.int().sum(dim=-1) if mask is not None else None
+            attention = entity_attention(qkv, entity_num, self.head_num,
+                                         1.0 / math.sqrt(self.head_dim))
+            return self.project(attention)
+        qkv = qkv.view(B, N, 3, self.head_num, self.head_dim)
         qkv = qkv.permute(2, 0, 3, 1, 4)  # 3, B, H, N, D
         query, key, value = qkv[0], qkv[1], qkv[2]
+        mask4 = None
+        if mask is not None:
+            # (B, N) key mask -> (B, 1, N, N) broadcast over heads/queries
+            mask4 = mask.unsqueeze(1).repeat(1, mask.shape[1], 1).unsqueeze(1)
         if x.is_cuda and self.dropout is None:
-            # fused scaled-dot-product path (flash/mem-efficient on ROCm).
+            # library fused path (A/B comparisons via DISTAR_AMD_DISABLE_HIP).
             # Additive -1e9 mask keeps the reference's finite-logit semantics
             # for fully-padded query rows (a bool mask would yield NaNs).
             attn_mask = None
-            if mask is not None:
-                attn_mask = torch.zeros(mask.shape, dtype=query.dtype,
+            if mask4 is not None:
+                attn_mask = torch.zeros(mask4.shape, dtype=query.dtype,
                                         device=query.device)
-                attn_mask.masked_fill_(~mask, -1e9)
+                attn_mask.masked_fill_(~mask4, -1e9)
             attention = F.scaled_dot_product_attention(query, key, value,
                                                        attn_mask=attn_mask)
         else:
             score = torch.matmul(query, key.transpose(-2, -1)) / math.sqrt(self.head_dim)
-            if mask is not None:
-                score = score.masked_fill(~mask, -1e9)
+            if mask4 is not None:
+                score = score.masked_fill(~mask4, -1e9)
             score = F.softmax(score, dim=-1)
             if self.dropout is not None:
                 score = self.dropout(score)
@@ -114,9 +133,8 @@ class Transformer(nn.Module):
         ])
 
     def forward(self, x, mask: Optional[torch.Tensor] = None):
-        if mask is not None:
-            # (B, N) key mask -> (B, 1, N, N) broadcast over heads/queries
-            mask = mask.unsqueeze(dim=1).repeat(1, mask.shape[1], 1).unsqueeze(dim=1)
+        # mask stays (B, N): Attention expands it for eager paths and turns
+        # it into per-row entity counts for the HIP kernel
         x = self.embedding(x)
         if self.dropout is not None:
             x = self.dropout(x)

@@ -17,7 +17,8 @@ import torch
 class _FusedMaskedCE(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits, labels, mask):
-        from .hip_ext import ops
+        from . import hip_ext
+        ops = hip_ext.require()
         loss, lse = ops.masked_ce_fwd(logits, labels, mask)
         ctx.save_for_backward(logits, labels, lse,
                               mask if mask is not None else torch.empty(0))
@@ -25,7 +26,8 @@ class _FusedMaskedCE(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, gout):
-        from .hip_ext import ops
+        from . import hip_ext
+        ops = hip_ext.require()
         logits, labels, lse, mask = ctx.saved_tensors
         mask = mask if mask.numel() else None
         dlogits = ops.masked_ce_bwd(logits, labels, mask, lse,

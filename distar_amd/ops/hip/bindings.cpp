@@ -392,7 +392,126 @@ torch::Tensor kl_bwd(torch::Tensor t_logits, torch::Tensor s_logits,
   return d;
 }
 
+// ------------------------------------------------------------- entity attn
+extern "C" __global__ void entity_attn_fwd_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+    const int*, __hip_bfloat16*, float*, float, int, int, int,
+    long, long, long, long, long);
+extern "C" __global__ void attn_drow_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, float*, int, int, int,
+    long, long);
+extern "C" __global__ void entity_attn_bwd_kv_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+    const __hip_bfloat16*, const float*, const float*, const int*,
+    __hip_bfloat16*, __hip_bfloat16*, float, int, int, int,
+    long, long, long, long, long, long, long);
+extern "C" __global__ void entity_attn_bwd_q_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+    const __hip_bfloat16*, const float*, const float*, const int*,
+    __hip_bfloat16*, float, int, int, int,
+    long, long, long, long, long, long, long);
+extern "C" __global__ void mfma_selftest_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, float*);
+
+static const __hip_bfloat16* bfp(const torch::Tensor& t) {
+  return reinterpret_cast<const __hip_bfloat16*>(t.data_ptr());
+}
+static __hip_bfloat16* bfp_mut(torch::Tensor& t) {
+  return reinterpret_cast<__hip_bfloat16*>(t.data_ptr());
+}
+
+// qkv: (B, N, 3*H*128) bf16 contiguous (packed [q|k|v][head][dim] order, the
+// attention_pre fc output); entity_num: (B,) int32 or None.
+std::vector<torch::Tensor> entity_attn_fwd(
+    torch::Tensor qkv, c10::optional<torch::Tensor> entity_num,
+    int64_t H, double scale) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && qkv.dim() == 3);
+  TORCH_CHECK(qkv.scalar_type() == torch::kBFloat16, "qkv must be bf16");
+  int64_t B = qkv.size(0), N = qkv.size(1), F = qkv.size(2);
+  int64_t HD = F / 3;
+  TORCH_CHECK(HD == H * 128, "entity_attn: head_dim must be 128");
+  auto out = torch::empty({B, N, HD}, qkv.options());
+  auto lse = torch::empty({B, H, N}, qkv.options().dtype(torch::kFloat32));
+  const int* en = nullptr;
+  if (entity_num.has_value()) {
+    TORCH_CHECK(entity_num->scalar_type() == torch::kInt32);
+    en = entity_num->data_ptr<int>();
+  }
+  dim3 grid((N + 63) / 64, H, B);
+  size_t lds = 64 * 256 * 2 + 128 * 128 + 64 * 128;    // Q,K,Vt,Ps
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(entity_attn_fwd_kernel, grid, dim3(256), lds,
+                     stream.stream(),
+                     bfp(qkv), bfp(qkv) + HD, bfp(qkv) + 2 * HD, en,
+                     bfp_mut(out), lse.data_ptr<float>(), (float)scale,
+                     (int)B, (int)H, (int)N,
+                     (long)(N * F), (long)F, 128L,
+                     (long)(N * HD), (long)HD);
+  return {out, lse};
+}
+
+torch::Tensor entity_attn_bwd(
+    torch::Tensor qkv, c10::optional<torch::Tensor> entity_num,
+    torch::Tensor out, torch::Tensor dout, torch::Tensor lse,
+    int64_t H, double scale) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous());
+  TORCH_CHECK(dout.is_contiguous() && dout.scalar_type() == torch::kBFloat16);
+  int64_t B = qkv.size(0), N = qkv.size(1), F = qkv.size(2);
+  int64_t HD = F / 3;
+  auto drow = torch::empty({B, H, N}, lse.options());
+  auto dqkv = torch::empty_like(qkv);
+  const int* en = nullptr;
+  if (entity_num.has_value()) en = entity_num->data_ptr<int>();
+  auto stream = c10::hip::getCurrentHIPStream();
+  long total_rows = B * H * N;
+  int drow_blocks = (int)std::min<long>((total_rows + 3) / 4, 8192);
+  hipLaunchKernelGGL(attn_drow_kernel, dim3(drow_blocks), dim3(256), 0,
+                     stream.stream(), bfp(dout), bfp(out),
+                     drow.data_ptr<float>(), (int)B, (int)H, (int)N,
+                     (long)(N * HD), (long)HD);
+  dim3 grid((N + 63) / 64, H, B);
+  size_t lds_kv = 4 * 64 * 256 + 2 * 128 * 128 + 64 * 128 +
+                  2 * 64 * sizeof(float);
+  hipLaunchKernelGGL(entity_attn_bwd_kv_kernel, grid, dim3(256), lds_kv,
+                     stream.stream(),
+                     bfp(qkv), bfp(qkv) + HD, bfp(qkv) + 2 * HD,
+                     bfp(dout), lse.data_ptr<float>(), drow.data_ptr<float>(),
+                     en, bfp_mut(dqkv) + HD, bfp_mut(dqkv) + 2 * HD,
+                     (float)scale, (int)B, (int)H, (int)N,
+                     (long)(N * F), (long)F, 128L,
+                     (long)(N * HD), (long)HD,
+                     (long)(N * F), (long)F);
+  size_t lds_q = 4 * 64 * 256 + 128 * 128 + 64 * 128 + 2 * 64 * sizeof(float);
+  hipLaunchKernelGGL(entity_attn_bwd_q_kernel, grid, dim3(256), lds_q,
+                     stream.stream(),
+                     bfp(qkv), bfp(qkv) + HD, bfp(qkv) + 2 * HD,
+                     bfp(dout), lse.data_ptr<float>(), drow.data_ptr<float>(),
+                     en, bfp_mut(dqkv),
+                     (float)scale, (int)B, (int)H, (int)N,
+                     (long)(N * F), (long)F, 128L,
+                     (long)(N * HD), (long)HD,
+                     (long)(N * F), (long)F);
+  return dqkv;
+}
+
+torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor Bm) {
+  TORCH_CHECK(A.is_cuda() && A.is_contiguous() && Bm.is_contiguous());
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(A.size(0) == 16 && A.size(1) == 32);
+  auto D = torch::empty({16, 16}, A.options().dtype(torch::kFloat32));
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(mfma_selftest_kernel, dim3(1), dim3(64), 0,
+                     stream.stream(), bfp(A), bfp(Bm), D.data_ptr<float>());
+  return D;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("entity_attn_fwd", &entity_attn_fwd,
+        "K1 entity-transformer flash attention forward (bf16 MFMA)");
+  m.def("entity_attn_bwd", &entity_attn_bwd,
+        "K1 entity-transformer flash attention backward");
+  m.def("mfma_selftest", &mfma_selftest,
+        "one 16x16x32 bf16 MFMA: A(16,32) . B(16,32)^T -> (16,16) fp32");
   m.def("entropy_fwd", &entropy_fwd, "fused rowwise entropy forward");
   m.def("entropy_bwd", &entropy_bwd, "fused rowwise entropy backward");
   m.def("kl_fwd", &kl_fwd, "fused rowwise KL(teacher||student) forward");

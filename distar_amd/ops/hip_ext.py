@@ -52,3 +52,13 @@ def maybe_ext(tensor):
             '`python setup_hip.py build` (hipcc --offload-arch=gfx950); '
             'refusing to fall back to eager on GPU.')
     return ext
+
+
+def require():
+    """The extension module, or ImportError (HIP-op call sites on GPU)."""
+    ext = _load()
+    if ext is None:
+        raise ImportError(
+            'distar_amd HIP extension (_hip_ops) is not built; run '
+            '`python setup_hip.py build` (hipcc --offload-arch=gfx950).')
+    return ext

@@ -17,14 +17,16 @@ import torch.nn.functional as F
 class _FusedEntropy(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits):
-        from .hip_ext import ops
+        from . import hip_ext
+        ops = hip_ext.require()
         ent, lse = ops.entropy_fwd(logits)
         ctx.save_for_backward(logits, lse, ent)
         return ent
 
     @staticmethod
     def backward(ctx, gout):
-        from .hip_ext import ops
+        from . import hip_ext
+        ops = hip_ext.require()
         logits, lse, ent = ctx.saved_tensors
         return ops.entropy_bwd(logits, lse, ent, gout.contiguous())
 
@@ -32,14 +34,16 @@ class _FusedEntropy(torch.autograd.Function):
 class _FusedKL(torch.autograd.Function):
     @staticmethod
     def forward(ctx, teacher_logits, student_logits):
-        from .hip_ext import ops
+        from . import hip_ext
+        ops = hip_ext.require()
         kl, t_lse, s_lse = ops.kl_fwd(teacher_logits, student_logits)
         ctx.save_for_backward(teacher_logits, student_logits, t_lse, s_lse)
         return kl
 
     @staticmethod
     def backward(ctx, gout):
-        from .hip_ext import ops
+        from . import hip_ext
+        ops = hip_ext.require()
         t_logits, s_logits, t_lse, s_lse = ctx.saved_tensors
         ds = ops.kl_bwd(t_logits, s_logits, t_lse, s_lse, gout.contiguous())
         return None, ds

@@ -333,3 +333,108 @@ def test_fused_rowwise_entropy_kl_match_eager():
         torch.testing.assert_close(s.grad, s_ref.grad, rtol=1e-4, atol=1e-6)
     finally:
         os.environ.pop('DISTAR_AMD_FUSED_RL_ROWWISE', None)
+
+
+# ------------------------------------------------------ K1 entity attention
+def test_mfma_selftest_layout():
+    """One 16x16x32 bf16 MFMA vs torch matmul — isolates fragment-layout
+    mistakes from attention logic (asymmetric random inputs per the guide's
+    transpose-detection rule)."""
+    from distar_amd.ops import hip_ext
+    ops = hip_ext.require()
+    torch.manual_seed(0)
+    A = torch.randn(16, 32, device='cuda').bfloat16()
+    B = torch.randn(16, 32, device='cuda').bfloat16()
+    D = ops.mfma_selftest(A, B)
+    ref = A.float() @ B.float().T
+    torch.testing.assert_close(D, ref, rtol=2e-2, atol=2e-2)
+
+
+def _eager_attn_fp32(qkv, entity_num, H, scale):
+    """fp32 reference with the reference's additive -1e9 convention."""
+    B, N, F3 = qkv.shape
+    D = F3 // 3 // H
+    q, k, v = qkv.float().split(F3 // 3, dim=-1)
+    q = q.view(B, N, H, D).permute(0, 2, 1, 3)
+    k = k.view(B, N, H, D).permute(0, 2, 1, 3)
+    v = v.view(B, N, H, D).permute(0, 2, 1, 3)
+    s = torch.matmul(q, k.transpose(-2, -1)) * scale
+    if entity_num is not None:
+        key = torch.arange(N, device=qkv.device).view(1, 1, 1, N)
+        s = s + torch.where(key < entity_num.view(B, 1, 1, 1), 0.0, -1e9)
+    p = torch.softmax(s, dim=-1)
+    o = torch.matmul(p, v)
+    return o.permute(0, 2, 1, 3).reshape(B, N, H * D)
+
+
+@pytest.mark.parametrize('N,nums', [
+    (512, [512, 300, 64, 1, 17]),
+    (256, [256, 250, 100, 3, 256]),
+    (160, [160, 96, 33, 160, 2]),      # N not a multiple of 64
+])
+def test_entity_attn_fwd_matches_eager(N, nums):
+    from distar_amd.ops.entity_attn import entity_attention
+    torch.manual_seed(0)
+    B, H, D = 5, 2, 128
+    qkv = (torch.randn(B, N, 3 * H * D, device='cuda') * 0.5).bfloat16()
+    en = torch.tensor(nums[:B], device='cuda', dtype=torch.int32)
+    out = entity_attention(qkv, en, H, 1.0 / D ** 0.5)
+    ref = _eager_attn_fp32(qkv, en, H, 1.0 / D ** 0.5)
+    torch.testing.assert_close(out.float(), ref, rtol=3e-2, atol=3e-2)
+    # unmasked path too
+    out2 = entity_attention(qkv, None, H, 1.0 / D ** 0.5)
+    ref2 = _eager_attn_fp32(qkv, None, H, 1.0 / D ** 0.5)
+    torch.testing.assert_close(out2.float(), ref2, rtol=3e-2, atol=3e-2)
+
+
+def test_entity_attn_bwd_matches_eager():
+    from distar_amd.ops.entity_attn import entity_attention
+    torch.manual_seed(1)
+    B, N, H, D = 4, 512, 2, 128
+    scale = 1.0 / D ** 0.5
+    qkv0 = (torch.randn(B, N, 3 * H * D, device='cuda') * 0.5).bfloat16()
+    en = torch.tensor([512, 317, 65, 1], device='cuda', dtype=torch.int32)
+    dout = (torch.randn(B, N, H * D, device='cuda') * 0.5).bfloat16()
+
+    qkv_h = qkv0.detach().clone().requires_grad_(True)
+    out_h = entity_attention(qkv_h, en, H, scale)
+    out_h.backward(dout)
+
+    qkv_e = qkv0.detach().clone().float().requires_grad_(True)
+    out_e = _eager_attn_fp32(qkv_e, en, H, scale)
+    out_e.backward(dout.float())
+
+    torch.testing.assert_close(out_h.float(), out_e.detach(),
+                               rtol=3e-2, atol=3e-2)
+    # bf16 recompute backward vs fp32 eager: absolute tolerance scaled to
+    # the gradient magnitude
+    g_h, g_e = qkv_h.grad.float(), qkv_e.grad
+    scale_tol = g_e.abs().max().item()
+    torch.testing.assert_close(g_h, g_e, rtol=5e-2, atol=0.05 * scale_tol)
+
+
+def test_entity_transformer_hip_vs_eager_end_to_end():
+    """The full 3-layer entity Transformer module: HIP attention path vs
+    DISTAR_AMD_DISABLE_HIP=1 eager, same bf16 weights."""
+    from distar_amd.models.nn.blocks import sequence_mask
+    from distar_amd.models.nn.transformer import Transformer
+    torch.manual_seed(2)
+    tr = Transformer(input_dim=997, head_dim=128, hidden_dim=1024,
+                     output_dim=256, head_num=2, mlp_num=2, layer_num=3,
+                     ln_type='post').cuda()
+    x = torch.randn(3, 512, 997, device='cuda')
+    en = torch.tensor([512, 200, 7], device='cuda')
+    mask = sequence_mask(en, max_len=512)
+    with torch.autocast('cuda', dtype=torch.bfloat16):
+        out_hip = tr(x, mask=mask)
+    os.environ['DISTAR_AMD_DISABLE_HIP'] = '1'
+    try:
+        with torch.autocast('cuda', dtype=torch.bfloat16):
+            out_eager = tr(x, mask=mask)
+    finally:
+        os.environ.pop('DISTAR_AMD_DISABLE_HIP', None)
+    # valid rows must agree; padded rows are don't-care downstream (the
+    # encoder masks them before reduction)
+    m = mask.unsqueeze(-1)
+    torch.testing.assert_close((out_hip * m).float(), (out_eager * m).float(),
+                               rtol=5e-2, atol=5e-2)
