@@ -49,26 +49,39 @@ class ReinforcementLoss:
             baseline_values_dict[field][-1] = baseline_values_dict[field][-1] * flag
 
         loss_info_dict = {}
-        target_policy_probs_dict = {}
-        target_policy_log_probs_dict = {}
+        # per-head row statistics via the fused rowwise kernels on GPU
+        # (ops/{ce_loss,rl_rowwise}.py; eager composition on CPU):
+        # action log-prob = -(lse - logit[a]), per-row entropy, per-row
+        # KL(teacher||target) — the reference's Categorical-based prepare
+        # stage materializes full (T,B,C) probs AND log-probs per head
+        # (`rl_loss.py:63-90`); here neither is ever built.
+        from ..ops.ce_loss import masked_cross_entropy
+        from ..ops.rl_rowwise import rowwise_entropy, rowwise_kl
         target_action_log_probs_dict = {}
+        entropy_rows_dict = {}
+        kl_rows_dict = {}
+        dapo_rows_dict = {}
+        num_classes_dict = {}
         clipped_rhos_dict = {}
         for head_type in ['action_type', 'delay', 'queued', 'target_unit',
                           'selected_units', 'target_location']:
             target_logits = target_logits_dict[head_type]
             actions = actions_dict[head_type]
-            pi_target = torch.distributions.Categorical(logits=target_logits)
-            target_policy_probs = pi_target.probs
-            target_policy_log_probs = pi_target.logits
-            target_action_log_probs = pi_target.log_prob(actions)
+            num_classes_dict[head_type] = target_logits.shape[-1]
+            target_action_log_probs = -masked_cross_entropy(target_logits,
+                                                            actions)
+            entropy_rows_dict[head_type] = rowwise_entropy(target_logits)
+            kl_rows_dict[head_type] = rowwise_kl(
+                teacher_logits_dict[head_type], target_logits)
+            if self.use_dapo:
+                dapo_rows_dict[head_type] = rowwise_kl(
+                    inputs['successive_logit'][head_type], target_logits)
             behaviour_action_log_probs = behaviour_logp_dict[head_type]
             with torch.no_grad():
                 log_rhos = target_action_log_probs - behaviour_action_log_probs
                 if head_type == 'selected_units':
                     log_rhos = (log_rhos * masks_dict['selected_units_mask']).sum(dim=-1)
                 clipped_rhos = torch.exp(log_rhos).clamp_(max=1)
-            target_policy_probs_dict[head_type] = target_policy_probs
-            target_policy_log_probs_dict[head_type] = target_policy_log_probs
             if head_type == 'selected_units':
                 target_action_log_probs = target_action_log_probs.masked_fill(
                     ~masks_dict['selected_units_mask'], 0).sum(-1)
@@ -104,22 +117,22 @@ class ReinforcementLoss:
                                          gamma=self.gammas.baseline[field], field=field)
             total_critic_loss = total_critic_loss + \
                 self.loss_weights.baseline[field] * critic_loss
-            loss_info_dict[field + '/td'] = critic_loss.item()
-            loss_info_dict[field + '/reward'] = reward.float().mean().item()
-            loss_info_dict[field + '/value'] = baseline.mean().item()
+            loss_info_dict[field + '/td'] = critic_loss.detach()
+            loss_info_dict[field + '/reward'] = reward.float().mean()
+            loss_info_dict[field + '/value'] = baseline.mean().detach()
         if 'battle' in rewards_dict:
-            loss_info_dict['battle/reward'] = rewards_dict['battle'].float().mean().item()
+            loss_info_dict['battle/reward'] = rewards_dict['battle'].float().mean()
 
         # entropy
         total_entropy_loss, entropy_info = entropy_loss(
-            target_policy_probs_dict, target_policy_log_probs_dict, masks_dict,
+            entropy_rows_dict, num_classes_dict, masks_dict,
             head_weights_dict=self.entropy_head_weights)
         total_entropy_loss = total_entropy_loss * self.loss_weights.entropy
         loss_info_dict.update(entropy_info)
 
         # teacher KL
         total_kl_loss, action_type_kl_loss, kl_info = kl_loss(
-            target_policy_log_probs_dict, teacher_logits_dict, masks_dict, game_steps,
+            kl_rows_dict, masks_dict, game_steps,
             action_type_kl_steps=self.action_type_kl_steps,
             head_weights_dict=self.kl_head_weights)
         total_kl_loss = total_kl_loss * self.loss_weights.kl
@@ -129,7 +142,7 @@ class ReinforcementLoss:
         # DAPO
         if self.use_dapo:
             total_dapo_loss, dapo_info = dapo_loss(
-                target_policy_log_probs_dict, inputs['successive_logit'], masks_dict,
+                dapo_rows_dict, masks_dict,
                 game_steps, dapo_steps=self.dapo_steps,
                 head_weights_dict=self.dapo_head_weights)
             total_dapo_loss = total_dapo_loss * self.loss_weights.dapo
@@ -143,6 +156,15 @@ class ReinforcementLoss:
             total_loss = (total_policy_gradient_loss + total_upgo_loss +
                           total_critic_loss + total_entropy_loss + total_kl_loss +
                           action_type_kl_loss + total_dapo_loss)
+        # ONE host sync for all scalar metrics (the reference's per-metric
+        # .item() calls stall the device queue ~40x per step)
+        tensor_keys = [k for k, v in loss_info_dict.items()
+                       if torch.is_tensor(v)]
+        if tensor_keys:
+            flat = torch.stack([loss_info_dict[k].detach().float().reshape(())
+                                for k in tensor_keys]).cpu()
+            for k, val in zip(tensor_keys, flat.tolist()):
+                loss_info_dict[k] = val
         loss_info_dict['total_loss'] = total_loss
         return loss_info_dict
 

@@ -77,8 +77,8 @@ def policy_gradient_loss(baseline_value, reward, target_action_log_probs_dict,
             loss = loss * mask[field + '_mask']
         loss = loss.mean()
         total = total + loss * head_weights_dict[head_type]
-        info[head_type] = loss.item()
-    info['total'] = total.item()
+        info[head_type] = loss.detach()
+    info['total'] = total.detach()
     return total, info
 
 
@@ -97,52 +97,54 @@ def upgo_loss(baseline_value, reward, target_action_log_probs_dict,
             loss = loss * mask[head_type]
         loss = loss.mean()
         total = total + loss * head_weights_dict[head_type]
-        info['upgo/' + head_type] = loss.item()
-    info['upgo/total'] = total.item()
+        info['upgo/' + head_type] = loss.detach()
+    info['upgo/total'] = total.detach()
     return total, info
 
 
-def entropy_loss(target_policy_probs_dict, target_policy_log_probs_dict, mask,
-                 head_weights_dict):
-    """Normalized entropy per head (reference as_rl_utils.py:52-75)."""
+def entropy_loss(entropy_rows_dict, num_classes_dict, mask, head_weights_dict):
+    """Normalized entropy per head (reference as_rl_utils.py:52-75).
+
+    `entropy_rows_dict[head]` is the per-row entropy -(p*logp).sum(-1)
+    (rowwise_entropy — the fused HIP kernel on GPU, eager on CPU), so the
+    (T,B,C) probs/log-probs tensors are never materialized here."""
+    import math as _math
     total = 0.
     info = {}
     for head_type in HEAD_TYPES:
-        ent = -target_policy_probs_dict[head_type] * target_policy_log_probs_dict[head_type]
+        ent = entropy_rows_dict[head_type]
         if head_type == 'selected_units':
-            ent = ent.sum(dim=-1) / (1e-9 + torch.log(
+            ent = ent / (1e-9 + torch.log(
                 mask['selected_units_logits_mask'].float().sum(dim=-1) + 1).unsqueeze(-1))
             ent = (ent * mask['selected_units_mask']).sum(-1)
             ent = ent.div(mask['selected_units_mask'].sum(-1) + 1e-9)
         elif head_type == 'target_unit':
-            ent = ent.sum(dim=-1) / (1e-9 + torch.log(
+            ent = ent / (1e-9 + torch.log(
                 mask['target_units_logits_mask'].float().sum(dim=-1) + 1))
         else:
-            ent = ent.sum(dim=-1) / torch.log(
-                torch.tensor([ent.shape[-1]], dtype=torch.float, device=ent.device))
+            ent = ent / _math.log(num_classes_dict[head_type])
         if head_type not in ('action_type', 'delay'):
             ent = ent * mask['actions_mask'][head_type]
         entropy = ent.mean()
-        info['entropy/' + head_type] = entropy.item()
+        info['entropy/' + head_type] = entropy.detach()
         total = total + (-entropy * head_weights_dict[head_type])
-    info['entropy/total'] = total.item()
+    info['entropy/total'] = total.detach()
     return total, info
 
 
-def kl_loss(target_policy_log_probs_dict, teacher_policy_logits_dict, mask,
-            game_steps, action_type_kl_steps, head_weights_dict):
+def kl_loss(kl_rows_dict, mask, game_steps, action_type_kl_steps,
+            head_weights_dict):
     """KL(teacher || target) per head, plus the early-game extra action-type
-    KL (reference as_rl_utils.py:78-110)."""
+    KL (reference as_rl_utils.py:78-110).
+
+    `kl_rows_dict[head]` is the per-row KL sum (rowwise_kl — fused on GPU),
+    so teacher probs/log-probs are never materialized here."""
     total = 0.
     action_type_kl_loss = torch.tensor(0.)
     info = {}
     for head_type in ['action_type', 'queued', 'delay', 'selected_units',
                       'target_unit', 'target_location']:
-        target_log_probs = target_policy_log_probs_dict[head_type]
-        teacher_logits = teacher_policy_logits_dict[head_type]
-        teacher_log_probs = F.log_softmax(teacher_logits, dim=-1)
-        teacher_probs = torch.exp(teacher_log_probs)
-        kl = (teacher_probs * (teacher_log_probs - target_log_probs)).sum(dim=-1)
+        kl = kl_rows_dict[head_type]
         if head_type == 'selected_units':
             kl = (kl * mask['selected_units_mask']).sum(-1)
         if head_type not in ('action_type', 'delay'):
@@ -151,33 +153,30 @@ def kl_loss(target_policy_log_probs_dict, teacher_policy_logits_dict, mask,
             flag = game_steps < action_type_kl_steps
             action_type_kl = kl * flag * mask['cum_action_mask']
             action_type_kl_loss = action_type_kl.mean()
-            info['kl/extra_at'] = action_type_kl_loss.item()
+            info['kl/extra_at'] = action_type_kl_loss.detach()
         kl_head = kl.mean()
         total = total + kl_head * head_weights_dict[head_type]
-        info['kl/' + head_type] = kl_head.item()
-    info['kl/total'] = total.item()
+        info['kl/' + head_type] = kl_head.detach()
+    info['kl/total'] = total.detach()
     return total, action_type_kl_loss, info
 
 
-def dapo_loss(target_policy_log_probs_dict, successive_policy_logits_dict, mask,
-              game_steps, dapo_steps, head_weights_dict):
+def dapo_loss(dapo_rows_dict, mask, game_steps, dapo_steps,
+              head_weights_dict):
     """KL against the successive model in the early game
-    (reference as_rl_utils.py:112-136)."""
+    (reference as_rl_utils.py:112-136).  `dapo_rows_dict[head]` is the
+    per-row KL(successive || target) sum."""
     total = 0.
     info = {}
     flag = game_steps < dapo_steps
     for head_type in HEAD_TYPES:
-        target_log_probs = target_policy_log_probs_dict[head_type]
-        succ_logits = successive_policy_logits_dict[head_type]
-        succ_log_probs = F.log_softmax(succ_logits, dim=-1)
-        succ_probs = torch.exp(succ_log_probs)
-        kl = (succ_probs * (succ_log_probs - target_log_probs)).sum(dim=-1)
+        kl = dapo_rows_dict[head_type]
         if head_type == 'selected_units':
             kl = (kl * mask['selected_units_mask']).sum(-1)
         if head_type not in ('action_type', 'delay'):
             kl = kl * mask['actions_mask'][head_type]
         kl = (kl * flag).mean()
         total = total + kl * head_weights_dict[head_type]
-        info['dapo/' + head_type] = kl.item()
-    info['dapo/total'] = total.item()
+        info['dapo/' + head_type] = kl.detach()
+    info['dapo/total'] = total.detach()
     return total, info

@@ -11,6 +11,7 @@ import torch
 import torch.nn.functional as F
 
 from ..models.nn.blocks import sequence_mask
+from ..ops.ce_loss import masked_cross_entropy
 from ..parallel.dist import allreduce, get_rank, get_world_size
 from ..utils.config import deep_merge_dicts, read_config
 
@@ -51,6 +52,7 @@ class SupervisedLoss:
         else:
             self.criterion = torch.nn.CrossEntropyLoss(reduction='none')
         self.su_criterion = torch.nn.CrossEntropyLoss(reduction='none')
+        self.label_smooth = self.cfg.get('label_smooth', False)
         self.su_mask = self.cfg.su_mask
         self.cross_rank_loss = self.cfg.get('cross_rank_loss', False)
         self.total_batch_size = None
@@ -78,6 +80,14 @@ class SupervisedLoss:
         return loss_dict
 
     # ----------------------------------------------------------- helpers
+    def _ce(self, logits, labels):
+        """Per-row CE: the fused HIP kernel on GPU (K13 — one pass, no
+        (N,C) log-softmax materialization), eager otherwise; label
+        smoothing keeps the explicit composition."""
+        if self.label_smooth:
+            return self.criterion(logits, labels)
+        return masked_cross_entropy(logits, labels)
+
     def _masked_mean(self, loss_tmp, mask, batch):
         if self.cross_rank_loss:
             loss = loss_tmp.mean()
@@ -88,19 +98,19 @@ class SupervisedLoss:
     def _action_type_loss(self, logits, labels, mask):
         with torch.no_grad():
             acc = (logits.argmax(dim=1) == labels).float().sum() / len(labels)
-        loss_tmp = self.criterion(logits, labels) * mask
+        loss_tmp = self._ce(logits, labels) * mask
         loss = self._masked_mean(loss_tmp, mask, labels.shape[0])
         return {'action_type_loss': loss, 'action_type_acc': acc}
 
     def _delay_loss(self, preds, labels, mask):
-        loss_tmp = self.criterion(preds, labels) * mask
+        loss_tmp = self._ce(preds, labels) * mask
         loss = self._masked_mean(loss_tmp, mask, labels.shape[0])
         with torch.no_grad():
             l1 = ((preds.argmax(dim=-1) - labels).abs() * mask).sum() / (mask.sum() + 1e-6)
         return {'delay_loss': loss, 'delay_distance_L1': l1}
 
     def _queued_loss(self, preds, labels, mask):
-        loss_tmp = self.criterion(preds, labels) * mask
+        loss_tmp = self._ce(preds, labels) * mask
         loss = self._masked_mean(loss_tmp, mask, labels.shape[0])
         with torch.no_grad():
             acc = ((preds.argmax(dim=-1) - labels).abs() * mask).sum() / (mask.sum() + 1e-6)
@@ -124,8 +134,8 @@ class SupervisedLoss:
             ext = ext.masked_fill(~logits_mask.bool(), -1e9)
             logits = ext[:, :, :-1]
         select_mask = sequence_mask(lengths, max_len=s)
-        loss_tmp = self.su_criterion(logits.reshape(-1, n),
-                                     labels[:, :s].reshape(-1)).view(b, s)
+        loss_tmp = masked_cross_entropy(logits.reshape(-1, n),
+                                        labels[:, :s].reshape(-1)).view(b, s)
         loss_tmp = loss_tmp.masked_fill(~select_mask, 0)
         loss_tmp = loss_tmp * mask.unsqueeze(1)
         if self.cross_rank_loss:
@@ -166,7 +176,7 @@ class SupervisedLoss:
         return (inter / (union + 1e-6) * mask).sum() / (mask.sum() + 1e-6)
 
     def _target_unit_loss(self, logits, labels, mask):
-        loss_tmp = self.criterion(logits, labels) * mask
+        loss_tmp = self._ce(logits, labels) * mask
         loss = self._masked_mean(loss_tmp, mask, labels.shape[0])
         with torch.no_grad():
             acc = ((logits.argmax(dim=-1) == labels) * mask).sum() / (mask.sum() + 1e-6)
@@ -174,7 +184,7 @@ class SupervisedLoss:
 
     def _target_location_loss(self, logits, labels, mask):
         W = 160
-        loss_tmp = self.criterion(logits, labels) * mask
+        loss_tmp = self._ce(logits, labels) * mask
         loss = self._masked_mean(loss_tmp, mask, labels.shape[0])
         with torch.no_grad():
             preds = logits.argmax(dim=-1)

@@ -38,12 +38,21 @@ class _FusedMaskedCE(torch.autograd.Function):
 def masked_cross_entropy(logits, labels, mask=None):
     """Per-row CE ``(logsumexp(logits_i) - logits_i[label_i]) * mask_i``.
 
-    logits: (N, C) float; labels: (N,) long; mask: (N,) float or None.
+    logits: (..., C) float; labels: (...,) long; mask: (...,) float or None.
     """
+    if logits.is_cuda:
+        logits = logits.float()
     use_hip = (logits.is_cuda and logits.dtype == torch.float32
                and os.environ.get('DISTAR_AMD_FUSED_CE') == '1')
+    C = logits.shape[-1]
+    shape = labels.shape
     if use_hip:
-        return _FusedMaskedCE.apply(logits.contiguous(), labels.contiguous(),
-                                    None if mask is None else mask.contiguous())
-    loss = torch.nn.functional.cross_entropy(logits, labels, reduction='none')
+        flat = _FusedMaskedCE.apply(
+            logits.contiguous().view(-1, C), labels.contiguous().view(-1),
+            None if mask is None else
+            mask.contiguous().view(-1).float())
+        return flat.view(shape)
+    loss = torch.nn.functional.cross_entropy(
+        logits.reshape(-1, C), labels.reshape(-1),
+        reduction='none').view(shape)
     return loss if mask is None else loss * mask

@@ -55,19 +55,31 @@ def _use_hip(t):
 
 
 def rowwise_entropy(logits):
-    """Per-row Shannon entropy of softmax(logits); (N, C) -> (N,)."""
+    """Per-row Shannon entropy of softmax(logits); (..., C) -> (...)."""
+    if logits.is_cuda:
+        logits = logits.float()
     if _use_hip(logits):
-        return _FusedEntropy.apply(logits.contiguous())
+        shape = logits.shape[:-1]
+        flat = _FusedEntropy.apply(
+            logits.contiguous().view(-1, logits.shape[-1]))
+        return flat.view(shape)
     log_p = F.log_softmax(logits, dim=-1)
     return -(log_p.exp() * log_p).sum(dim=-1)
 
 
 def rowwise_kl(teacher_logits, student_logits):
-    """Per-row KL(softmax(teacher) || softmax(student)); teacher gets no
-    gradient (detached), matching the RL loss semantics."""
+    """Per-row KL(softmax(teacher) || softmax(student)); (..., C) -> (...).
+    Teacher gets no gradient (detached), matching the RL loss semantics."""
+    if student_logits.is_cuda:
+        student_logits = student_logits.float()
+        teacher_logits = teacher_logits.float()
     if _use_hip(student_logits):
-        return _FusedKL.apply(teacher_logits.detach().contiguous(),
-                              student_logits.contiguous())
+        shape = student_logits.shape[:-1]
+        C = student_logits.shape[-1]
+        flat = _FusedKL.apply(
+            teacher_logits.detach().contiguous().view(-1, C),
+            student_logits.contiguous().view(-1, C))
+        return flat.view(shape)
     t_log_p = F.log_softmax(teacher_logits.detach(), dim=-1)
     s_log_p = F.log_softmax(student_logits, dim=-1)
     return (t_log_p.exp() * (t_log_p - s_log_p)).sum(dim=-1)
