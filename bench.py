@@ -187,9 +187,10 @@ def main():
         else torch.device('cpu')
     if has_gpu:
         torch.cuda.set_device(device)
-        # MIOpen find mode: benchmarked conv solvers WITH workspace (the
-        # immediate-mode fallback runs workspace-starved — r01 stderr)
-        torch.backends.cudnn.benchmark = True
+        # NOTE: cudnn.benchmark=True (MIOpen exhaustive find) costs 10+ min
+        # of solver compilation on a fresh box per unique conv shape —
+        # measured r2: it blew an 840 s bench window.  Immediate mode keeps
+        # the r01 conv behavior; the real fix is the hand-written K4 conv.
     use_amp = has_gpu and not args.no_amp
 
     import copy
