@@ -40,7 +40,8 @@ class Attention(nn.Module):
         qkv = self.attention_pre(x)                      # (B, N, 3*H*D)
         use_hip = (x.is_cuda and self.dropout is None and self.head_dim == 128
                    and qkv.dtype == torch.bfloat16
-                   and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1')
+                   and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'
+                   and os.environ.get('DISTAR_AMD_ATTN') != '0')
         if use_hip:
             from ...ops.entity_attn import entity_attention
             entity_num = mask.int().sum(dim=-1) if mask is not None else None

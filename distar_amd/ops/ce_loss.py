@@ -5,9 +5,8 @@ Autograd wrapper over `ops/hip/ce_loss.hip` for the SL location-head loss
 (N, 24320) fp32 tensor).  Forward returns per-row losses like
 ``F.cross_entropy(..., reduction='none') * mask``; callers reduce.
 
-EXPERIMENTAL: GPU numerics validation is scheduled for round 2 — the HIP
-path is only taken with ``DISTAR_AMD_FUSED_CE=1`` (and on CUDA); otherwise
-this is the eager composition.
+Validated on MI355X (round-2 GPU numerics tests); the HIP path is the
+default on CUDA fp32 — set ``DISTAR_AMD_FUSED_CE=0`` to force eager.
 """
 import os
 
@@ -43,7 +42,7 @@ def masked_cross_entropy(logits, labels, mask=None):
     if logits.is_cuda:
         logits = logits.float()
     use_hip = (logits.is_cuda and logits.dtype == torch.float32
-               and os.environ.get('DISTAR_AMD_FUSED_CE') == '1')
+               and os.environ.get('DISTAR_AMD_FUSED_CE', '1') != '0')
     C = logits.shape[-1]
     shape = labels.shape
     if use_hip:

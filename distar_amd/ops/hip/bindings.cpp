@@ -494,6 +494,47 @@ torch::Tensor entity_attn_bwd(
   return dqkv;
 }
 
+extern "C" __global__ void scatter_add_bf16_kernel(
+    const __hip_bfloat16*, const int*, const int*, __hip_bfloat16*,
+    int, int, int, int, int);
+extern "C" __global__ void scatter_add_bf16_bwd_kernel(
+    const __hip_bfloat16*, const int*, __hip_bfloat16*,
+    int, int, int, int, int);
+
+torch::Tensor scatter_add_map(torch::Tensor src, torch::Tensor xy,
+                              c10::optional<torch::Tensor> entity_num,
+                              int64_t H, int64_t W) {
+  TORCH_CHECK(src.is_cuda() && src.is_contiguous() && src.dim() == 3);
+  TORCH_CHECK(src.scalar_type() == torch::kBFloat16, "src must be bf16");
+  TORCH_CHECK(xy.scalar_type() == torch::kInt32 && xy.is_contiguous());
+  int64_t B = src.size(0), N = src.size(1), C = src.size(2);
+  auto out = torch::zeros({B, C, H, W}, src.options());
+  const int* en = nullptr;
+  if (entity_num.has_value()) en = entity_num->data_ptr<int>();
+  long total = B * N * C;
+  int blocks = (int)std::min<long>((total + 255) / 256, 4096);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(scatter_add_bf16_kernel, dim3(blocks), dim3(256), 0,
+                     stream.stream(), bfp(src), xy.data_ptr<int>(), en,
+                     bfp_mut(out), (int)B, (int)N, (int)C, (int)H, (int)W);
+  return out;
+}
+
+torch::Tensor scatter_add_map_bwd(torch::Tensor dout, torch::Tensor xy,
+                                  int64_t N) {
+  TORCH_CHECK(dout.is_cuda() && dout.is_contiguous() && dout.dim() == 4);
+  TORCH_CHECK(dout.scalar_type() == torch::kBFloat16);
+  int64_t B = dout.size(0), C = dout.size(1), H = dout.size(2), W = dout.size(3);
+  auto dsrc = torch::empty({B, N, C}, dout.options());
+  long total = B * N * C;
+  int blocks = (int)std::min<long>((total + 255) / 256, 4096);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(scatter_add_bf16_bwd_kernel, dim3(blocks), dim3(256), 0,
+                     stream.stream(), bfp(dout), xy.data_ptr<int>(),
+                     bfp_mut(dsrc), (int)B, (int)N, (int)C, (int)H, (int)W);
+  return dsrc;
+}
+
 torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor Bm) {
   TORCH_CHECK(A.is_cuda() && A.is_contiguous() && Bm.is_contiguous());
   TORCH_CHECK(A.scalar_type() == torch::kBFloat16);
@@ -512,6 +553,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K1 entity-transformer flash attention backward");
   m.def("mfma_selftest", &mfma_selftest,
         "one 16x16x32 bf16 MFMA: A(16,32) . B(16,32)^T -> (16,16) fp32");
+  m.def("scatter_add_map", &scatter_add_map,
+        "K3 entity->NCHW map scatter-add (packed bf16 atomics)");
+  m.def("scatter_add_map_bwd", &scatter_add_map_bwd,
+        "K3 scatter-add backward (row gather)");
   m.def("entropy_fwd", &entropy_fwd, "fused rowwise entropy forward");
   m.def("entropy_bwd", &entropy_bwd, "fused rowwise entropy backward");
   m.def("kl_fwd", &kl_fwd, "fused rowwise KL(teacher||student) forward");

@@ -5,8 +5,8 @@ and KL terms (reference `rl_training/as_rl_utils.py` entropy/kl; eager
 materializes (N,C) softmax+log_softmax intermediates).  Per-row outputs;
 callers mask/normalize/mean exactly as the eager path does.
 
-EXPERIMENTAL: HIP path only with ``DISTAR_AMD_FUSED_RL_ROWWISE=1`` (and on
-CUDA); GPU numerics validation scheduled for round 2.
+Validated on MI355X (round-2 GPU numerics tests); the HIP path is the
+default on CUDA fp32 — set ``DISTAR_AMD_FUSED_RL_ROWWISE=0`` to force eager.
 """
 import os
 
@@ -51,7 +51,7 @@ class _FusedKL(torch.autograd.Function):
 
 def _use_hip(t):
     return (t.is_cuda and t.dtype == torch.float32
-            and os.environ.get('DISTAR_AMD_FUSED_RL_ROWWISE') == '1')
+            and os.environ.get('DISTAR_AMD_FUSED_RL_ROWWISE', '1') != '0')
 
 
 def rowwise_entropy(logits):

@@ -12,7 +12,28 @@ maps to a single HIP kernel with coalesced row writes.  A hand-written HIP
 kernel (atomic bf16x2 adds staged through LDS) can be slotted in behind the
 same signature; autograd's backward is a row gather with the same index.
 """
+import os
+
 import torch
+
+from . import hip_ext
+
+
+class _ScatterAddMap(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, src, xy, H, W):
+        ops = hip_ext.require()
+        out = ops.scatter_add_map(src, xy, None, H, W)
+        ctx.save_for_backward(xy)
+        ctx.N = src.shape[1]
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        ops = hip_ext.require()
+        (xy,) = ctx.saved_tensors
+        return ops.scatter_add_map_bwd(dout.contiguous(), xy, ctx.N), \
+            None, None, None
 
 
 def scatter_connection(shape, project_embeddings, entity_location, scatter_dim,
@@ -30,6 +51,15 @@ def scatter_connection(shape, project_embeddings, entity_location, scatter_dim,
     N = project_embeddings.shape[1]
     C = scatter_dim
     device = project_embeddings.device
+    if (scatter_type == 'add' and project_embeddings.is_cuda
+            and project_embeddings.dtype == torch.bfloat16
+            and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'):
+        # K3 HIP kernel: direct NCHW scatter with packed-bf16 atomics —
+        # no (B*HW, C) staging buffer, no permute().contiguous() pass
+        xy = torch.stack([entity_location[..., 0].clamp(0, W - 1),
+                          entity_location[..., 1].clamp(0, H - 1)],
+                         dim=-1).to(torch.int32).contiguous()
+        return _ScatterAddMap.apply(project_embeddings.contiguous(), xy, H, W)
     x = entity_location[..., 0].long().clamp_(0, W - 1)
     y = entity_location[..., 1].long().clamp_(0, H - 1)
     bias = (torch.arange(B, device=device) * (H * W)).unsqueeze(1)

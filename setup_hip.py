@@ -28,6 +28,7 @@ SOURCES = [
     os.path.join(HIP_DIR, 'ce_loss.hip'),
     os.path.join(HIP_DIR, 'rl_rowwise.hip'),
     os.path.join(HIP_DIR, 'entity_attn.hip'),
+    os.path.join(HIP_DIR, 'scatter.hip'),
     os.path.join(HIP_DIR, 'bindings.cpp'),
 ]
 

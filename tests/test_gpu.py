@@ -283,8 +283,6 @@ def test_su_sample_kernel_semantics():
 
 
 @pytest.mark.gpu
-@pytest.mark.skipif(os.environ.get('DISTAR_AMD_EXPERIMENTAL') != '1',
-                    reason='experimental kernel, round-2 validation pending')
 def test_fused_masked_ce_matches_eager():
     """K13 fused masked CE vs F.cross_entropy fp32 (forward + backward)."""
     import distar_amd.ops.ce_loss as ce
@@ -297,7 +295,7 @@ def test_fused_masked_ce_matches_eager():
     ref = torch.nn.functional.cross_entropy(ref_l, labels,
                                             reduction='none') * mask
     ref.sum().backward()
-    os.environ['DISTAR_AMD_FUSED_CE'] = '1'
+    os.environ['DISTAR_AMD_FUSED_CE'] = '1'   # explicit (also the default)
     try:
         fused_l = logits.detach().clone().requires_grad_(True)
         out = ce.masked_cross_entropy(fused_l, labels, mask)
@@ -310,8 +308,6 @@ def test_fused_masked_ce_matches_eager():
 
 
 @pytest.mark.gpu
-@pytest.mark.skipif(os.environ.get('DISTAR_AMD_EXPERIMENTAL') != '1',
-                    reason='experimental kernel, round-2 validation pending')
 def test_fused_rowwise_entropy_kl_match_eager():
     """K13 fused entropy/KL vs eager fp32 (forward + backward)."""
     from distar_amd.ops.rl_rowwise import rowwise_entropy, rowwise_kl
@@ -319,7 +315,8 @@ def test_fused_rowwise_entropy_kl_match_eager():
     N, C = 64, 24320
     t = torch.randn(N, C, device='cuda') * 2
     s_ref = (torch.randn(N, C, device='cuda') * 2).requires_grad_(True)
-    ent_ref = rowwise_entropy(s_ref)            # eager (env off)
+    os.environ['DISTAR_AMD_FUSED_RL_ROWWISE'] = '0'
+    ent_ref = rowwise_entropy(s_ref)            # eager reference
     kl_ref = rowwise_kl(t, s_ref)
     (ent_ref.sum() + kl_ref.sum()).backward()
     os.environ['DISTAR_AMD_FUSED_RL_ROWWISE'] = '1'
@@ -438,3 +435,31 @@ def test_entity_transformer_hip_vs_eager_end_to_end():
     m = mask.unsqueeze(-1)
     torch.testing.assert_close((out_hip * m).float(), (out_eager * m).float(),
                                rtol=5e-2, atol=5e-2)
+
+
+def test_scatter_add_map_matches_eager():
+    """K3 HIP NCHW scatter-add (packed bf16 atomics) vs the eager index_add
+    formulation, forward + backward."""
+    from distar_amd.ops.scatter import scatter_connection
+    torch.manual_seed(3)
+    B, N, C, H, W = 6, 512, 32, 152, 160
+    emb0 = (torch.randn(B, N, C, device='cuda') * 0.5).bfloat16()
+    loc = torch.stack([torch.randint(0, W, (B, N), device='cuda'),
+                       torch.randint(0, H, (B, N), device='cuda')], dim=-1)
+    # force some collisions
+    loc[:, 1] = loc[:, 0]
+    emb_h = emb0.detach().clone().requires_grad_(True)
+    out_h = scatter_connection((B, H, W), emb_h, loc, C, 'add')
+    dout = (torch.randn_like(out_h) * 0.5).bfloat16()
+    out_h.backward(dout)
+    os.environ['DISTAR_AMD_DISABLE_HIP'] = '1'
+    try:
+        emb_e = emb0.detach().clone().requires_grad_(True)
+        out_e = scatter_connection((B, H, W), emb_e, loc, C, 'add')
+        out_e.backward(dout)
+    finally:
+        os.environ.pop('DISTAR_AMD_DISABLE_HIP', None)
+    torch.testing.assert_close(out_h.float(), out_e.float(),
+                               rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(emb_h.grad.float(), emb_e.grad.float(),
+                               rtol=2e-2, atol=2e-2)
