@@ -100,11 +100,12 @@ class TransformerLayer(nn.Module):
             a = self.attention(x, mask)
             if self.dropout is not None:
                 a = self.dropout(a)
-            x = self.layernorm1(x + a)
+            from ...ops.residual_ln import fused_residual_ln
+            x = fused_residual_ln(a, x, self.layernorm1)
             m = self.mlp(x)
             if self.dropout is not None:
                 m = self.dropout(m)
-            x = self.layernorm2(x + m)
+            x = fused_residual_ln(m, x, self.layernorm2)
         elif self.ln_type == 'pre':
             a = self.attention(self.layernorm1(x), mask)
             if self.dropout is not None:

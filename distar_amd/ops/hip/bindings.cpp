@@ -470,8 +470,7 @@ torch::Tensor entity_attn_bwd(
                      drow.data_ptr<float>(), (int)B, (int)H, (int)N,
                      (long)(N * HD), (long)HD);
   dim3 grid((N + 63) / 64, H, B);
-  size_t lds_kv = 4 * 64 * 256 + 2 * 128 * 128 + 64 * 128 +
-                  2 * 64 * sizeof(float);
+  size_t lds_kv = 4 * 64 * 256 + 64 * 128 + 2 * 64 * sizeof(float);
   hipLaunchKernelGGL(entity_attn_bwd_kv_kernel, grid, dim3(256), lds_kv,
                      stream.stream(),
                      bfp(qkv), bfp(qkv) + HD, bfp(qkv) + 2 * HD,
@@ -481,7 +480,7 @@ torch::Tensor entity_attn_bwd(
                      (long)(N * F), (long)F, 128L,
                      (long)(N * HD), (long)HD,
                      (long)(N * F), (long)F);
-  size_t lds_q = 4 * 64 * 256 + 128 * 128 + 64 * 128 + 2 * 64 * sizeof(float);
+  size_t lds_q = 4 * 64 * 256 + 64 * 128 + 2 * 64 * sizeof(float);
   hipLaunchKernelGGL(entity_attn_bwd_q_kernel, grid, dim3(256), lds_q,
                      stream.stream(),
                      bfp(qkv), bfp(qkv) + HD, bfp(qkv) + 2 * HD,
@@ -535,6 +534,63 @@ torch::Tensor scatter_add_map_bwd(torch::Tensor dout, torch::Tensor xy,
   return dsrc;
 }
 
+extern "C" __global__ void residual_ln_fwd_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, const float*, const float*,
+    __hip_bfloat16*, __hip_bfloat16*, float*, float*, long, int, float);
+extern "C" __global__ void residual_ln_bwd_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, const float*, const float*,
+    const float*, __hip_bfloat16*, float*, float*, long, int);
+
+std::vector<torch::Tensor> residual_ln_fwd(
+    torch::Tensor x, c10::optional<torch::Tensor> a, torch::Tensor w,
+    torch::Tensor b, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "x must be bf16");
+  int64_t C = x.size(-1);
+  TORCH_CHECK(C % 64 == 0 && C <= 1024, "C must be 64..1024, mult of 64");
+  long R = x.numel() / C;
+  auto y = torch::empty_like(x);
+  auto s = torch::empty_like(x);
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto mean = torch::empty({R}, fopt);
+  auto rstd = torch::empty({R}, fopt);
+  const __hip_bfloat16* ap = nullptr;
+  if (a.has_value()) {
+    TORCH_CHECK(a->is_contiguous() && a->scalar_type() == torch::kBFloat16);
+    ap = bfp(*a);
+  }
+  int blocks = (int)std::min<long>((R + 3) / 4, 4096);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(residual_ln_fwd_kernel, dim3(blocks), dim3(256), 0,
+                     stream.stream(), bfp(x), ap,
+                     w.data_ptr<float>(), b.data_ptr<float>(),
+                     bfp_mut(y), bfp_mut(s), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(), R, (int)C, (float)eps);
+  return {y, s, mean, rstd};
+}
+
+std::vector<torch::Tensor> residual_ln_bwd(
+    torch::Tensor dy, torch::Tensor s, torch::Tensor mean,
+    torch::Tensor rstd, torch::Tensor w) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous());
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16);
+  int64_t C = dy.size(-1);
+  long R = dy.numel() / C;
+  auto dsum = torch::empty_like(dy);
+  auto fopt = dy.options().dtype(torch::kFloat32);
+  auto dw = torch::zeros({C}, fopt);
+  auto db = torch::zeros({C}, fopt);
+  int blocks = (int)std::min<long>((R + 3) / 4, 2048);
+  size_t lds = 2 * C * sizeof(float);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(residual_ln_bwd_kernel, dim3(blocks), dim3(256), lds,
+                     stream.stream(), bfp(dy), bfp(s),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     w.data_ptr<float>(), bfp_mut(dsum),
+                     dw.data_ptr<float>(), db.data_ptr<float>(), R, (int)C);
+  return {dsum, dw, db};
+}
+
 torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor Bm) {
   TORCH_CHECK(A.is_cuda() && A.is_contiguous() && Bm.is_contiguous());
   TORCH_CHECK(A.scalar_type() == torch::kBFloat16);
@@ -557,6 +613,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K3 entity->NCHW map scatter-add (packed bf16 atomics)");
   m.def("scatter_add_map_bwd", &scatter_add_map_bwd,
         "K3 scatter-add backward (row gather)");
+  m.def("residual_ln_fwd", &residual_ln_fwd,
+        "fused residual-add + LayerNorm forward (bf16, fp32 stats)");
+  m.def("residual_ln_bwd", &residual_ln_bwd,
+        "fused residual-add + LayerNorm backward");
   m.def("entropy_fwd", &entropy_fwd, "fused rowwise entropy forward");
   m.def("entropy_bwd", &entropy_bwd, "fused rowwise entropy backward");
   m.def("kl_fwd", &kl_fwd, "fused rowwise KL(teacher||student) forward");

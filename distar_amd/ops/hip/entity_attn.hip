@@ -60,6 +60,18 @@ __device__ __forceinline__ bf16x8 lds_read8(const char* base, int byte_off) {
   return *(const bf16x8*)(base + byte_off);
 }
 
+// 8 elements T[row0+j][col] of a row-major [..][128] swizzled tile —
+// the transposed-operand fragment (strided u16 reads; saves staging a
+// second, transposed copy of the tile and 32 KB of LDS per block).
+__device__ __forceinline__ bf16x8 lds_read_col8(const char* base, int row0,
+                                                int col) {
+  bf16x8 out;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    out[j] = *(const __hip_bfloat16*)(base + swz_off(row0 + j, col));
+  return out;
+}
+
 // stage a [rows][128] bf16 tile from global (row stride `gstride` elems,
 // element base `gbase`) into swizzled LDS; rows >= limit are zero-filled.
 __device__ __forceinline__ void stage_tile128(
@@ -245,7 +257,7 @@ extern "C" __global__ void attn_drow_kernel(
 }
 
 // dK/dV: one block per (b, h, kv-tile); loops q-tiles.
-extern "C" __global__ __launch_bounds__(256, 1)
+extern "C" __global__ __launch_bounds__(256, 2)
 void entity_attn_bwd_kv_kernel(
     const __hip_bfloat16* __restrict__ q,
     const __hip_bfloat16* __restrict__ k,
@@ -273,10 +285,8 @@ void entity_attn_bwd_kv_kernel(
   char* Ks  = lds;                      // [64][128] 16 KB
   char* Vs  = Ks + TILE * 256;          // [64][128] 16 KB
   char* Qs  = Vs + TILE * 256;          // [64][128] 16 KB (per q-tile)
-  char* Qt  = Qs + TILE * 256;          // [128][64] 16 KB
-  char* dOs = Qt + D_DIM * 128;         // [64][128] 16 KB
-  char* dOt = dOs + TILE * 256;         // [128][64] 16 KB
-  char* Ps  = dOt + D_DIM * 128;        // [64][64]   8 KB (P', then dS')
+  char* dOs = Qs + TILE * 256;          // [64][128] 16 KB
+  char* Ps  = dOs + TILE * 256;         // [64][64]   8 KB (P', then dS')
   float* lse_s  = (float*)(Ps + TILE * 128);   // [64]
   float* drow_s = lse_s + TILE;                // [64]
 
@@ -310,9 +320,7 @@ void entity_attn_bwd_kv_kernel(
     const int q0 = qt * TILE;
     __syncthreads();
     stage_tile128(Qs, qg, n_stride, q0, TILE, N);
-    stage_tile_t(Qt, qg, n_stride, q0, N);
     stage_tile128(dOs, dog, on_stride, q0, TILE, N);
-    stage_tile_t(dOt, dog, on_stride, q0, N);
     for (int i = tid; i < TILE; i += 256) {
       lse_s[i] = (q0 + i < N) ? lse_g[q0 + i] : 0.f;
       drow_s[i] = (q0 + i < N) ? drow_g[q0 + i] : 0.f;
@@ -353,7 +361,7 @@ void entity_attn_bwd_kv_kernel(
     for (int ks = 0; ks < 2; ++ks) {
       bf16x8 bp = lds_read8(Ps, swz_off64(band + l16, ks * 32 + lq * 8));
       for (int t = 0; t < 8; ++t) {
-        bf16x8 a = lds_read8(dOt, swz_off64(t * 16 + l16, ks * 32 + lq * 8));
+        bf16x8 a = lds_read_col8(dOs, ks * 32 + lq * 8, t * 16 + l16);
         acc_dvT[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bp,
                                                              acc_dvT[t], 0, 0, 0);
       }
@@ -380,11 +388,12 @@ void entity_attn_bwd_kv_kernel(
       }
     }
     __builtin_amdgcn_s_waitcnt(0);
-    // dK += dS' . Q : m=key band, n=d 8 tiles, k=q 2 steps (B from Qt)
+    // dK += dS' . Q : m=key band, n=d 8 tiles, k=q 2 steps (B = Q^T via
+    // strided column reads of the row-major Q tile)
     for (int ks = 0; ks < 2; ++ks) {
       bf16x8 a = lds_read8(Ps, swz_off64(band + l16, ks * 32 + lq * 8));
       for (int t = 0; t < 8; ++t) {
-        bf16x8 bq = lds_read8(Qt, swz_off64(t * 16 + l16, ks * 32 + lq * 8));
+        bf16x8 bq = lds_read_col8(Qs, ks * 32 + lq * 8, t * 16 + l16);
         acc_dk[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bq,
                                                             acc_dk[t], 0, 0, 0);
       }
@@ -419,7 +428,7 @@ void entity_attn_bwd_kv_kernel(
 }
 
 // dQ: one block per (b, h, q-tile); loops kv-tiles.
-extern "C" __global__ __launch_bounds__(256, 1)
+extern "C" __global__ __launch_bounds__(256, 2)
 void entity_attn_bwd_q_kernel(
     const __hip_bfloat16* __restrict__ q,
     const __hip_bfloat16* __restrict__ k,
@@ -443,8 +452,7 @@ void entity_attn_bwd_q_kernel(
   char* Qs  = lds;                      // [64][128] 16 KB
   char* dOs = Qs + TILE * 256;          // [64][128] 16 KB
   char* Ks  = dOs + TILE * 256;         // [64][128] 16 KB
-  char* Kt  = Ks + TILE * 256;          // [128][64] 16 KB
-  char* Vs  = Kt + D_DIM * 128;         // [64][128] 16 KB
+  char* Vs  = Ks + TILE * 256;          // [64][128] 16 KB
   char* Ps  = Vs + TILE * 256;          // [64][64]   8 KB
   float* lse_s  = (float*)(Ps + TILE * 128);
   float* drow_s = lse_s + TILE;
@@ -481,7 +489,6 @@ void entity_attn_bwd_q_kernel(
     const int kv0 = kt * TILE;
     __syncthreads();
     stage_tile128(Ks, kg, n_stride, kv0, TILE, N);
-    stage_tile_t(Kt, kg, n_stride, kv0, N);
     stage_tile128(Vs, vg, n_stride, kv0, TILE, N);
     __syncthreads();
 
@@ -522,11 +529,12 @@ void entity_attn_bwd_q_kernel(
       }
     }
     __builtin_amdgcn_s_waitcnt(0);
-    // dQ band += dS . K : m=q band, n=d 8 tiles, k=key 2 steps (B from Kt)
+    // dQ band += dS . K : m=q band, n=d 8 tiles, k=key 2 steps (B = K^T
+    // via strided column reads of the row-major K tile)
     for (int ks = 0; ks < 2; ++ks) {
       bf16x8 a = lds_read8(Ps, swz_off64(band + l16, ks * 32 + lq * 8));
       for (int t = 0; t < 8; ++t) {
-        bf16x8 bk = lds_read8(Kt, swz_off64(t * 16 + l16, ks * 32 + lq * 8));
+        bf16x8 bk = lds_read_col8(Ks, ks * 32 + lq * 8, t * 16 + l16);
         acc_dq[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bk,
                                                             acc_dq[t], 0, 0, 0);
       }

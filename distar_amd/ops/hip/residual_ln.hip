@@ -1,0 +1,126 @@
+// Fused residual-add + LayerNorm, bf16 in/out, fp32 row stats.
+//
+// The entity transformer's post-LN sites (x = LN(x + a), twice per layer —
+// reference module_utils.py:152-199) cost four eager passes each under
+// autocast: bf16 add, bf16->fp32 cast (autocast runs LN in fp32), the LN
+// kernel, and the fp32->bf16 cast at the next matmul.  This kernel does one
+// pass: s = x + a (bf16 read, fp32 math), y = (s - mean) * rstd * w + b,
+// written back as bf16, saving s/mean/rstd for the backward.
+//
+// Rows are C<=1024 bf16 elements; one wave per row, 4 rows per block.
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+typedef __attribute__((ext_vector_type(4))) short short4v;
+
+__device__ __forceinline__ float warp_sum(float v) {
+  for (int off = 1; off < 64; off <<= 1)
+    v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+// C must be a multiple of 4 (256 / 1024 in practice); EPW = elems per lane.
+extern "C" __global__ void residual_ln_fwd_kernel(
+    const __hip_bfloat16* __restrict__ x,
+    const __hip_bfloat16* __restrict__ a,      // nullptr: plain LN
+    const float* __restrict__ w, const float* __restrict__ b,
+    __hip_bfloat16* __restrict__ y,
+    __hip_bfloat16* __restrict__ s_out,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out,
+    long R, int C, float eps) {
+  const int lane = threadIdx.x & 63;
+  const int epw = C / 64;                       // elems per lane (4 or 16)
+  for (long row = blockIdx.x * 4 + (threadIdx.x >> 6); row < R;
+       row += (long)gridDim.x * 4) {
+    const __hip_bfloat16* xr = x + row * C;
+    const __hip_bfloat16* ar = a ? a + row * C : nullptr;
+    float vals[16];
+    __hip_bfloat16 xv[16], av[16];
+    __builtin_memcpy(xv, xr + lane * epw, epw * 2);   // one vector load
+    if (ar) __builtin_memcpy(av, ar + lane * epw, epw * 2);
+    float sum = 0.f;
+    for (int i = 0; i < epw; ++i) {
+      float v = __bfloat162float(xv[i]);
+      if (ar) v += __bfloat162float(av[i]);
+      vals[i] = v;
+      sum += v;
+    }
+    float mean = warp_sum(sum) / C;
+    float var = 0.f;
+    for (int i = 0; i < epw; ++i) {
+      float d = vals[i] - mean;
+      var += d * d;
+    }
+    var = warp_sum(var) / C;
+    float rstd = rsqrtf(var + eps);
+    __hip_bfloat16 yv[16], sv[16];
+    for (int i = 0; i < epw; ++i) {
+      int c = lane * epw + i;
+      sv[i] = __float2bfloat16(vals[i]);
+      yv[i] = __float2bfloat16((vals[i] - mean) * rstd * w[c] + b[c]);
+    }
+    __builtin_memcpy(y + row * C + lane * epw, yv, epw * 2);
+    __builtin_memcpy(s_out + row * C + lane * epw, sv, epw * 2);
+    if (lane == 0) {
+      mean_out[row] = mean;
+      rstd_out[row] = rstd;
+    }
+  }
+}
+
+// dsum = (w*dy - mean(w*dy) - xhat * mean(w*dy*xhat)) * rstd
+// dw += sum_rows dy * xhat ; db += sum_rows dy     (fp32 atomics)
+extern "C" __global__ void residual_ln_bwd_kernel(
+    const __hip_bfloat16* __restrict__ dy,
+    const __hip_bfloat16* __restrict__ s,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const float* __restrict__ w,
+    __hip_bfloat16* __restrict__ dsum,
+    float* __restrict__ dw, float* __restrict__ db,
+    long R, int C) {
+  const int lane = threadIdx.x & 63;
+  const int epw = C / 64;
+  extern __shared__ float lds[];                // dw/db partials: 2*C floats
+  float* dw_part = lds;
+  float* db_part = lds + C;
+  for (int i = threadIdx.x; i < C; i += blockDim.x) {
+    dw_part[i] = 0.f;
+    db_part[i] = 0.f;
+  }
+  __syncthreads();
+  for (long row = blockIdx.x * 4 + (threadIdx.x >> 6); row < R;
+       row += (long)gridDim.x * 4) {
+    const __hip_bfloat16* dyr = dy + row * C;
+    const __hip_bfloat16* sr = s + row * C;
+    float m = mean[row], r = rstd[row];
+    float g[16], xh[16];
+    __hip_bfloat16 dyv_v[16], sv_v[16];
+    __builtin_memcpy(dyv_v, dyr + lane * epw, epw * 2);
+    __builtin_memcpy(sv_v, sr + lane * epw, epw * 2);
+    float s1 = 0.f, s2 = 0.f;
+    for (int i = 0; i < epw; ++i) {
+      int c = lane * epw + i;
+      float dyv = __bfloat162float(dyv_v[i]);
+      float xhat = (__bfloat162float(sv_v[i]) - m) * r;
+      float wg = w[c] * dyv;
+      g[i] = wg;
+      xh[i] = xhat;
+      s1 += wg;
+      s2 += wg * xhat;
+      // per-block LN-param partials (LDS, fp32)
+      atomicAdd(&dw_part[c], dyv * xhat);
+      atomicAdd(&db_part[c], dyv);
+    }
+    s1 = warp_sum(s1) / C;
+    s2 = warp_sum(s2) / C;
+    __hip_bfloat16 drv[16];
+    for (int i = 0; i < epw; ++i)
+      drv[i] = __float2bfloat16((g[i] - s1 - xh[i] * s2) * r);
+    __builtin_memcpy(dsum + row * C + lane * epw, drv, epw * 2);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < C; i += blockDim.x) {
+    atomicAdd(&dw[i], dw_part[i]);
+    atomicAdd(&db[i], db_part[i]);
+  }
+}

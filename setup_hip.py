@@ -29,6 +29,7 @@ SOURCES = [
     os.path.join(HIP_DIR, 'rl_rowwise.hip'),
     os.path.join(HIP_DIR, 'entity_attn.hip'),
     os.path.join(HIP_DIR, 'scatter.hip'),
+    os.path.join(HIP_DIR, 'residual_ln.hip'),
     os.path.join(HIP_DIR, 'bindings.cpp'),
 ]
 

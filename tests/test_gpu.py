@@ -463,3 +463,41 @@ def test_scatter_add_map_matches_eager():
                                rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(emb_h.grad.float(), emb_e.grad.float(),
                                rtol=2e-2, atol=2e-2)
+
+
+def test_fused_residual_ln_matches_eager():
+    """Fused residual-add + LayerNorm vs fp32 eager (fwd + bwd, incl. the
+    LN parameter grads accumulated via atomics)."""
+    from distar_amd.ops.residual_ln import fused_residual_ln
+    torch.manual_seed(4)
+    R, C = 3000, 256
+    ln = torch.nn.LayerNorm(C).cuda()
+    x0 = (torch.randn(R, C, device='cuda') * 0.7).bfloat16()
+    a0 = (torch.randn(R, C, device='cuda') * 0.7).bfloat16()
+    dout = (torch.randn(R, C, device='cuda') * 0.5).bfloat16()
+
+    x_h = x0.detach().clone().requires_grad_(True)
+    a_h = a0.detach().clone().requires_grad_(True)
+    y_h = fused_residual_ln(x_h, a_h, ln)
+    y_h.backward(dout)
+    gw_h, gb_h = ln.weight.grad.clone(), ln.bias.grad.clone()
+    ln.weight.grad = None
+    ln.bias.grad = None
+
+    x_e = x0.detach().clone().float().requires_grad_(True)
+    a_e = a0.detach().clone().float().requires_grad_(True)
+    y_e = torch.nn.functional.layer_norm(x_e + a_e, (C,), ln.weight, ln.bias,
+                                         ln.eps)
+    y_e.backward(dout.float())
+
+    torch.testing.assert_close(y_h.float(), y_e.detach(), rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(x_h.grad.float(), x_e.grad, rtol=5e-2,
+                               atol=5e-2)
+    torch.testing.assert_close(a_h.grad.float(), a_e.grad, rtol=5e-2,
+                               atol=5e-2)
+    # parameter grads sum over 3000 rows: scale tolerance to magnitude
+    tol = ln.weight.grad.abs().max().item() if False else None
+    torch.testing.assert_close(gw_h, ln.weight.grad, rtol=2e-2,
+                               atol=0.02 * float(ln.weight.grad.abs().max()))
+    torch.testing.assert_close(gb_h, ln.bias.grad, rtol=2e-2,
+                               atol=0.02 * float(ln.bias.grad.abs().max()))
