@@ -222,7 +222,8 @@ class SpatialEncoder(nn.Module):
             if self.cfg.downsample_type == 'avgpool':
                 out = torch.nn.functional.avg_pool2d(out, 2, 2)
             elif self.cfg.downsample_type == 'maxpool':
-                out = torch.nn.functional.max_pool2d(out, 2, 2)
+                from ...ops.conv2d import max_pool2x2
+                out = max_pool2x2(out)
             out = self.downsample[i](out)
         for block in self.res:
             map_skip.append(out)
@@ -419,9 +420,10 @@ class ValueEncoder(nn.Module):
         sp = self.cfg.spatial
         self.project = conv2d_block(sp.input_dim, sp.project_dim, 1, 1, 0, activation=self.act)
         dims = [sp.project_dim] + list(sp.down_channels)
+        from ...ops.conv2d import MaxPool2x2HIP
         down_layers = []
         for i in range(len(sp.down_channels)):
-            down_layers.append(nn.MaxPool2d(2, 2))
+            down_layers.append(MaxPool2x2HIP())
             down_layers.append(conv2d_block(dims[i], dims[i + 1], 3, 1, 1, activation=self.act))
         self.downsample = nn.Sequential(*down_layers)
         self.resblock_num = sp.resblock_num

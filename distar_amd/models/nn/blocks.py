@@ -123,7 +123,8 @@ def conv2d_block(in_channels, out_channels, kernel_size, stride=1, padding=0,
     elif pad_type == 'replication':
         block.append(nn.ReplicationPad2d(padding))
         padding = 0
-    conv = nn.Conv2d(in_channels, out_channels, kernel_size, stride,
+    from ...ops.conv2d import Conv2dHIP
+    conv = Conv2dHIP(in_channels, out_channels, kernel_size, stride,
                      padding=padding, dilation=dilation, groups=groups)
     _weight_init(conv.weight, init_type, activation)
     block.append(conv)

@@ -1,0 +1,132 @@
+"""K4: NCHW bf16 conv + 2x2 maxpool through the hand-written MFMA kernels.
+
+`Conv2dHIP` is a drop-in nn.Conv2d (same parameters/state_dict) whose
+forward dispatches stride-1 1x1/3x3 convs on GPU bf16 inputs to
+`ops/hip/conv2d.hip`: implicit-GEMM with in-kernel im2col staging (no NHWC
+transposes, no MIOpen workspace), backward-data through the same kernel
+with flipped/transposed weights, backward-weight through the wgrad kernel.
+`MaxPool2x2HIP` replaces nn.MaxPool2d(2, 2) with an argmax-saving fwd and a
+gather (atomics-free) backward.
+"""
+import os
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from . import hip_ext
+
+
+def _kpad(k):
+    return (k + 63) // 64 * 64
+
+
+def _pack_weight(w, KH, KW):
+    """(Cout, Cin, KH, KW) -> (Cout, Kpad) bf16 contiguous."""
+    Cout = w.shape[0]
+    flat = w.detach().reshape(Cout, -1).to(torch.bfloat16)
+    K = flat.shape[1]
+    Kp = _kpad(K)
+    if Kp != K:
+        flat = F.pad(flat, (0, Kp - K))
+    return flat.contiguous()
+
+
+def _pack_weight_flipped(w, KH, KW):
+    """Backward-data weights: W'[ci][co][KH-1-dy][KW-1-dx] packed."""
+    wf = w.detach().flip(2, 3).permute(1, 0, 2, 3)
+    return _pack_weight(wf, KH, KW)
+
+
+class _Conv2dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, KH, KW, padH, padW):
+        ops = hip_ext.require()
+        wp = _pack_weight(weight, KH, KW)
+        b32 = bias.detach().float().contiguous() if bias is not None else None
+        out = ops.conv2d_fwd(x, wp, b32, weight.shape[0], KH, KW, padH, padW,
+                             False)
+        ctx.save_for_backward(x, weight)
+        ctx.dims = (KH, KW, padH, padW)
+        ctx.has_bias = bias is not None
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        ops = hip_ext.require()
+        x, weight = ctx.saved_tensors
+        KH, KW, padH, padW = ctx.dims
+        dy = dy.contiguous().to(torch.bfloat16)
+        Cout, Cin = weight.shape[0], weight.shape[1]
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            wpf = _pack_weight_flipped(weight, KH, KW)
+            dx = ops.conv2d_fwd(dy, wpf, None, Cin, KH, KW,
+                                KH - 1 - padH, KW - 1 - padW, False)
+        if ctx.needs_input_grad[1]:
+            K = Cin * KH * KW
+            dwp = ops.conv2d_wgrad(x, dy, KH, KW, padH, padW, _kpad(K))
+            dw = dwp[:K].view(Cin, KH, KW, Cout).permute(3, 0, 1, 2) \
+                .contiguous().to(weight.dtype)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = dy.sum(dim=(0, 2, 3)).to(weight.dtype)
+        return dx, dw, db, None, None, None, None
+
+
+class Conv2dHIP(nn.Conv2d):
+    def _use_hip(self, x):
+        kh, kw = self.kernel_size
+        return (x.is_cuda and x.dtype == torch.bfloat16
+                and self.stride == (1, 1) and self.dilation == (1, 1)
+                and self.groups == 1 and (kh, kw) in ((1, 1), (3, 3))
+                and self.padding == (kh // 2, kw // 2)
+                and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'
+                and os.environ.get('DISTAR_AMD_CONV') != '0')
+
+    def forward(self, x):
+        if self._use_hip(x):
+            kh, kw = self.kernel_size
+            return _Conv2dFn.apply(x.contiguous(), self.weight, self.bias,
+                                   kh, kw, kh // 2, kw // 2)
+        return super().forward(x)
+
+
+class _MaxPool2x2Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ops = hip_ext.require()
+        out, idx = ops.maxpool2x2_fwd(x)
+        ctx.save_for_backward(idx)
+        ctx.hw = (x.shape[2], x.shape[3])
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        ops = hip_ext.require()
+        (idx,) = ctx.saved_tensors
+        H, W = ctx.hw
+        return ops.maxpool2x2_bwd(dy.contiguous().to(torch.bfloat16), idx,
+                                  H, W)
+
+
+def max_pool2x2(x):
+    """Functional 2x2/2 maxpool through the HIP kernels when possible."""
+    if (x.is_cuda and x.dtype == torch.bfloat16
+            and x.shape[2] % 2 == 0 and x.shape[3] % 2 == 0
+            and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'
+            and os.environ.get('DISTAR_AMD_CONV') != '0'):
+        return _MaxPool2x2Fn.apply(x.contiguous())
+    return F.max_pool2d(x, 2, 2)
+
+
+class MaxPool2x2HIP(nn.MaxPool2d):
+    def __init__(self):
+        super().__init__(2, 2)
+
+    def forward(self, x):
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and x.shape[2] % 2 == 0 and x.shape[3] % 2 == 0
+                and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'
+                and os.environ.get('DISTAR_AMD_CONV') != '0'):
+            return _MaxPool2x2Fn.apply(x.contiguous())
+        return super().forward(x)

@@ -591,6 +591,99 @@ std::vector<torch::Tensor> residual_ln_bwd(
   return {dsum, dw, db};
 }
 
+extern "C" __global__ void conv2d_fwd_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+    __hip_bfloat16*, int, int, int, int, int, int, int, int, int, int, int);
+extern "C" __global__ void conv2d_wgrad_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, float*,
+    int, int, int, int, int, int, int, int, int, int, int);
+extern "C" __global__ void maxpool2x2_fwd_kernel(
+    const __hip_bfloat16*, __hip_bfloat16*, unsigned char*, long, int, int);
+extern "C" __global__ void maxpool2x2_bwd_kernel(
+    const __hip_bfloat16*, const unsigned char*, __hip_bfloat16*,
+    long, int, int);
+
+torch::Tensor conv2d_fwd(torch::Tensor input, torch::Tensor wp,
+                         c10::optional<torch::Tensor> bias,
+                         int64_t Cout, int64_t KH, int64_t KW,
+                         int64_t padH, int64_t padW, bool relu) {
+  TORCH_CHECK(input.is_cuda() && input.is_contiguous() && input.dim() == 4);
+  TORCH_CHECK(input.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(wp.is_contiguous() && wp.scalar_type() == torch::kBFloat16);
+  int64_t B = input.size(0), Cin = input.size(1);
+  int64_t H = input.size(2), W = input.size(3);
+  int64_t Kpad = wp.size(1);
+  auto out = torch::empty({B, Cout, H, W}, input.options());
+  const float* bp = nullptr;
+  if (bias.has_value()) bp = bias->data_ptr<float>();
+  int HW = (int)(H * W);
+  dim3 grid((HW + 63) / 64, (unsigned)((Cout + 63) / 64), (unsigned)B);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(conv2d_fwd_kernel, grid, dim3(256), 64 * 128,
+                     stream.stream(), bfp(input), bfp(wp), bp, bfp_mut(out),
+                     (int)B, (int)Cin, (int)Cout, (int)H, (int)W,
+                     (int)KH, (int)KW, (int)padH, (int)padW, (int)Kpad,
+                     relu ? 1 : 0);
+  return out;
+}
+
+torch::Tensor conv2d_wgrad(torch::Tensor input, torch::Tensor dout,
+                           int64_t KH, int64_t KW, int64_t padH, int64_t padW,
+                           int64_t Kpad) {
+  TORCH_CHECK(input.is_cuda() && input.is_contiguous());
+  TORCH_CHECK(dout.is_contiguous() && dout.scalar_type() == torch::kBFloat16);
+  int64_t B = input.size(0), Cin = input.size(1);
+  int64_t H = input.size(2), W = input.size(3);
+  int64_t Cout = dout.size(1);
+  auto dwp = torch::zeros({Kpad, Cout},
+                          input.options().dtype(torch::kFloat32));
+  // pick images-per-block so total blocks lands in a healthy range
+  int kt = (int)((Kpad + 63) / 64), nt = (int)((Cout + 63) / 64);
+  long want_z = 6144 / std::max(1, kt * nt);
+  int ipb = (int)std::max<long>(1, (B + want_z - 1) / std::max<long>(1, want_z));
+  dim3 grid(kt, nt, (unsigned)((B + ipb - 1) / ipb));
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(conv2d_wgrad_kernel, grid, dim3(256), 64 * 128,
+                     stream.stream(), bfp(input), bfp(dout),
+                     dwp.data_ptr<float>(),
+                     (int)B, (int)Cin, (int)Cout, (int)H, (int)W,
+                     (int)KH, (int)KW, (int)padH, (int)padW, (int)Kpad, ipb);
+  return dwp;
+}
+
+std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor input) {
+  TORCH_CHECK(input.is_cuda() && input.is_contiguous() && input.dim() == 4);
+  TORCH_CHECK(input.scalar_type() == torch::kBFloat16);
+  int64_t B = input.size(0), C = input.size(1);
+  int64_t H = input.size(2), W = input.size(3);
+  auto out = torch::empty({B, C, H / 2, W / 2}, input.options());
+  auto idx = torch::empty({B, C, H / 2, W / 2},
+                          input.options().dtype(torch::kUInt8));
+  long total = out.numel();
+  int blocks = (int)std::min<long>((total + 255) / 256, 4096);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(maxpool2x2_fwd_kernel, dim3(blocks), dim3(256), 0,
+                     stream.stream(), bfp(input), bfp_mut(out),
+                     idx.data_ptr<unsigned char>(), (long)(B * C),
+                     (int)H, (int)W);
+  return {out, idx};
+}
+
+torch::Tensor maxpool2x2_bwd(torch::Tensor dout, torch::Tensor idx,
+                             int64_t H, int64_t W) {
+  TORCH_CHECK(dout.is_cuda() && dout.is_contiguous());
+  int64_t B = dout.size(0), C = dout.size(1);
+  auto din = torch::empty({B, C, H, W}, dout.options());
+  long total = din.numel();
+  int blocks = (int)std::min<long>((total + 255) / 256, 4096);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(maxpool2x2_bwd_kernel, dim3(blocks), dim3(256), 0,
+                     stream.stream(), bfp(dout),
+                     idx.data_ptr<unsigned char>(), bfp_mut(din),
+                     (long)(B * C), (int)H, (int)W);
+  return din;
+}
+
 torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor Bm) {
   TORCH_CHECK(A.is_cuda() && A.is_contiguous() && Bm.is_contiguous());
   TORCH_CHECK(A.scalar_type() == torch::kBFloat16);
@@ -617,6 +710,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused residual-add + LayerNorm forward (bf16, fp32 stats)");
   m.def("residual_ln_bwd", &residual_ln_bwd,
         "fused residual-add + LayerNorm backward");
+  m.def("conv2d_fwd", &conv2d_fwd,
+        "K4 NCHW bf16 MFMA implicit-GEMM conv (stride 1), fused bias/relu");
+  m.def("conv2d_wgrad", &conv2d_wgrad, "K4 conv weight gradient");
+  m.def("maxpool2x2_fwd", &maxpool2x2_fwd, "2x2 maxpool fwd + argmax");
+  m.def("maxpool2x2_bwd", &maxpool2x2_bwd, "2x2 maxpool gather backward");
   m.def("entropy_fwd", &entropy_fwd, "fused rowwise entropy forward");
   m.def("entropy_bwd", &entropy_bwd, "fused rowwise entropy backward");
   m.def("kl_fwd", &kl_fwd, "fused rowwise KL(teacher||student) forward");

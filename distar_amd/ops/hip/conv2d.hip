@@ -1,0 +1,248 @@
+// K4: spatial conv stack — hand-written CDNA4 MFMA implicit-GEMM conv,
+// NCHW bf16, stride 1 (the whole DI-star conv surface: 1x1 projections and
+// 3x3 pad-1 convs; downsampling is MaxPool, reference
+// obs_encoder/spatial_encoder.py:20-90, head/action_arg_head.py:417-450).
+//
+// Formulation: out[b,co,p] = sum_k im2col[b,p,k] . W[co,k], k = ci*KH*KW +
+// offset.  A-operand = LDS-staged im2col tile [64 px][64 k-chunk] (the
+// gather does the halo/bounds logic, so NO NHWC transposes and no
+// workspace, unlike the MIOpen path it replaces); B-operand = packed
+// weights Wp[Cout][Kpad] read straight from global (small, L2-resident;
+// the B-fragment wants per-lane contiguous k, which row-major Wp gives).
+// Bias + optional ReLU fused into the epilogue.
+//
+// Backward-data reuses the SAME kernel with flipped/transposed packed
+// weights (stride-1 conv duality).  Backward-weight is a second kernel:
+// dW[k,co] += im2col^T . dOut — A = LDS im2col-transposed tile, B = dOut
+// read from global NCHW (for fixed co, pixels are contiguous), fp32
+// atomics into dW.
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+typedef __bf16 bf16x8c __attribute__((ext_vector_type(8)));
+typedef float f32x4c __attribute__((ext_vector_type(4)));
+
+#define CTILE 64                  // px per block (fwd) / k-rows (wgrad)
+#define KC 64                     // k-chunk
+
+// [64][64] bf16 tile, 128-byte rows, granule XOR swizzle
+__device__ __forceinline__ int cswz(int row, int col) {
+  int g = (col >> 3) ^ (row & 7);
+  return row * 128 + (g & 7) * 16 + (col & 7) * 2;
+}
+
+__device__ __forceinline__ bf16x8c clds8(const char* base, int off) {
+  return *(const bf16x8c*)(base + off);
+}
+
+// im2col element (p, k) of image (b): in[b][ci][y+dy][x+dx] with
+// k = ci*KH*KW + (dy+padH)*KW + (dx+padW); 0 outside / beyond K_real.
+__device__ __forceinline__ float im2col_elem(
+    const __hip_bfloat16* __restrict__ inb, int p, int k,
+    int H, int W, int KH, int KW, int padH, int padW, int K_real) {
+  if (k >= K_real) return 0.f;
+  int ci = k / (KH * KW);
+  int off = k % (KH * KW);
+  int y = p / W + off / KW - padH;
+  int x = p % W + off % KW - padW;
+  if (y < 0 || y >= H || x < 0 || x >= W) return 0.f;
+  return __bfloat162float(inb[((long)ci * H + y) * W + x]);
+}
+
+extern "C" __global__ __launch_bounds__(256, 4)
+void conv2d_fwd_kernel(
+    const __hip_bfloat16* __restrict__ input,   // (B, Cin, H, W)
+    const __hip_bfloat16* __restrict__ wp,      // (Cout, Kpad) packed bf16
+    const float* __restrict__ bias,             // (Cout) or nullptr
+    __hip_bfloat16* __restrict__ out,           // (B, Cout, H, W)
+    int B, int Cin, int Cout, int H, int W,
+    int KH, int KW, int padH, int padW, int Kpad, int relu) {
+  const int HW = H * W;
+  const int p0 = blockIdx.x * CTILE;
+  const int n0 = blockIdx.y * 64;
+  const long b = blockIdx.z;
+  if (p0 >= HW || n0 >= Cout) return;
+  const int K_real = Cin * KH * KW;
+
+  extern __shared__ char lds[];                 // A tile [64px][KC] 8 KB
+  const __hip_bfloat16* inb = input + b * Cin * HW;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int l16 = lane & 15;
+  const int lq = lane >> 4;
+  const int band = wave * 16;
+  const int tid = threadIdx.x;
+
+  f32x4c acc[4];
+  for (int nt = 0; nt < 4; ++nt) acc[nt] = (f32x4c){0, 0, 0, 0};
+
+  for (int k0 = 0; k0 < Kpad; k0 += KC) {
+    __syncthreads();
+    // stage im2col chunk: thread t covers (kk = t/64 + 4*i, p = t%64)
+    for (int i = 0; i < KC / 4; ++i) {
+      int kk = (tid >> 6) + 4 * i;
+      int p = tid & 63;
+      float v = (p0 + p < HW)
+          ? im2col_elem(inb, p0 + p, k0 + kk, H, W, KH, KW, padH, padW,
+                        K_real)
+          : 0.f;
+      *(__hip_bfloat16*)(lds + cswz(p, kk)) = __float2bfloat16(v);
+    }
+    __syncthreads();
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8c a = clds8(lds, cswz(band + l16, ks * 32 + lq * 8));
+      for (int nt = 0; nt < 4; ++nt) {
+        int co = n0 + nt * 16 + l16;
+        bf16x8c bw = co < Cout
+            ? *(const bf16x8c*)(wp + (long)co * Kpad + k0 + ks * 32 + lq * 8)
+            : (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
+        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bw, acc[nt],
+                                                          0, 0, 0);
+      }
+    }
+  }
+  // epilogue: bias + relu, scattered bf16 writes
+  for (int nt = 0; nt < 4; ++nt) {
+    int co = n0 + nt * 16 + l16;
+    if (co >= Cout) continue;
+    float bv = bias ? bias[co] : 0.f;
+    for (int r = 0; r < 4; ++r) {
+      int p = p0 + band + lq * 4 + r;
+      if (p >= HW) continue;
+      float v = acc[nt][r] + bv;
+      if (relu) v = fmaxf(v, 0.f);
+      out[(b * Cout + co) * HW + p] = __float2bfloat16(v);
+    }
+  }
+}
+
+// dW[k][co] += sum_px im2col[px][k] * dout[co][px]   (fp32 atomics)
+extern "C" __global__ __launch_bounds__(256, 2)
+void conv2d_wgrad_kernel(
+    const __hip_bfloat16* __restrict__ input,   // (B, Cin, H, W)
+    const __hip_bfloat16* __restrict__ dout,    // (B, Cout, H, W)
+    float* __restrict__ dwp,                    // (Kpad, Cout) fp32
+    int B, int Cin, int Cout, int H, int W,
+    int KH, int KW, int padH, int padW, int Kpad, int ipb) {
+  const int HW = H * W;
+  const int k_base = blockIdx.x * CTILE;        // k tile (rows of dW)
+  const int n0 = blockIdx.y * 64;               // co tile
+  const int b_base = blockIdx.z * ipb;          // image range
+  if (k_base >= Kpad || n0 >= Cout) return;
+  const int K_real = Cin * KH * KW;
+
+  extern __shared__ char lds[];                 // A^T tile [64 k][64 px] 8 KB
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int l16 = lane & 15;
+  const int lq = lane >> 4;
+  const int band = wave * 16;                   // k band of this wave
+  const int tid = threadIdx.x;
+
+  f32x4c acc[4];
+  for (int nt = 0; nt < 4; ++nt) acc[nt] = (f32x4c){0, 0, 0, 0};
+
+  for (int bi = 0; bi < ipb && b_base + bi < B; ++bi) {
+    const long b = b_base + bi;
+    const __hip_bfloat16* inb = input + b * Cin * HW;
+    const __hip_bfloat16* dob = dout + b * Cout * HW;
+    for (int p0 = 0; p0 < HW; p0 += 64) {
+      __syncthreads();
+      // stage im2col^T chunk: rows = k, cols = px
+      for (int i = 0; i < 16; ++i) {
+        int kk = (tid >> 6) + 4 * i;
+        int p = tid & 63;
+        float v = (p0 + p < HW)
+            ? im2col_elem(inb, p0 + p, k_base + kk, H, W, KH, KW, padH,
+                          padW, K_real)
+            : 0.f;
+        *(__hip_bfloat16*)(lds + cswz(kk, p)) = __float2bfloat16(v);
+      }
+      __syncthreads();
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8c a = clds8(lds, cswz(band + l16, ks * 32 + lq * 8));
+        for (int nt = 0; nt < 4; ++nt) {
+          int co = n0 + nt * 16 + l16;
+          bf16x8c bdo = (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
+          if (co < Cout) {
+            int p = p0 + ks * 32 + lq * 8;
+            if (p + 8 <= HW) {
+              // HW*2B is not always 16B-aligned per-row: memcpy lets the
+              // compiler emit the widest legal loads
+              __builtin_memcpy(&bdo, dob + (long)co * HW + p, 16);
+            } else {
+              for (int j = 0; j < 8; ++j)
+                if (p + j < HW)
+                  bdo[j] = ((const __bf16*)dob)[(long)co * HW + p + j];
+            }
+          }
+          acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bdo, acc[nt],
+                                                            0, 0, 0);
+        }
+      }
+    }
+  }
+  // accumulate into dwp (Kpad, Cout) fp32
+  for (int nt = 0; nt < 4; ++nt) {
+    int co = n0 + nt * 16 + l16;
+    if (co >= Cout) continue;
+    for (int r = 0; r < 4; ++r) {
+      int k = k_base + band + lq * 4 + r;
+      if (k >= Kpad) continue;
+      atomicAdd(&dwp[(long)k * Cout + co], acc[nt][r]);
+    }
+  }
+}
+
+// MaxPool2d(2,2) forward with packed argmax (2 bits would do; store u8).
+extern "C" __global__ void maxpool2x2_fwd_kernel(
+    const __hip_bfloat16* __restrict__ in,      // (N, H, W) flattened b*c
+    __hip_bfloat16* __restrict__ out,           // (N, H/2, W/2)
+    unsigned char* __restrict__ idx,            // (N, H/2, W/2)
+    long NC, int H, int W) {
+  int Ho = H / 2, Wo = W / 2;
+  long total = NC * Ho * Wo;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int xo = i % Wo;
+    int yo = (i / Wo) % Ho;
+    long n = i / ((long)Wo * Ho);
+    const __hip_bfloat16* p = in + (n * H + yo * 2) * W + xo * 2;
+    float v00 = __bfloat162float(p[0]);
+    float v01 = __bfloat162float(p[1]);
+    float v10 = __bfloat162float(p[W]);
+    float v11 = __bfloat162float(p[W + 1]);
+    float m = v00;
+    int a = 0;
+    if (v01 > m) { m = v01; a = 1; }
+    if (v10 > m) { m = v10; a = 2; }
+    if (v11 > m) { m = v11; a = 3; }
+    out[i] = __float2bfloat16(m);
+    idx[i] = (unsigned char)a;
+  }
+}
+
+// gather-style backward: each INPUT position checks its pool cell's argmax
+// (no atomics, one coalesced pass — ATen's scatter-add bwd was 12 ms/step).
+extern "C" __global__ void maxpool2x2_bwd_kernel(
+    const __hip_bfloat16* __restrict__ dout,    // (N, H/2, W/2)
+    const unsigned char* __restrict__ idx,
+    __hip_bfloat16* __restrict__ din,           // (N, H, W)
+    long NC, int H, int W) {
+  int Ho = H / 2, Wo = W / 2;
+  long total = NC * H * W;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int x = i % W;
+    int y = (i / W) % H;
+    long n = i / ((long)W * H);
+    float g = 0.f;
+    if (y / 2 < Ho && x / 2 < Wo) {
+      long o = (n * Ho + y / 2) * Wo + x / 2;
+      int a = (y & 1) * 2 + (x & 1);
+      if (idx[o] == a) g = __bfloat162float(dout[o]);
+    }
+    din[i] = __float2bfloat16(g);
+  }
+}

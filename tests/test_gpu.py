@@ -501,3 +501,64 @@ def test_fused_residual_ln_matches_eager():
                                atol=0.02 * float(ln.weight.grad.abs().max()))
     torch.testing.assert_close(gb_h, ln.bias.grad, rtol=2e-2,
                                atol=0.02 * float(ln.bias.grad.abs().max()))
+
+
+# --------------------------------------------------------------- K4 conv
+@pytest.mark.parametrize('cin,cout,hw,kh', [
+    (56, 32, (152, 160), 1),     # spatial project
+    (32, 64, (76, 80), 3),       # downsample convs
+    (128, 128, (19, 20), 3),     # ResBlocks
+    (132, 128, (19, 20), 1),     # location head 1x1
+])
+def test_conv2d_hip_matches_eager(cin, cout, hw, kh):
+    """K4 MFMA implicit-GEMM conv vs fp32 F.conv2d (fwd + both bwds)."""
+    from distar_amd.ops.conv2d import Conv2dHIP
+    torch.manual_seed(5)
+    H, W = hw
+    B = 3
+    conv = Conv2dHIP(cin, cout, kh, 1, padding=kh // 2).cuda()
+    x0 = (torch.randn(B, cin, H, W, device='cuda') * 0.5).bfloat16()
+    dout = (torch.randn(B, cout, H, W, device='cuda') * 0.5).bfloat16()
+
+    x_h = x0.detach().clone().requires_grad_(True)
+    out_h = conv(x_h)
+    out_h.backward(dout)
+    gw_h = conv.weight.grad.clone()
+    gb_h = conv.bias.grad.clone()
+    conv.weight.grad = None
+    conv.bias.grad = None
+
+    x_e = x0.detach().clone().float().requires_grad_(True)
+    out_e = torch.nn.functional.conv2d(x_e, conv.weight, conv.bias,
+                                       padding=kh // 2)
+    out_e.backward(dout.float())
+
+    torch.testing.assert_close(out_h.float(), out_e.detach(),
+                               rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(x_h.grad.float(), x_e.grad.float(),
+                               rtol=5e-2, atol=5e-2)
+    sw = float(conv.weight.grad.abs().max())
+    torch.testing.assert_close(gw_h, conv.weight.grad, rtol=3e-2,
+                               atol=0.03 * sw)
+    sb = float(conv.bias.grad.abs().max())
+    torch.testing.assert_close(gb_h, conv.bias.grad, rtol=3e-2,
+                               atol=0.03 * sb)
+
+
+def test_maxpool2x2_hip_matches_eager():
+    from distar_amd.ops.conv2d import max_pool2x2
+    torch.manual_seed(6)
+    x0 = (torch.randn(4, 32, 76, 80, device='cuda')).bfloat16()
+    x_h = x0.detach().clone().requires_grad_(True)
+    out_h = max_pool2x2(x_h)
+    dout = torch.randn_like(out_h).bfloat16()
+    out_h.backward(dout)
+    x_e = x0.detach().clone().requires_grad_(True)
+    os.environ['DISTAR_AMD_CONV'] = '0'
+    try:
+        out_e = max_pool2x2(x_e)
+        out_e.backward(dout)
+    finally:
+        os.environ.pop('DISTAR_AMD_CONV', None)
+    torch.testing.assert_close(out_h.float(), out_e.float())
+    torch.testing.assert_close(x_h.grad.float(), x_e.grad.float())
