@@ -43,12 +43,21 @@ class TextLogger:
 
 
 class ScalarLogger:
-    """JSONL scalar sink with a tensorboard-like API."""
+    """Scalar sink with a tensorboard-like API: JSONL (greppable) + a real
+    TensorBoard event file (utils/tb.py — drop-in for the reference's
+    tensorboardX streams, `ctools/utils/log_helper.py:23-64`)."""
 
-    def __init__(self, path, name='scalars'):
+    def __init__(self, path, name='scalars', tensorboard=True):
         os.makedirs(path, exist_ok=True)
         self._file = open(os.path.join(path, f'{name}.jsonl'), 'a')
         self._vars = set()
+        self._tb = None
+        if tensorboard:
+            try:
+                from .tb import SummaryWriter
+                self._tb = SummaryWriter(os.path.join(path, 'tb', name))
+            except Exception:  # noqa: BLE001 - TB stream is best-effort
+                self._tb = None
 
     def register_var(self, name):
         self._vars.add(name)
@@ -56,12 +65,18 @@ class ScalarLogger:
     def add_scalar(self, name, value, global_step=0):
         self._file.write(json.dumps({'step': int(global_step), 'key': name,
                                      'value': float(value), 't': time.time()}) + '\n')
+        if self._tb is not None:
+            self._tb.add_scalar(name, value, global_step)
 
     def flush(self):
         self._file.flush()
+        if self._tb is not None:
+            self._tb.flush()
 
     def close(self):
         self._file.close()
+        if self._tb is not None:
+            self._tb.close()
 
 
 class AverageMeter:
