@@ -74,41 +74,68 @@ void conv2d_fwd_kernel(
   const int band = wave * 16;
   const int tid = threadIdx.x;
 
-  f32x4c acc[4];
+  // Swapped MFMA: D[m=co][n=px] = W[co,k] . im2col^T[k,px] — the A
+  // operand streams straight from packed global weights (L1/L2-resident,
+  // per-lane contiguous k) and the accumulator's col=lane&15 becomes the
+  // PIXEL, so the epilogue stores 16 consecutive pixels per lane group.
+  f32x4c acc[4];                        // [pixel n-tile]
   for (int nt = 0; nt < 4; ++nt) acc[nt] = (f32x4c){0, 0, 0, 0};
 
   for (int k0 = 0; k0 < Kpad; k0 += KC) {
     __syncthreads();
-    // stage im2col chunk: thread t covers (kk = t/64 + 4*i, p = t%64)
-    for (int i = 0; i < KC / 4; ++i) {
-      int kk = (tid >> 6) + 4 * i;
-      int p = tid & 63;
-      float v = (p0 + p < HW)
-          ? im2col_elem(inb, p0 + p, k0 + kk, H, W, KH, KW, padH, padW,
-                        K_real)
-          : 0.f;
-      *(__hip_bfloat16*)(lds + cswz(p, kk)) = __float2bfloat16(v);
+    // k-major staging: ci/dy/dx computed once per k row; pixel groups of 8
+    // are one vector load when the run stays inside a map row
+    for (int task = tid; task < KC * 8; task += 256) {
+      int kk = task >> 3, g = task & 7;
+      int k = k0 + kk;
+      __hip_bfloat16 vals[8] = {};
+      if (k < K_real) {
+        int ci = k / (KH * KW);
+        int off = k % (KH * KW);
+        int dy = off / KW - padH, dx = off % KW - padW;
+        int pbase = p0 + g * 8;
+        int y = pbase / W + dy;
+        int x = pbase % W + dx;
+        const __hip_bfloat16* src = inb + ((long)ci * H + y) * W + x;
+        if (pbase + 7 < HW && (pbase % W) + 7 < W && y >= 0 && y < H &&
+            x >= 0 && x + 7 < W) {
+          __builtin_memcpy(vals, src, 16);
+        } else {
+          for (int j = 0; j < 8; ++j) {
+            int p = pbase + j;
+            if (p < HW) {
+              int yj = p / W + dy, xj = p % W + dx;
+              if (yj >= 0 && yj < H && xj >= 0 && xj < W)
+                vals[j] = inb[((long)ci * H + yj) * W + xj];
+            }
+          }
+        }
+      }
+      for (int j = 0; j < 8; ++j)
+        *(__hip_bfloat16*)(lds + cswz(g * 8 + j, kk)) = vals[j];
     }
     __syncthreads();
     for (int ks = 0; ks < 2; ++ks) {
-      bf16x8c a = clds8(lds, cswz(band + l16, ks * 32 + lq * 8));
+      int co_a = band + l16;            // A row: this wave's co band
+      bf16x8c a = (n0 + co_a < Cout)
+          ? *(const bf16x8c*)(wp + (long)(n0 + co_a) * Kpad + k0 + ks * 32
+                              + lq * 8)
+          : (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
       for (int nt = 0; nt < 4; ++nt) {
-        int co = n0 + nt * 16 + l16;
-        bf16x8c bw = co < Cout
-            ? *(const bf16x8c*)(wp + (long)co * Kpad + k0 + ks * 32 + lq * 8)
-            : (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
-        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bw, acc[nt],
+        bf16x8c bi = clds8(lds, cswz(nt * 16 + l16, ks * 32 + lq * 8));
+        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bi, acc[nt],
                                                           0, 0, 0);
       }
     }
   }
-  // epilogue: bias + relu, scattered bf16 writes
-  for (int nt = 0; nt < 4; ++nt) {
-    int co = n0 + nt * 16 + l16;
+  // epilogue: acc rows = co (band + lq*4 + r), cols = px (nt*16 + l16):
+  // 16-lane groups store 32 contiguous bytes
+  for (int r = 0; r < 4; ++r) {
+    int co = n0 + band + lq * 4 + r;
     if (co >= Cout) continue;
     float bv = bias ? bias[co] : 0.f;
-    for (int r = 0; r < 4; ++r) {
-      int p = p0 + band + lq * 4 + r;
+    for (int nt = 0; nt < 4; ++nt) {
+      int p = p0 + nt * 16 + l16;
       if (p >= HW) continue;
       float v = acc[nt][r] + bv;
       if (relu) v = fmaxf(v, 0.f);
@@ -149,15 +176,36 @@ void conv2d_wgrad_kernel(
     const __hip_bfloat16* dob = dout + b * Cout * HW;
     for (int p0 = 0; p0 < HW; p0 += 64) {
       __syncthreads();
-      // stage im2col^T chunk: rows = k, cols = px
-      for (int i = 0; i < 16; ++i) {
-        int kk = (tid >> 6) + 4 * i;
-        int p = tid & 63;
-        float v = (p0 + p < HW)
-            ? im2col_elem(inb, p0 + p, k_base + kk, H, W, KH, KW, padH,
-                          padW, K_real)
-            : 0.f;
-        *(__hip_bfloat16*)(lds + cswz(kk, p)) = __float2bfloat16(v);
+      // stage im2col^T chunk: rows = k (ci/dy/dx once per row), cols =
+      // px in groups of 8 -> one 16B ds_write per group
+      for (int task = tid; task < CTILE * 8; task += 256) {
+        int kk = task >> 3, g = task & 7;
+        int k = k_base + kk;
+        __hip_bfloat16 vals[8] = {};
+        if (k < K_real) {
+          int ci = k / (KH * KW);
+          int off = k % (KH * KW);
+          int dy = off / KW - padH, dx = off % KW - padW;
+          int pbase = p0 + g * 8;
+          int y = pbase / W + dy;
+          int x = pbase % W + dx;
+          const __hip_bfloat16* src = inb + ((long)ci * H + y) * W + x;
+          if (pbase + 7 < HW && (pbase % W) + 7 < W && y >= 0 && y < H &&
+              x >= 0 && x + 7 < W) {
+            __builtin_memcpy(vals, src, 16);
+          } else {
+            for (int j = 0; j < 8; ++j) {
+              int p = pbase + j;
+              if (p < HW) {
+                int yj = p / W + dy, xj = p % W + dx;
+                if (yj >= 0 && yj < H && xj >= 0 && xj < W)
+                  vals[j] = inb[((long)ci * H + yj) * W + xj];
+              }
+            }
+          }
+        }
+        // cols g*8..g*8+7 share one swizzled granule: single 16B write
+        __builtin_memcpy(lds + cswz(kk, g * 8), vals, 16);
       }
       __syncthreads();
       for (int ks = 0; ks < 2; ++ks) {

@@ -258,9 +258,29 @@ def test_su_sample_kernel_semantics():
     sel = l0e > -1e8
     torch.testing.assert_close(l0h[sel], l0e[sel], rtol=5e-2, atol=5e-2)
     assert (l0h[~sel] < -1e8).all()
-    # first picks overwhelmingly agree
-    agree = (results_h[live, 0] == results_e[live, 0]).float().mean()
-    assert agree >= 0.7, float(agree)
+    # Exact agreement on every CDF-stable pick: a pick can only legally
+    # diverge when the shared uniform lands within eps of an (eager) CDF
+    # boundary — there bf16/fp32 rounding differences between the kernel
+    # and eager softmax flip the inverse-CDF bin.  Away from boundaries
+    # the kernel MUST reproduce eager exactly, step by step, while the
+    # prefix agrees.  (replaces the 0.7 first-pick gate.)
+    eps = 0.05
+    probs_e = torch.softmax(logits_e.float(), dim=-1)
+    stable_total = stable_agree = 0
+    for b in live.tolist():
+        steps = min(int(num_e[b]), int(num_h[b]))
+        for s in range(steps):
+            if s > 0 and not torch.equal(results_h[b, :s], results_e[b, :s]):
+                break
+            cdf = probs_e[b, s].cumsum(0)
+            u = float(uniforms[b, s])
+            dist = float((cdf - u).abs().min())
+            if dist > eps:
+                stable_total += 1
+                stable_agree += int(results_h[b, s] == results_e[b, s])
+    assert stable_total >= 10, 'stability check is vacuous'
+    assert stable_agree == stable_total, \
+        f'{stable_agree}/{stable_total} stable picks agree'
     # invariants per live row
     for b in live.tolist():
         n = int(num_h[b])
