@@ -73,22 +73,41 @@ def offline_entries(data_dir, min_loop):
     return entries
 
 
-def replay_entries(data_dir, min_loop, cfg=None):
+def replay_entries(data_dir, min_loop, cfg=None, parse_race='Z'):
+    """Live replay decode (reference `bin/gen_z.py` worker_loop): decode both
+    sides of each replay, keep WINNING sides of the requested race whose
+    games ran at least min_loop, key by the decoder's real metadata
+    (map / race pair / born location)."""
     from ..data.replay_decoder import ReplayDecoder
     from ..utils.config import Config
     decoder = ReplayDecoder(cfg or Config({}))
     entries = []
+    race_initial = {'zerg': 'Z', 'terran': 'T', 'protoss': 'P', 'random': 'R'}
     for name in sorted(os.listdir(data_dir)):
         if not name.endswith('.SC2Replay'):
             continue
         for player_idx in range(2):
             traj = decoder.run(os.path.join(data_dir, name), player_idx)
-            if not traj:
+            meta = decoder.last_meta
+            if not traj or meta is None:
                 continue
-            feature = decoder._last_feature if hasattr(decoder, '_last_feature') else None
-            # winning-side filter happens in the decoder outcome check
-            entries.append(('KingsCove', 'zerg', 0,
-                            z_from_traj(feature, traj, traj[-1].get('game_loop', 0))))
+            if meta['result'] != 1:                   # winning side only
+                continue
+            if meta['end_loop'] < min_loop:
+                continue
+            if parse_race and \
+                    race_initial.get(meta['home_race'], 'Z') not in parse_race:
+                continue
+            # Z extraction from the already-stamped steps: every step
+            # carries identical beginning_order/cumulative_stat tensors
+            s0 = traj[0]['scalar_info']
+            cum_idx = torch.nonzero(
+                torch.as_tensor(s0['cumulative_stat']) > 0).squeeze(1).tolist()
+            z = [s0['beginning_order'].tolist(), cum_idx,
+                 s0['bo_location'].tolist(), int(meta['end_loop'])]
+            mix = meta['home_race'] if meta['home_race'] == meta['away_race'] \
+                else meta['home_race'] + meta['away_race']
+            entries.append((meta['map_name'], mix, meta['born_location'], z))
     decoder.close()
     return entries
 

@@ -67,6 +67,10 @@ class ReplayDecoder:
         self._controller = None
         self._decode_count = 0
         self._cur_version = None
+        # metadata of the last successful decode (gen_z consumes this):
+        # {'map_name','home_race','away_race','born_location','result',
+        #  'end_loop'}
+        self.last_meta = None
 
     def _version_of(self, replay_path):
         """SC2 version sniff from the replay's MPQ `replay.gamemetadata.json`
@@ -117,8 +121,11 @@ class ReplayDecoder:
         map_size = (game_info.start_raw.map_size.x,
                     game_info.start_raw.map_size.y)
         raw_actions = []
+        outcome = None
+        end_loop = 0
         while True:
             obs = ctrl.observe()
+            end_loop = obs['game_loop']
             for act in getattr(obs['raw_obs'], 'actions', []):
                 if act.HasField('action_raw'):
                     uc = act.action_raw.unit_command \
@@ -126,7 +133,13 @@ class ReplayDecoder:
                     raw_actions.append((obs['game_loop'],
                                         uc.ability_id if uc else 0,
                                         tuple(uc.unit_tags) if uc else (), act))
-            if ctrl.outcome(obs) is not None:
+            results = getattr(obs['raw_obs'], 'player_result', [])
+            if results:
+                # the OBSERVED player's result (player_result lists all)
+                raw = next((r.result for r in results
+                            if r.player_id == player_idx + 1),
+                           results[0].result)
+                outcome = {1: 1, 2: -1, 3: 0}.get(raw, 0)
                 break
             ctrl.step(PASS1_STEP)
         raw_actions = self._filter.run(raw_actions)
@@ -161,6 +174,21 @@ class ReplayDecoder:
             step['scalar_info']['beginning_order'] = bo
             step['scalar_info']['bo_location'] = bo_loc
             step['scalar_info']['cumulative_stat'] = cum
+        # decode metadata for gen_z (map, races, born location, outcome of
+        # the observed player — reference gen_z keeps winning sides only)
+        race_names = {1: 'terran', 2: 'zerg', 3: 'protoss', 4: 'random'}
+        races = {info.player_id: race_names.get(int(info.race_requested), 'zerg')
+                 for info in game_info.player_info}
+        home_id = player_idx + 1
+        away_id = next((pid for pid in races if pid != home_id), home_id)
+        self.last_meta = {
+            'map_name': game_info.map_name,
+            'home_race': races.get(home_id, 'zerg'),
+            'away_race': races.get(away_id, 'zerg'),
+            'born_location': getattr(feature, 'home_born_location', 0),
+            'result': outcome,
+            'end_loop': end_loop,
+        }
         return traj_data
 
     def close(self):

@@ -18,7 +18,7 @@ from . import hip_ext
 
 
 def _kpad(k):
-    return (k + 63) // 64 * 64
+    return (k + 127) // 128 * 128     # KC=128 chunks in the kernels
 
 
 def _pack_weight(w, KH, KW):

@@ -559,7 +559,7 @@ std::vector<torch::Tensor> residual_ln_fwd(
     TORCH_CHECK(a->is_contiguous() && a->scalar_type() == torch::kBFloat16);
     ap = bfp(*a);
   }
-  int blocks = (int)std::min<long>((R + 3) / 4, 4096);
+  int blocks = (int)std::min<long>((R + 3) / 4, 1 << 20);
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(residual_ln_fwd_kernel, dim3(blocks), dim3(256), 0,
                      stream.stream(), bfp(x), ap,
@@ -580,8 +580,8 @@ std::vector<torch::Tensor> residual_ln_bwd(
   auto fopt = dy.options().dtype(torch::kFloat32);
   auto dw = torch::zeros({C}, fopt);
   auto db = torch::zeros({C}, fopt);
-  int blocks = (int)std::min<long>((R + 3) / 4, 2048);
-  size_t lds = 2 * C * sizeof(float);
+  int blocks = (int)std::min<long>((R + 3) / 4, 8192);
+  size_t lds = 8 * C * sizeof(float);    // [4 waves][2*C] combine buffer
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(residual_ln_bwd_kernel, dim3(blocks), dim3(256), lds,
                      stream.stream(), bfp(dy), bfp(s),
@@ -619,7 +619,7 @@ torch::Tensor conv2d_fwd(torch::Tensor input, torch::Tensor wp,
   int HW = (int)(H * W);
   dim3 grid((HW + 63) / 64, (unsigned)((Cout + 63) / 64), (unsigned)B);
   auto stream = c10::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(conv2d_fwd_kernel, grid, dim3(256), 64 * 128,
+  hipLaunchKernelGGL(conv2d_fwd_kernel, grid, dim3(256), 64 * 256,
                      stream.stream(), bfp(input), bfp(wp), bp, bfp_mut(out),
                      (int)B, (int)Cin, (int)Cout, (int)H, (int)W,
                      (int)KH, (int)KW, (int)padH, (int)padW, (int)Kpad,
@@ -643,7 +643,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor input, torch::Tensor dout,
   int ipb = (int)std::max<long>(1, (B + want_z - 1) / std::max<long>(1, want_z));
   dim3 grid(kt, nt, (unsigned)((B + ipb - 1) / ipb));
   auto stream = c10::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(conv2d_wgrad_kernel, grid, dim3(256), 64 * 128,
+  hipLaunchKernelGGL(conv2d_wgrad_kernel, grid, dim3(256), 64 * 256,
                      stream.stream(), bfp(input), bfp(dout),
                      dwp.data_ptr<float>(),
                      (int)B, (int)Cin, (int)Cout, (int)H, (int)W,

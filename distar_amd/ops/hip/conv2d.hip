@@ -23,12 +23,12 @@ typedef __bf16 bf16x8c __attribute__((ext_vector_type(8)));
 typedef float f32x4c __attribute__((ext_vector_type(4)));
 
 #define CTILE 64                  // px per block (fwd) / k-rows (wgrad)
-#define KC 64                     // k-chunk
+#define KC 128                    // k-chunk (4 MFMA k-steps per barrier)
 
-// [64][64] bf16 tile, 128-byte rows, granule XOR swizzle
+// [64][128] bf16 tile, 256-byte rows, granule XOR swizzle
 __device__ __forceinline__ int cswz(int row, int col) {
-  int g = (col >> 3) ^ (row & 7);
-  return row * 128 + (g & 7) * 16 + (col & 7) * 2;
+  int g = (col >> 3) ^ (row & 7);   // 16 granules/row; XOR spreads banks
+  return row * 256 + g * 16 + (col & 7) * 2;
 }
 
 __device__ __forceinline__ bf16x8c clds8(const char* base, int off) {
@@ -115,7 +115,7 @@ void conv2d_fwd_kernel(
         *(__hip_bfloat16*)(lds + cswz(g * 8 + j, kk)) = vals[j];
     }
     __syncthreads();
-    for (int ks = 0; ks < 2; ++ks) {
+    for (int ks = 0; ks < 4; ++ks) {
       int co_a = band + l16;            // A row: this wave's co band
       bf16x8c a = (n0 + co_a < Cout)
           ? *(const bf16x8c*)(wp + (long)(n0 + co_a) * Kpad + k0 + ks * 32
@@ -174,12 +174,12 @@ void conv2d_wgrad_kernel(
     const long b = b_base + bi;
     const __hip_bfloat16* inb = input + b * Cin * HW;
     const __hip_bfloat16* dob = dout + b * Cout * HW;
-    for (int p0 = 0; p0 < HW; p0 += 64) {
+    for (int p0 = 0; p0 < HW; p0 += KC) {
       __syncthreads();
       // stage im2col^T chunk: rows = k (ci/dy/dx once per row), cols =
       // px in groups of 8 -> one 16B ds_write per group
-      for (int task = tid; task < CTILE * 8; task += 256) {
-        int kk = task >> 3, g = task & 7;
+      for (int task = tid; task < CTILE * (KC / 8); task += 256) {
+        int kk = task >> 4, g = task & 15;
         int k = k_base + kk;
         __hip_bfloat16 vals[8] = {};
         if (k < K_real) {
@@ -208,7 +208,7 @@ void conv2d_wgrad_kernel(
         __builtin_memcpy(lds + cswz(kk, g * 8), vals, 16);
       }
       __syncthreads();
-      for (int ks = 0; ks < 2; ++ks) {
+      for (int ks = 0; ks < 4; ++ks) {
         bf16x8c a = clds8(lds, cswz(band + l16, ks * 32 + lq * 8));
         for (int nt = 0; nt < 4; ++nt) {
           int co = n0 + nt * 16 + l16;

@@ -80,14 +80,10 @@ extern "C" __global__ void residual_ln_bwd_kernel(
     long R, int C) {
   const int lane = threadIdx.x & 63;
   const int epw = C / 64;
-  extern __shared__ float lds[];                // dw/db partials: 2*C floats
-  float* dw_part = lds;
-  float* db_part = lds + C;
-  for (int i = threadIdx.x; i < C; i += blockDim.x) {
-    dw_part[i] = 0.f;
-    db_part[i] = 0.f;
-  }
-  __syncthreads();
+  extern __shared__ float lds[];                // [4 waves][2*C] partials
+  // each (lane, i) owns column lane*epw+i in EVERY row it visits: the
+  // LN-param partials accumulate in registers, no per-element atomics
+  float dwacc[16] = {}, dbacc[16] = {};
   for (long row = blockIdx.x * 4 + (threadIdx.x >> 6); row < R;
        row += (long)gridDim.x * 4) {
     const __hip_bfloat16* dyr = dy + row * C;
@@ -107,9 +103,8 @@ extern "C" __global__ void residual_ln_bwd_kernel(
       xh[i] = xhat;
       s1 += wg;
       s2 += wg * xhat;
-      // per-block LN-param partials (LDS, fp32)
-      atomicAdd(&dw_part[c], dyv * xhat);
-      atomicAdd(&db_part[c], dyv);
+      dwacc[i] += dyv * xhat;
+      dbacc[i] += dyv;
     }
     s1 = warp_sum(s1) / C;
     s2 = warp_sum(s2) / C;
@@ -118,9 +113,21 @@ extern "C" __global__ void residual_ln_bwd_kernel(
       drv[i] = __float2bfloat16((g[i] - s1 - xh[i] * s2) * r);
     __builtin_memcpy(dsum + row * C + lane * epw, drv, epw * 2);
   }
+  // combine the 4 waves' register partials through LDS, then one global
+  // atomic pass per block
+  const int wv = threadIdx.x >> 6;
+  for (int i = 0; i < epw; ++i) {
+    lds[(wv * C + lane * epw + i) * 2 + 0] = dwacc[i];
+    lds[(wv * C + lane * epw + i) * 2 + 1] = dbacc[i];
+  }
   __syncthreads();
-  for (int i = threadIdx.x; i < C; i += blockDim.x) {
-    atomicAdd(&dw[i], dw_part[i]);
-    atomicAdd(&db[i], db_part[i]);
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    float dw_s = 0.f, db_s = 0.f;
+    for (int wvi = 0; wvi < 4; ++wvi) {
+      dw_s += lds[(wvi * C + c) * 2 + 0];
+      db_s += lds[(wvi * C + c) * 2 + 1];
+    }
+    atomicAdd(&dw[c], dw_s);
+    atomicAdd(&db[c], db_s);
   }
 }

@@ -477,3 +477,40 @@ def test_transform_obs_on_real_protos():
     assert not invalid
     assert action_info['action_type'] is not None
     assert int(su_num) == 2      # one selected unit + the end flag
+
+
+def test_gen_z_live_decode_end_to_end(tmp_path, monkeypatch):
+    """The gen_z live path over the fake-websocket SC2: decode, winner
+    filter, Z extraction, and map/race/born aggregation (reference
+    bin/gen_z.py worker_loop+result_loop)."""
+    from distar_amd.bin.gen_z import aggregate, replay_entries
+    replay = _synthetic_replay(str(tmp_path))
+
+    act1 = P.raw_action(3674, unit_tags=[2000], target_pos=(30., 40.),
+                        game_loop=10)
+
+    def fake_launch(cfg=None, port=None, version=None, **kw):
+        class Proc:
+            def kill(self):
+                pass
+        return Proc(), 23456
+
+    def fake_connect(host, port, resource='/sc2api', timeout=120.0):
+        # player 1 wins, player 2 loses -> only one side kept
+        return FakeSC2(end_after_observes=2, actions_on_observe={1: [act1]})
+
+    from distar_amd.data import replay_decoder as rd_mod
+    monkeypatch.setattr(rd_mod, 'launch_game_process', fake_launch)
+    monkeypatch.setattr(protocol.WebSocket, 'connect',
+                        staticmethod(fake_connect))
+
+    entries = replay_entries(str(tmp_path), min_loop=0)
+    # both sides decoded; FakeSC2 reports Victory for player 1 only
+    assert len(entries) == 1
+    map_name, mix, born, z = entries[0]
+    assert map_name == 'KingsCove'
+    assert mix == 'zerg'
+    bo, cum_idx, bo_loc, end_loop = z
+    assert isinstance(bo, list) and isinstance(cum_idx, list)
+    agg = aggregate(entries)
+    assert 'KingsCove' in agg and 'zerg' in agg['KingsCove']
