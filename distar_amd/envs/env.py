@@ -107,19 +107,72 @@ class SC2Env:
         self._episode_count += 1
         self._episode_steps = 0
         self._next_obs_step = [0] * self._agent_num
-        return {i: ctrl.observe() for i, ctrl in enumerate(self._controllers)}
+        self._game_infos = [ctrl.game_info() for ctrl in self._controllers]
+        return {i: self._wrap_obs(i, ctrl.observe())
+                for i, ctrl in enumerate(self._controllers)}
+
+    def _wrap_obs(self, idx, obs):
+        obs['game_info_proto'] = self._game_infos[idx]
+        return obs
+
+    # --------------------------------------------------------- action protos
+    def transform_action(self, actions):
+        """Agent action dicts -> raw-action protos (reference
+        `envs/env.py:457-483`): `{'func_id','skip_steps','queued',
+        'unit_tags','target_unit_tag','location'}` routed by the raw
+        function table's function_type."""
+        from .protocol import sc_pb
+        from ..lib.features import RAW_FUNC_BY_ID
+        sc2_actions = []
+        skip = None
+        for a in actions:
+            skip = int(a.get('skip_steps', 0)) if skip is None \
+                else min(skip, int(a.get('skip_steps', 0)))
+            f = RAW_FUNC_BY_ID[int(a['func_id'])]
+            ftype = f['function_type']
+            if ftype == 'raw_no_op':
+                continue
+            act = sc_pb.Action()
+            if ftype == 'raw_move_camera':
+                loc = a['location']
+                act.action_raw.camera_move.center_world_space.x = float(loc[0])
+                act.action_raw.camera_move.center_world_space.y = float(loc[1])
+            elif ftype == 'raw_autocast':
+                ta = act.action_raw.toggle_autocast
+                ta.ability_id = f['ability_id']
+                ta.unit_tags.extend(int(t) for t in a.get('unit_tags', []))
+            else:
+                uc = act.action_raw.unit_command
+                uc.ability_id = f['ability_id']
+                uc.queue_command = bool(a.get('queued', 0))
+                uc.unit_tags.extend(int(t) for t in a.get('unit_tags', []))
+                if ftype == 'raw_cmd_pt':
+                    loc = a['location']
+                    uc.target_world_space_pos.x = float(loc[0])
+                    uc.target_world_space_pos.y = float(loc[1])
+                elif ftype == 'raw_cmd_unit':
+                    uc.target_unit_tag = int(a['target_unit_tag'])
+            sc2_actions.append(act)
+        return sc2_actions, (skip or 0)
 
     # ----------------------------------------------------------------- step
     def step(self, actions):
         """Per-agent skip scheduling (reference `envs/env.py:333-375`): each
         action carries skip_steps; the env advances to min(next_obs_step)
-        plus 0-3 random delay steps, then observes agents that are due."""
+        plus 0-3 random delay steps, then observes agents that are due.
+        Accepts pre-built `{'raw_actions': [...]}` or agent action dicts
+        (single or list) with `func_id`."""
         for idx, action in (actions or {}).items():
             if action is None:
                 continue
-            self._controllers[idx].acts(action)
-            self._next_obs_step[idx] = self._episode_steps + \
-                int(action.get('skip_steps', 0)) + 1
+            if isinstance(action, dict) and 'raw_actions' in action:
+                protos = action['raw_actions']
+                skip = int(action.get('skip_steps', 0))
+            else:
+                alist = action if isinstance(action, list) else [action]
+                protos, skip = self.transform_action(alist)
+            self._controllers[idx].acts({'raw_actions': protos})
+            self._next_obs_step[idx] = self._episode_steps + skip + 1
         target = min(self._next_obs_step)
         delay = self._rng.choices([0, 1, 2, 3], weights=DELAY_WEIGHTS)[0]
         step_count = max(target + delay - self._episode_steps, 1)
@@ -130,7 +183,7 @@ class SC2Env:
         obs, rewards, done = {}, {}, False
         for i, ctrl in enumerate(self._controllers):
             if self._next_obs_step[i] <= self._episode_steps:
-                o = ctrl.observe()
+                o = self._wrap_obs(i, ctrl.observe())
                 obs[i] = o
                 outcome = ctrl.outcome(o)
                 if outcome is not None:

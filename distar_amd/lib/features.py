@@ -55,6 +55,9 @@ for _func in RAW_FUNCTIONS:
     if _func['ability_id']:
         RAW_ABILITY_IDS[_func['ability_id']].append(_func)
 
+# func_id -> raw function entry (env.transform_action routing)
+RAW_FUNC_BY_ID = {f['id']: f for f in RAW_FUNCTIONS}
+
 
 def unpack_feature_layer(image_data):
     """Decode a protobuf ImageData-like object (bits_per_pixel, size, data)

