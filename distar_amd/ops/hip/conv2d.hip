@@ -78,7 +78,10 @@ void conv2d_fwd_kernel(
   // operand streams straight from packed global weights (L1/L2-resident,
   // per-lane contiguous k) and the accumulator's col=lane&15 becomes the
   // PIXEL, so the epilogue stores 16 consecutive pixels per lane group.
-  f32x4c acc[4];                        // [pixel n-tile]
+  // Each WAVE owns one 16-pixel band and ALL 64 cout rows (4 m-tiles):
+  // one LDS B-fragment feeds 4 MFMAs (the A-operands are cheap global
+  // reads), 4:1 MFMA:ds_read instead of 1:1.
+  f32x4c acc[4];                        // [cout m-tile]
   for (int nt = 0; nt < 4; ++nt) acc[nt] = (f32x4c){0, 0, 0, 0};
 
   for (int k0 = 0; k0 < Kpad; k0 += KC) {
@@ -116,28 +119,27 @@ void conv2d_fwd_kernel(
     }
     __syncthreads();
     for (int ks = 0; ks < 4; ++ks) {
-      int co_a = band + l16;            // A row: this wave's co band
-      bf16x8c a = (n0 + co_a < Cout)
-          ? *(const bf16x8c*)(wp + (long)(n0 + co_a) * Kpad + k0 + ks * 32
-                              + lq * 8)
-          : (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
-      for (int nt = 0; nt < 4; ++nt) {
-        bf16x8c bi = clds8(lds, cswz(nt * 16 + l16, ks * 32 + lq * 8));
-        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bi, acc[nt],
+      bf16x8c bi = clds8(lds, cswz(band + l16, ks * 32 + lq * 8));
+      for (int mt = 0; mt < 4; ++mt) {
+        int co = n0 + mt * 16 + l16;
+        bf16x8c a = (co < Cout)
+            ? *(const bf16x8c*)(wp + (long)co * Kpad + k0 + ks * 32 + lq * 8)
+            : (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
+        acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bi, acc[mt],
                                                           0, 0, 0);
       }
     }
   }
-  // epilogue: acc rows = co (band + lq*4 + r), cols = px (nt*16 + l16):
-  // 16-lane groups store 32 contiguous bytes
-  for (int r = 0; r < 4; ++r) {
-    int co = n0 + band + lq * 4 + r;
-    if (co >= Cout) continue;
-    float bv = bias ? bias[co] : 0.f;
-    for (int nt = 0; nt < 4; ++nt) {
-      int p = p0 + nt * 16 + l16;
+  // epilogue: acc[mt] rows = co (mt*16 + lq*4 + r), cols = px (this wave's
+  // band + l16): 16-lane groups store 32 contiguous bytes
+  for (int mt = 0; mt < 4; ++mt) {
+    for (int r = 0; r < 4; ++r) {
+      int co = n0 + mt * 16 + lq * 4 + r;
+      if (co >= Cout) continue;
+      float bv = bias ? bias[co] : 0.f;
+      int p = p0 + band + l16;
       if (p >= HW) continue;
-      float v = acc[nt][r] + bv;
+      float v = acc[mt][r] + bv;
       if (relu) v = fmaxf(v, 0.f);
       out[(b * Cout + co) * HW + p] = __float2bfloat16(v);
     }

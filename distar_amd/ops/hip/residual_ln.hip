@@ -19,7 +19,10 @@ __device__ __forceinline__ float warp_sum(float v) {
   return v;
 }
 
-// C must be a multiple of 4 (256 / 1024 in practice); EPW = elems per lane.
+// Specialized to C=256 (EPW=4 compile-time): with a runtime elems-per-lane
+// the per-thread arrays are runtime-indexed and spill to scratch memory.
+#define LN_C 256
+#define EPW 4
 extern "C" __global__ void residual_ln_fwd_kernel(
     const __hip_bfloat16* __restrict__ x,
     const __hip_bfloat16* __restrict__ a,      // nullptr: plain LN
@@ -29,13 +32,13 @@ extern "C" __global__ void residual_ln_fwd_kernel(
     float* __restrict__ mean_out, float* __restrict__ rstd_out,
     long R, int C, float eps) {
   const int lane = threadIdx.x & 63;
-  const int epw = C / 64;                       // elems per lane (4 or 16)
+  const int epw = EPW;
   for (long row = blockIdx.x * 4 + (threadIdx.x >> 6); row < R;
        row += (long)gridDim.x * 4) {
     const __hip_bfloat16* xr = x + row * C;
     const __hip_bfloat16* ar = a ? a + row * C : nullptr;
-    float vals[16];
-    __hip_bfloat16 xv[16], av[16];
+    float vals[EPW];
+    __hip_bfloat16 xv[EPW], av[EPW];
     __builtin_memcpy(xv, xr + lane * epw, epw * 2);   // one vector load
     if (ar) __builtin_memcpy(av, ar + lane * epw, epw * 2);
     float sum = 0.f;
@@ -53,7 +56,7 @@ extern "C" __global__ void residual_ln_fwd_kernel(
     }
     var = warp_sum(var) / C;
     float rstd = rsqrtf(var + eps);
-    __hip_bfloat16 yv[16], sv[16];
+    __hip_bfloat16 yv[EPW], sv[EPW];
     for (int i = 0; i < epw; ++i) {
       int c = lane * epw + i;
       sv[i] = __float2bfloat16(vals[i]);
@@ -79,18 +82,18 @@ extern "C" __global__ void residual_ln_bwd_kernel(
     float* __restrict__ dw, float* __restrict__ db,
     long R, int C) {
   const int lane = threadIdx.x & 63;
-  const int epw = C / 64;
+  const int epw = EPW;
   extern __shared__ float lds[];                // [4 waves][2*C] partials
   // each (lane, i) owns column lane*epw+i in EVERY row it visits: the
   // LN-param partials accumulate in registers, no per-element atomics
-  float dwacc[16] = {}, dbacc[16] = {};
+  float dwacc[EPW] = {}, dbacc[EPW] = {};
   for (long row = blockIdx.x * 4 + (threadIdx.x >> 6); row < R;
        row += (long)gridDim.x * 4) {
     const __hip_bfloat16* dyr = dy + row * C;
     const __hip_bfloat16* sr = s + row * C;
     float m = mean[row], r = rstd[row];
-    float g[16], xh[16];
-    __hip_bfloat16 dyv_v[16], sv_v[16];
+    float g[EPW], xh[EPW];
+    __hip_bfloat16 dyv_v[EPW], sv_v[EPW];
     __builtin_memcpy(dyv_v, dyr + lane * epw, epw * 2);
     __builtin_memcpy(sv_v, sr + lane * epw, epw * 2);
     float s1 = 0.f, s2 = 0.f;
@@ -108,7 +111,7 @@ extern "C" __global__ void residual_ln_bwd_kernel(
     }
     s1 = warp_sum(s1) / C;
     s2 = warp_sum(s2) / C;
-    __hip_bfloat16 drv[16];
+    __hip_bfloat16 drv[EPW];
     for (int i = 0; i < epw; ++i)
       drv[i] = __float2bfloat16((g[i] - s1 - xh[i] * s2) * r);
     __builtin_memcpy(dsum + row * C + lane * epw, drv, epw * 2);

@@ -37,7 +37,7 @@ def fused_residual_ln(x, residual, ln):
     """LN(x + residual) (residual may be None) through the HIP kernel when
     on GPU in bf16 with a supported width; eager otherwise."""
     C = x.shape[-1]
-    if (x.is_cuda and x.dtype == torch.bfloat16 and C % 64 == 0 and C <= 1024
+    if (x.is_cuda and x.dtype == torch.bfloat16 and C == 256
             and (residual is None or residual.dtype == torch.bfloat16)
             and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'
             and os.environ.get('DISTAR_AMD_FUSED_LN') != '0'):
