@@ -638,10 +638,11 @@ torch::Tensor conv2d_wgrad(torch::Tensor input, torch::Tensor dout,
   auto dwp = torch::zeros({Kpad, Cout},
                           input.options().dtype(torch::kFloat32));
   // pick images-per-block so total blocks lands in a healthy range
-  int kt = (int)((Kpad + 63) / 64), nt = (int)((Cout + 63) / 64);
-  long want_z = 6144 / std::max(1, kt * nt);
+  int64_t K_real = Cin * KH * KW;
+  int kt = (int)((std::min<int64_t>(Kpad, (K_real + 31) / 32 * 32) + 63) / 64);
+  long want_z = 6144 / std::max(1, kt);
   int ipb = (int)std::max<long>(1, (B + want_z - 1) / std::max<long>(1, want_z));
-  dim3 grid(kt, nt, (unsigned)((B + ipb - 1) / ipb));
+  dim3 grid(kt, 1, (unsigned)((B + ipb - 1) / ipb));
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(conv2d_wgrad_kernel, grid, dim3(256), 64 * 256,
                      stream.stream(), bfp(input), bfp(dout),

@@ -84,11 +84,15 @@ void conv2d_fwd_kernel(
   f32x4c acc[4];                        // [cout m-tile]
   for (int nt = 0; nt < 4; ++nt) acc[nt] = (f32x4c){0, 0, 0, 0};
 
-  for (int k0 = 0; k0 < Kpad; k0 += KC) {
+  // skip whole chunks/k-rows that are pure padding (e.g. the 56->128 pad
+  // of the 1x1 projection would otherwise double the staged volume)
+  const int k_hi = min(Kpad, (K_real + 31) / 32 * 32);
+  for (int k0 = 0; k0 < k_hi; k0 += KC) {
+    const int krows = min(k_hi - k0, KC);
     __syncthreads();
     // k-major staging: ci/dy/dx computed once per k row; pixel groups of 8
     // are one vector load when the run stays inside a map row
-    for (int task = tid; task < KC * 8; task += 256) {
+    for (int task = tid; task < krows * 8; task += 256) {
       int kk = task >> 3, g = task & 7;
       int k = k0 + kk;
       __hip_bfloat16 vals[8] = {};
@@ -118,28 +122,29 @@ void conv2d_fwd_kernel(
         *(__hip_bfloat16*)(lds + cswz(g * 8 + j, kk)) = vals[j];
     }
     __syncthreads();
-    for (int ks = 0; ks < 4; ++ks) {
-      bf16x8c bi = clds8(lds, cswz(band + l16, ks * 32 + lq * 8));
-      for (int mt = 0; mt < 4; ++mt) {
-        int co = n0 + mt * 16 + l16;
-        bf16x8c a = (co < Cout)
-            ? *(const bf16x8c*)(wp + (long)co * Kpad + k0 + ks * 32 + lq * 8)
-            : (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
-        acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bi, acc[mt],
+    const int ks_count = min(k_hi - k0, KC) / 32;
+    for (int ks = 0; ks < ks_count; ++ks) {
+      int co_a = n0 + band + l16;       // A row: this wave's co band
+      bf16x8c a = (co_a < Cout)
+          ? *(const bf16x8c*)(wp + (long)co_a * Kpad + k0 + ks * 32 + lq * 8)
+          : (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
+      for (int nt = 0; nt < 4; ++nt) {
+        bf16x8c bi = clds8(lds, cswz(nt * 16 + l16, ks * 32 + lq * 8));
+        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bi, acc[nt],
                                                           0, 0, 0);
       }
     }
   }
-  // epilogue: acc[mt] rows = co (mt*16 + lq*4 + r), cols = px (this wave's
-  // band + l16): 16-lane groups store 32 contiguous bytes
-  for (int mt = 0; mt < 4; ++mt) {
-    for (int r = 0; r < 4; ++r) {
-      int co = n0 + mt * 16 + lq * 4 + r;
-      if (co >= Cout) continue;
-      float bv = bias ? bias[co] : 0.f;
-      int p = p0 + band + l16;
+  // epilogue: acc rows = co (band + lq*4 + r), cols = px (nt*16 + l16):
+  // 16-lane groups store 32 contiguous bytes
+  for (int r = 0; r < 4; ++r) {
+    int co = n0 + band + lq * 4 + r;
+    if (co >= Cout) continue;
+    float bv = bias ? bias[co] : 0.f;
+    for (int nt = 0; nt < 4; ++nt) {
+      int p = p0 + nt * 16 + l16;
       if (p >= HW) continue;
-      float v = acc[mt][r] + bv;
+      float v = acc[nt][r] + bv;
       if (relu) v = fmaxf(v, 0.f);
       out[(b * Cout + co) * HW + p] = __float2bfloat16(v);
     }
@@ -156,10 +161,10 @@ void conv2d_wgrad_kernel(
     int KH, int KW, int padH, int padW, int Kpad, int ipb) {
   const int HW = H * W;
   const int k_base = blockIdx.x * CTILE;        // k tile (rows of dW)
-  const int n0 = blockIdx.y * 64;               // co tile
   const int b_base = blockIdx.z * ipb;          // image range
-  if (k_base >= Kpad || n0 >= Cout) return;
   const int K_real = Cin * KH * KW;
+  if (k_base >= K_real) return;                 // pure-padding k tile
+  const int n_tiles_co = (Cout + 15) / 16;      // <= 8 (Cout <= 128)
 
   extern __shared__ char lds[];                 // A^T tile [64 k][64 px] 8 KB
   const int lane = threadIdx.x & 63;
@@ -169,8 +174,8 @@ void conv2d_wgrad_kernel(
   const int band = wave * 16;                   // k band of this wave
   const int tid = threadIdx.x;
 
-  f32x4c acc[4];
-  for (int nt = 0; nt < 4; ++nt) acc[nt] = (f32x4c){0, 0, 0, 0};
+  f32x4c acc[8];                        // co tiles (<= 128 cout)
+  for (int nt = 0; nt < 8; ++nt) acc[nt] = (f32x4c){0, 0, 0, 0};
 
   for (int bi = 0; bi < ipb && b_base + bi < B; ++bi) {
     const long b = b_base + bi;
@@ -212,8 +217,8 @@ void conv2d_wgrad_kernel(
       __syncthreads();
       for (int ks = 0; ks < 4; ++ks) {
         bf16x8c a = clds8(lds, cswz(band + l16, ks * 32 + lq * 8));
-        for (int nt = 0; nt < 4; ++nt) {
-          int co = n0 + nt * 16 + l16;
+        for (int nt = 0; nt < n_tiles_co; ++nt) {
+          int co = nt * 16 + l16;
           bf16x8c bdo = (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
           if (co < Cout) {
             int p = p0 + ks * 32 + lq * 8;
@@ -234,8 +239,8 @@ void conv2d_wgrad_kernel(
     }
   }
   // accumulate into dwp (Kpad, Cout) fp32
-  for (int nt = 0; nt < 4; ++nt) {
-    int co = n0 + nt * 16 + l16;
+  for (int nt = 0; nt < n_tiles_co; ++nt) {
+    int co = nt * 16 + l16;
     if (co >= Cout) continue;
     for (int r = 0; r < 4; ++r) {
       int k = k_base + band + lq * 4 + r;
