@@ -101,20 +101,22 @@ void conv2d_fwd_kernel(
         int off = k % (KH * KW);
         int dy = off / KW - padH, dx = off % KW - padW;
         int pbase = p0 + g * 8;
-        int y = pbase / W + dy;
-        int x = pbase % W + dx;
+        int y0 = pbase / W, x0 = pbase % W;  // the only division per group
+        int y = y0 + dy;
+        int x = x0 + dx;
         const __hip_bfloat16* src = inb + ((long)ci * H + y) * W + x;
-        if (pbase + 7 < HW && (pbase % W) + 7 < W && y >= 0 && y < H &&
+        if (pbase + 7 < HW && x0 + 7 < W && y >= 0 && y < H &&
             x >= 0 && x + 7 < W) {
           __builtin_memcpy(vals, src, 16);
         } else {
+          const __hip_bfloat16* cib = inb + (long)ci * H * W;
           for (int j = 0; j < 8; ++j) {
-            int p = pbase + j;
-            if (p < HW) {
-              int yj = p / W + dy, xj = p % W + dx;
+            if (pbase + j < HW) {
+              int yj = y0 + dy, xj = x0 + dx;
               if (yj >= 0 && yj < H && xj >= 0 && xj < W)
-                vals[j] = inb[((long)ci * H + yj) * W + xj];
+                vals[j] = cib[(long)yj * W + xj];
             }
+            if (++x0 == W) { x0 = 0; ++y0; }  // incremental, no div/mod
           }
         }
       }
@@ -194,20 +196,22 @@ void conv2d_wgrad_kernel(
           int off = k % (KH * KW);
           int dy = off / KW - padH, dx = off % KW - padW;
           int pbase = p0 + g * 8;
-          int y = pbase / W + dy;
-          int x = pbase % W + dx;
+          int y0 = pbase / W, x0 = pbase % W;  // the only division per group
+          int y = y0 + dy;
+          int x = x0 + dx;
           const __hip_bfloat16* src = inb + ((long)ci * H + y) * W + x;
-          if (pbase + 7 < HW && (pbase % W) + 7 < W && y >= 0 && y < H &&
+          if (pbase + 7 < HW && x0 + 7 < W && y >= 0 && y < H &&
               x >= 0 && x + 7 < W) {
             __builtin_memcpy(vals, src, 16);
           } else {
+            const __hip_bfloat16* cib = inb + (long)ci * H * W;
             for (int j = 0; j < 8; ++j) {
-              int p = pbase + j;
-              if (p < HW) {
-                int yj = p / W + dy, xj = p % W + dx;
+              if (pbase + j < HW) {
+                int yj = y0 + dy, xj = x0 + dx;
                 if (yj >= 0 && yj < H && xj >= 0 && xj < W)
-                  vals[j] = inb[((long)ci * H + yj) * W + xj];
+                  vals[j] = cib[(long)yj * W + xj];
               }
+              if (++x0 == W) { x0 = 0; ++y0; }  // incremental, no div/mod
             }
           }
         }
