@@ -211,9 +211,10 @@ class SpatialEncoder(nn.Module):
         # layout) and differentiable wrt both weight slices and scatter_map
         conv = self.project[0]
         w = conv.weight
-        out = torch.nn.functional.conv2d(const, w[:, :n_const]) + \
-            torch.nn.functional.conv2d(scatter_map.to(const.dtype), w[:, n_const:],
-                                       bias=conv.bias)
+        from ...ops.conv2d import conv2d as hip_conv2d
+        out = hip_conv2d(const, w[:, :n_const].contiguous()) + \
+            hip_conv2d(scatter_map.to(const.dtype),
+                       w[:, n_const:].contiguous(), bias=conv.bias)
         for layer in list(self.project)[1:]:        # norm/act of the block
             out = layer(out)
         map_skip = []

@@ -109,6 +109,22 @@ class _MaxPool2x2Fn(torch.autograd.Function):
                                   H, W)
 
 
+def conv2d(x, weight, bias=None, padding=0):
+    """Functional conv through the HIP kernels when the shape qualifies
+    (1x1 p0 / 3x3 p1, stride 1, GPU bf16); F.conv2d otherwise.  Used by the
+    spatial encoder's split projection (sliced-weight calls that bypass the
+    Conv2dHIP module)."""
+    kh, kw = weight.shape[2], weight.shape[3]
+    pad = (padding, padding) if isinstance(padding, int) else tuple(padding)
+    if (x.is_cuda and x.dtype == torch.bfloat16 and (kh, kw) in ((1, 1), (3, 3))
+            and pad == (kh // 2, kw // 2)
+            and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'
+            and os.environ.get('DISTAR_AMD_CONV') != '0'):
+        return _Conv2dFn.apply(x.contiguous(), weight, bias, kh, kw,
+                               kh // 2, kw // 2)
+    return F.conv2d(x, weight, bias, padding=padding)
+
+
 def max_pool2x2(x):
     """Functional 2x2/2 maxpool through the HIP kernels when possible."""
     if (x.is_cuda and x.dtype == torch.bfloat16

@@ -594,6 +594,9 @@ std::vector<torch::Tensor> residual_ln_bwd(
 extern "C" __global__ void conv2d_fwd_kernel(
     const __hip_bfloat16*, const __hip_bfloat16*, const float*,
     __hip_bfloat16*, int, int, int, int, int, int, int, int, int, int, int);
+extern "C" __global__ void conv2d_small_fwd_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+    __hip_bfloat16*, int, int, int, int, int, int, int, int, int, int, int);
 extern "C" __global__ void conv2d_wgrad_kernel(
     const __hip_bfloat16*, const __hip_bfloat16*, float*,
     int, int, int, int, int, int, int, int, int, int, int);
@@ -617,8 +620,20 @@ torch::Tensor conv2d_fwd(torch::Tensor input, torch::Tensor wp,
   const float* bp = nullptr;
   if (bias.has_value()) bp = bias->data_ptr<float>();
   int HW = (int)(H * W);
-  dim3 grid((HW + 63) / 64, (unsigned)((Cout + 63) / 64), (unsigned)B);
   auto stream = c10::hip::getCurrentHIPStream();
+  if (Cout < 16) {
+    // memory-bound tiny-channel conv (e.g. the 32->1 location head):
+    // direct VALU kernel, one thread per output pixel
+    long total = B * Cout * (long)HW;
+    int blocks = (int)std::min<long>((total + 255) / 256, 8192);
+    hipLaunchKernelGGL(conv2d_small_fwd_kernel, dim3(blocks), dim3(256), 0,
+                       stream.stream(), bfp(input), bfp(wp), bp,
+                       bfp_mut(out), (int)B, (int)Cin, (int)Cout,
+                       (int)H, (int)W, (int)KH, (int)KW, (int)padH,
+                       (int)padW, (int)Kpad, relu ? 1 : 0);
+    return out;
+  }
+  dim3 grid((HW + 63) / 64, (unsigned)((Cout + 63) / 64), (unsigned)B);
   hipLaunchKernelGGL(conv2d_fwd_kernel, grid, dim3(256), 64 * 256,
                      stream.stream(), bfp(input), bfp(wp), bp, bfp_mut(out),
                      (int)B, (int)Cin, (int)Cout, (int)H, (int)W,

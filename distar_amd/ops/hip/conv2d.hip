@@ -125,15 +125,19 @@ void conv2d_fwd_kernel(
     }
     __syncthreads();
     const int ks_count = min(k_hi - k0, KC) / 32;
-    for (int ks = 0; ks < ks_count; ++ks) {
-      int co_a = n0 + band + l16;       // A row: this wave's co band
-      bf16x8c a = (co_a < Cout)
-          ? *(const bf16x8c*)(wp + (long)co_a * Kpad + k0 + ks * 32 + lq * 8)
-          : (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
-      for (int nt = 0; nt < 4; ++nt) {
-        bf16x8c bi = clds8(lds, cswz(nt * 16 + l16, ks * 32 + lq * 8));
-        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bi, acc[nt],
-                                                          0, 0, 0);
+    // waves whose whole co band is beyond Cout only help stage (the 32->1
+    // location-head conv would otherwise waste 3/4 of the block's MFMAs)
+    if (n0 + band < Cout) {
+      for (int ks = 0; ks < ks_count; ++ks) {
+        int co_a = n0 + band + l16;     // A row: this wave's co band
+        bf16x8c a = (co_a < Cout)
+            ? *(const bf16x8c*)(wp + (long)co_a * Kpad + k0 + ks * 32 + lq * 8)
+            : (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
+        for (int nt = 0; nt < 4; ++nt) {
+          bf16x8c bi = clds8(lds, cswz(nt * 16 + l16, ks * 32 + lq * 8));
+          acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bi, acc[nt],
+                                                            0, 0, 0);
+        }
       }
     }
   }
@@ -303,5 +307,44 @@ extern "C" __global__ void maxpool2x2_bwd_kernel(
       if (idx[o] == a) g = __bfloat162float(dout[o]);
     }
     din[i] = __float2bfloat16(g);
+  }
+}
+
+
+// Direct conv for small Cout (< 16): one thread per output pixel, VALU
+// MACs — these convs are memory-bound and MFMA n-tiles would waste >=16x.
+extern "C" __global__ void conv2d_small_fwd_kernel(
+    const __hip_bfloat16* __restrict__ input,   // (B, Cin, H, W)
+    const __hip_bfloat16* __restrict__ wp,      // (Cout, Kpad)
+    const float* __restrict__ bias,
+    __hip_bfloat16* __restrict__ out,           // (B, Cout, H, W)
+    int B, int Cin, int Cout, int H, int W,
+    int KH, int KW, int padH, int padW, int Kpad, int relu) {
+  const long HW = (long)H * W;
+  const long total = (long)B * Cout * HW;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int x = i % W;
+    int y = (i / W) % H;
+    int co = (i / HW) % Cout;
+    long b = i / (HW * Cout);
+    const __hip_bfloat16* inb = input + b * Cin * HW;
+    const __hip_bfloat16* wrow = wp + (long)co * Kpad;
+    float acc = bias ? bias[co] : 0.f;
+    int k = 0;
+    for (int ci = 0; ci < Cin; ++ci) {
+      const __hip_bfloat16* cib = inb + (long)ci * HW;
+      for (int dy = -padH; dy < KH - padH; ++dy) {
+        int yy = y + dy;
+        for (int dx = -padW; dx < KW - padW; ++dx, ++k) {
+          int xx = x + dx;
+          if (yy >= 0 && yy < H && xx >= 0 && xx < W)
+            acc += __bfloat162float(wrow[k]) *
+                   __bfloat162float(cib[(long)yy * W + xx]);
+        }
+      }
+    }
+    if (relu) acc = fmaxf(acc, 0.f);
+    out[i] = __float2bfloat16(acc);
   }
 }
