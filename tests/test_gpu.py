@@ -529,6 +529,7 @@ def test_fused_residual_ln_matches_eager():
     (32, 64, (76, 80), 3),       # downsample convs
     (128, 128, (19, 20), 3),     # ResBlocks
     (132, 128, (19, 20), 1),     # location head 1x1
+    (32, 1, (76, 80), 3),        # tiny-Cout direct path (location head out)
 ])
 def test_conv2d_hip_matches_eager(cin, cout, hw, kh):
     """K4 MFMA implicit-GEMM conv vs fp32 F.conv2d (fwd + both bwds)."""
