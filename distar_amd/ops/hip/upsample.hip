@@ -38,24 +38,35 @@ __device__ inline void taps2x(int d, int n_src, int& s0, int& s1, float& w0) {
   w0 = 1.f - frac;
 }
 
+// One thread per 8 consecutive output pixels: the 64-bit div/mod and the
+// y-taps amortize 8x (at 1.6G output elements the per-element divisions
+// dominated), and the 8 stores become one 16B write for bf16.
 template <typename T>
 __device__ void upsample2x_fwd(const T* __restrict__ in, T* __restrict__ out,
                                int NC, int H, int W) {
-  const int H2 = H * 2, W2 = W * 2;
-  const long total = (long)NC * H2 * W2;
-  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
-    int x = idx % W2;
-    int y = (idx / W2) % H2;
-    long nc = idx / ((long)W2 * H2);
-    int x0, x1, y0, y1;
-    float wx0, wy0;
-    taps2x(x, W, x0, x1, wx0);
+  const int H2 = H * 2, W2 = W * 2;     // W2 is a multiple of 8 here
+  const int WG = W2 / 8;
+  const long total_g = (long)NC * H2 * WG;
+  for (long g = (long)blockIdx.x * blockDim.x + threadIdx.x; g < total_g;
+       g += (long)gridDim.x * blockDim.x) {
+    int xg = (int)(g % WG) * 8;
+    int y = (int)((g / WG) % H2);
+    long nc = g / ((long)WG * H2);
+    int y0, y1;
+    float wy0;
     taps2x(y, H, y0, y1, wy0);
-    const T* base = in + nc * H * W;
-    float v = wy0 * (wx0 * ld(base + y0 * W + x0) + (1.f - wx0) * ld(base + y0 * W + x1))
-        + (1.f - wy0) * (wx0 * ld(base + y1 * W + x0) + (1.f - wx0) * ld(base + y1 * W + x1));
-    st(out + idx, v);
+    const T* r0 = in + nc * (long)H * W + (long)y0 * W;
+    const T* r1 = in + nc * (long)H * W + (long)y1 * W;
+    T vals[8];
+    for (int j = 0; j < 8; ++j) {
+      int x0, x1;
+      float wx0;
+      taps2x(xg + j, W, x0, x1, wx0);
+      float v = wy0 * (wx0 * ld(r0 + x0) + (1.f - wx0) * ld(r0 + x1)) +
+                (1.f - wy0) * (wx0 * ld(r1 + x0) + (1.f - wx0) * ld(r1 + x1));
+      st(vals + j, v);
+    }
+    __builtin_memcpy(out + (nc * (long)H2 + y) * W2 + xg, vals, sizeof(vals));
   }
 }
 
@@ -65,14 +76,18 @@ __device__ void upsample2x_fwd(const T* __restrict__ in, T* __restrict__ out,
 template <typename T>
 __device__ void upsample2x_bwd(const T* __restrict__ gout, T* __restrict__ gin,
                                int NC, int H, int W) {
-  const int H2 = H * 2, W2 = W * 2;
-  const long total = (long)NC * H * W;
-  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (long)gridDim.x * blockDim.x) {
-    int sx = idx % W;
-    int sy = (idx / W) % H;
-    long nc = idx / ((long)W * H);
+  const int H2 = H * 2, W2 = W * 2;     // W is a multiple of 4 here
+  const int WG = W / 4;
+  const long total_g = (long)NC * H * WG;
+  for (long g = (long)blockIdx.x * blockDim.x + threadIdx.x; g < total_g;
+       g += (long)gridDim.x * blockDim.x) {
+    int sxg = (int)(g % WG) * 4;
+    int sy = (int)((g / WG) % H);
+    long nc = g / ((long)WG * H);
     const T* base = gout + nc * (long)H2 * W2;
+    T outv[4];
+    for (int j = 0; j < 4; ++j) {
+    int sx = sxg + j;
     float acc = 0.f;
     // destination rows/cols that can reference (sy, sx): d in [2s-1, 2s+2]
     for (int dy = sy * 2 - 1; dy <= sy * 2 + 2; ++dy) {
@@ -90,7 +105,9 @@ __device__ void upsample2x_bwd(const T* __restrict__ gout, T* __restrict__ gin,
         acc += wy * wx * ld(base + (long)dy * W2 + dx);
       }
     }
-    st(gin + idx, acc);
+    st(outv + j, acc);
+    }
+    __builtin_memcpy(gin + (nc * (long)H + sy) * W + sxg, outv, sizeof(outv));
   }
 }
 

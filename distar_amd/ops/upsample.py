@@ -22,6 +22,8 @@ class _Upsample2x(torch.autograd.Function):
 
 def upsample2x_bilinear(x):
     import os
-    if x.is_cuda and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1':
+    # kernel processes 8-output / 4-source pixel groups per thread
+    if (x.is_cuda and x.shape[-1] % 4 == 0
+            and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'):
         return _Upsample2x.apply(x)
     return F.interpolate(x, scale_factor=2., mode='bilinear')
