@@ -557,3 +557,44 @@ def test_play_path_agent_over_fake_sc2(monkeypatch):
     stepped = any(r.HasField('step') for f in fakes for r in f.requests)
     assert stepped
     env.close()
+
+
+def test_replay_actor_worker_decodes_and_pushes(tmp_path, monkeypatch):
+    """ReplayActor worker over the fake SC2: decodes each (replay, player)
+    and Adapter.pushes the step lists (reference replay_actor.py:10-73)."""
+    from distar_amd.data import replay_actor as ra_mod
+    from distar_amd.data import replay_decoder as rd_mod
+    from distar_amd.utils.config import Config
+    replay = _synthetic_replay(str(tmp_path))
+
+    act1 = P.raw_action(3674, unit_tags=[2000], target_pos=(30., 40.),
+                        game_loop=10)
+
+    def fake_launch(cfg=None, port=None, version=None, **kw):
+        class Proc:
+            def kill(self):
+                pass
+        return Proc(), 23456
+
+    def fake_connect(host, port, resource='/sc2api', timeout=120.0):
+        return FakeSC2(end_after_observes=2, actions_on_observe={1: [act1]})
+
+    monkeypatch.setattr(rd_mod, 'launch_game_process', fake_launch)
+    monkeypatch.setattr(protocol.WebSocket, 'connect',
+                        staticmethod(fake_connect))
+
+    pushes = []
+
+    class FakeAdapter:
+        def __init__(self, cfg):
+            pass
+
+        def push(self, data, token=None, fs_type=None):
+            pushes.append((token, fs_type, len(data)))
+
+    monkeypatch.setattr(ra_mod, 'Adapter', FakeAdapter)
+    ra_mod._worker(Config({'env': {}}), [replay], 0)
+    # both players decoded successfully (1 transformed step each)
+    assert len(pushes) == 2
+    assert all(t == 'replay' and fs == 'nppickle' and n == 1
+               for t, fs, n in pushes)
