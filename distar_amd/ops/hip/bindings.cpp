@@ -655,7 +655,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor input, torch::Tensor dout,
   // pick images-per-block so total blocks lands in a healthy range
   int64_t K_real = Cin * KH * KW;
   int kt = (int)((std::min<int64_t>(Kpad, (K_real + 31) / 32 * 32) + 63) / 64);
-  long want_z = 6144 / std::max(1, kt);
+  long want_z = 32768 / std::max(1, kt);
   int ipb = (int)std::max<long>(1, (B + want_z - 1) / std::max<long>(1, want_z));
   dim3 grid(kt, 1, (unsigned)((B + ipb - 1) / ipb));
   auto stream = c10::hip::getCurrentHIPStream();
