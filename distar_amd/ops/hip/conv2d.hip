@@ -93,39 +93,48 @@ void conv2d_fwd_kernel(
   const int kwin = KH * KW;
   for (int k0 = 0; k0 < k_hi; k0 += KC) {
     const int krows = min(k_hi - k0, KC);
-    const int nkg = krows / 8;
     __syncthreads();
-    // px-major staging: each task fills ONE pixel row's 8-k granule and
-    // stores it as a single 16 B ds_write (conflict-free: granule XOR
-    // spreads banks; writes are row-major like the MFMA-fragment reads)
-    for (int task = tid; task < 64 * nkg; task += 256) {
-      int px = task / nkg;
-      int kg = task - px * nkg;
-      int p = p0 + px;
+    // k-major staging with kk FASTEST across threads: the 8-pixel global
+    // run stays one vector load, and a wave's 64 LDS writes span all 32
+    // banks (kk>>3 varies per lane; px-major order had 16-way conflicts,
+    // px-major single-writes lost the vectorized global reads — PMC r2l/m)
+    for (int task = tid; task < krows * 8; task += 256) {
+      int g = task / krows;               // pixel group (fixed per wave)
+      int kk = task - g * krows;          // k row (fastest across lanes)
+      int k = k0 + kk;
       __hip_bfloat16 vals[8] = {};
-      if (p < HW) {
-        int y0 = p / W, x0 = p - (p / W) * W;
-        const int kb = k0 + kg * 8;
-        for (int j = 0; j < 8; ++j) {
-          int k = kb + j;
-          if (k < K_real) {
-            int ci, dy, dx;
-            if (kwin == 1) {
-              ci = k; dy = 0; dx = 0;
-            } else {                       // 3x3: mul-shift div by 9 / 3
-              ci = (k * 7282) >> 16;
-              int off = k - ci * 9;
-              dy = ((off * 21846) >> 16) - padH;
-              dx = off - ((off * 21846) >> 16) * 3 - padW;
+      if (k < K_real) {
+        int ci, dy, dx;
+        if (kwin == 1) {
+          ci = k; dy = -padH; dx = -padW;
+        } else {                           // 3x3: mul-shift div by 9 / 3
+          ci = (k * 7282) >> 16;
+          int off = k - ci * 9;
+          dy = ((off * 21846) >> 16) - padH;
+          dx = off - ((off * 21846) >> 16) * 3 - padW;
+        }
+        int pbase = p0 + g * 8;
+        int y0 = pbase / W, x0 = pbase - (pbase / W) * W;
+        int y = y0 + dy;
+        int x = x0 + dx;
+        const __hip_bfloat16* src = inb + ((long)ci * H + y) * W + x;
+        if (pbase + 7 < HW && x0 + 7 < W && y >= 0 && y < H &&
+            x >= 0 && x + 7 < W) {
+          __builtin_memcpy(vals, src, 16);
+        } else {
+          const __hip_bfloat16* cib = inb + (long)ci * H * W;
+          for (int j = 0; j < 8; ++j) {
+            if (pbase + j < HW) {
+              int yj = y0 + dy, xj = x0 + dx;
+              if (yj >= 0 && yj < H && xj >= 0 && xj < W)
+                vals[j] = cib[(long)yj * W + xj];
             }
-            int yy = y0 + dy, xx = x0 + dx;
-            if (yy >= 0 && yy < H && xx >= 0 && xx < W)
-              vals[j] = inb[((long)ci * H + yy) * W + xx];
+            if (++x0 == W) { x0 = 0; ++y0; }
           }
         }
       }
-      int gsw = (kg ^ (px & 7)) & 15;
-      __builtin_memcpy(lds + px * 256 + gsw * 16, vals, 16);
+      for (int j = 0; j < 8; ++j)
+        *(__hip_bfloat16*)(lds + cswz(g * 8 + j, kk)) = vals[j];
     }
     __syncthreads();
     const int ks_count = min(k_hi - k0, KC) / 32;
