@@ -711,7 +711,58 @@ torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor Bm) {
   return D;
 }
 
+extern "C" __global__ void multi_norm_sq_kernel(
+    const unsigned long long*, const long*, int, float*);
+extern "C" __global__ void multi_scale_kernel(
+    const unsigned long long*, const long*, int, const float*,
+    float, float, int);
+
+static std::pair<torch::Tensor, int> build_chunk_table(
+    const std::vector<torch::Tensor>& grads) {
+  int n = (int)grads.size();
+  auto meta = torch::empty({2, n}, torch::dtype(torch::kInt64));
+  auto acc = meta.accessor<int64_t, 2>();
+  for (int i = 0; i < n; ++i) {
+    TORCH_CHECK(grads[i].is_cuda() && grads[i].is_contiguous());
+    TORCH_CHECK(grads[i].scalar_type() == torch::kFloat32);
+    acc[0][i] = (int64_t)grads[i].data_ptr();
+    acc[1][i] = grads[i].numel();
+  }
+  return {meta.to(grads[0].device(), /*non_blocking=*/true), n};
+}
+
+torch::Tensor multi_norm_sq(std::vector<torch::Tensor> grads) {
+  TORCH_CHECK(!grads.empty());
+  auto [meta, n] = build_chunk_table(grads);
+  auto out = torch::zeros({}, grads[0].options());
+  auto stream = c10::hip::getCurrentHIPStream();
+  int blocks = std::min(n, 2048);
+  hipLaunchKernelGGL(multi_norm_sq_kernel, dim3(blocks), dim3(256), 0,
+                     stream.stream(),
+                     reinterpret_cast<const unsigned long long*>(
+                         meta[0].data_ptr<int64_t>()),
+                     meta[1].data_ptr<int64_t>(), n, out.data_ptr<float>());
+  return out;
+}
+
+void multi_clip(std::vector<torch::Tensor> grads, torch::Tensor norm_sq,
+                double thresh, double eps) {
+  auto [meta, n] = build_chunk_table(grads);
+  auto stream = c10::hip::getCurrentHIPStream();
+  dim3 grid(64, std::min(n, 1024));
+  hipLaunchKernelGGL(multi_scale_kernel, grid, dim3(256), 0,
+                     stream.stream(),
+                     reinterpret_cast<const unsigned long long*>(
+                         meta[0].data_ptr<int64_t>()),
+                     meta[1].data_ptr<int64_t>(), n,
+                     norm_sq.data_ptr<float>(), (float)thresh, (float)eps, 1);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("multi_norm_sq", &multi_norm_sq,
+        "K14 multi-tensor global sum-of-squares");
+  m.def("multi_clip", &multi_clip,
+        "K14 clip-by-global-norm with a device-resident norm (no host sync)");
   m.def("entity_attn_fwd", &entity_attn_fwd,
         "K1 entity-transformer flash attention forward (bf16 MFMA)");
   m.def("entity_attn_bwd", &entity_attn_bwd,

@@ -6,6 +6,8 @@ On GPU the norm computations use torch's fused multi-tensor paths
 (`torch.nn.utils.clip_grad_norm_` with foreach=True) — one kernel per dtype
 group rather than a per-parameter loop.
 """
+import os
+
 import torch
 
 
@@ -35,6 +37,19 @@ class GradClip:
                     self.norm_type)
             return total.item()
         if self.clip_type == 'pytorch_norm':
+            grads = [p.grad for p in params]
+            if (self.norm_type == 2 and grads[0].is_cuda
+                    and all(g.is_cuda and g.dtype == torch.float32
+                            and g.is_contiguous() for g in grads)
+                    and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'):
+                # K14 hand-written multi-tensor path: one norm kernel + one
+                # clip kernel with a DEVICE-resident norm (the scale never
+                # round-trips to the host inside the step)
+                from ..ops import hip_ext
+                ops = hip_ext.require()
+                norm_sq = ops.multi_norm_sq(grads)
+                ops.multi_clip(grads, norm_sq, float(self.threshold), 1e-6)
+                return float(norm_sq.sqrt())
             total = torch.nn.utils.clip_grad_norm_(params, self.threshold,
                                                    norm_type=self.norm_type,
                                                    foreach=True)

@@ -31,6 +31,7 @@ SOURCES = [
     os.path.join(HIP_DIR, 'scatter.hip'),
     os.path.join(HIP_DIR, 'residual_ln.hip'),
     os.path.join(HIP_DIR, 'conv2d.hip'),
+    os.path.join(HIP_DIR, 'multi_tensor.hip'),
     os.path.join(HIP_DIR, 'bindings.cpp'),
 ]
 

@@ -583,3 +583,21 @@ def test_maxpool2x2_hip_matches_eager():
         os.environ.pop('DISTAR_AMD_CONV', None)
     torch.testing.assert_close(out_h.float(), out_e.float())
     torch.testing.assert_close(x_h.grad.float(), x_e.grad.float())
+
+
+def test_multi_tensor_norm_clip_matches_eager():
+    """K14 multi-tensor global-norm + clip vs clip_grad_norm_."""
+    from distar_amd.ops import hip_ext
+    ops = hip_ext.require()
+    torch.manual_seed(7)
+    grads = [torch.randn(n, device='cuda') * 3
+             for n in (17, 1024, 100003, 4096)]
+    ref = [g.clone() for g in grads]
+    norm_sq = ops.multi_norm_sq(grads)
+    expected_norm = torch.norm(torch.cat([g.view(-1) for g in ref]))
+    torch.testing.assert_close(norm_sq.sqrt(), expected_norm,
+                               rtol=1e-5, atol=1e-5)
+    ops.multi_clip(grads, norm_sq, 1.0, 1e-6)
+    scale = min(1.0, 1.0 / (float(expected_norm) + 1e-6))
+    for g, r in zip(grads, ref):
+        torch.testing.assert_close(g, r * scale, rtol=1e-5, atol=1e-6)
