@@ -633,7 +633,7 @@ torch::Tensor conv2d_fwd(torch::Tensor input, torch::Tensor wp,
                        (int)padW, (int)Kpad, relu ? 1 : 0);
     return out;
   }
-  dim3 grid((HW + 63) / 64, (unsigned)((Cout + 63) / 64), (unsigned)B);
+  dim3 grid((HW + 63) / 64, (unsigned)((Cout + 127) / 128), (unsigned)B);
   hipLaunchKernelGGL(conv2d_fwd_kernel, grid, dim3(256), 64 * 256,
                      stream.stream(), bfp(input), bfp(wp), bp, bfp_mut(out),
                      (int)B, (int)Cin, (int)Cout, (int)H, (int)W,

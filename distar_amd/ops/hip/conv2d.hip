@@ -59,10 +59,11 @@ void conv2d_fwd_kernel(
     int KH, int KW, int padH, int padW, int Kpad, int relu) {
   const int HW = H * W;
   const int p0 = blockIdx.x * CTILE;
-  const int n0 = blockIdx.y * 64;
+  const int n0 = blockIdx.y * 128;      // one block covers up to 128 cout
   const long b = blockIdx.z;
   if (p0 >= HW || n0 >= Cout) return;
   const int K_real = Cin * KH * KW;
+  const int wrecip = (1048576 + W - 1) / W;     // 2^20/W, exact for p<2^31/recip
 
   extern __shared__ char lds[];                 // A tile [64px][KC] 8 KB
   const __hip_bfloat16* inb = input + b * Cin * HW;
@@ -81,8 +82,11 @@ void conv2d_fwd_kernel(
   // Each WAVE owns one 16-pixel band and ALL 64 cout rows (4 m-tiles):
   // one LDS B-fragment feeds 4 MFMAs (the A-operands are cheap global
   // reads), 4:1 MFMA:ds_read instead of 1:1.
-  f32x4c acc[4];                        // [cout m-tile]
-  for (int nt = 0; nt < 4; ++nt) acc[nt] = (f32x4c){0, 0, 0, 0};
+  // two 64-cout halves per block: one staged im2col tile feeds 2x the
+  // MFMAs (PMC r2n: staging VALU dominated at 59 insts per MFMA)
+  f32x4c acc[2][4];
+  for (int h = 0; h < 2; ++h)
+    for (int nt = 0; nt < 4; ++nt) acc[h][nt] = (f32x4c){0, 0, 0, 0};
 
   // skip whole chunks/k-rows that are pure padding (e.g. the 56->128 pad
   // of the 1x1 projection would otherwise double the staged volume)
@@ -98,8 +102,9 @@ void conv2d_fwd_kernel(
     // run stays one vector load, and a wave's 64 LDS writes span all 32
     // banks (kk>>3 varies per lane; px-major order had 16-way conflicts,
     // px-major single-writes lost the vectorized global reads — PMC r2l/m)
+    const int krecip = (65536 + krows - 1) / krows;
     for (int task = tid; task < krows * 8; task += 256) {
-      int g = task / krows;               // pixel group (fixed per wave)
+      int g = (task * krecip) >> 16;      // pixel group (fixed per wave)
       int kk = task - g * krows;          // k row (fastest across lanes)
       int k = k0 + kk;
       __hip_bfloat16 vals[8] = {};
@@ -114,7 +119,8 @@ void conv2d_fwd_kernel(
           dx = off - ((off * 21846) >> 16) * 3 - padW;
         }
         int pbase = p0 + g * 8;
-        int y0 = pbase / W, x0 = pbase - (pbase / W) * W;
+        int y0 = (int)(((long)pbase * wrecip) >> 20);
+        int x0 = pbase - y0 * W;
         int y = y0 + dy;
         int x = x0 + dx;
         const __hip_bfloat16* src = inb + ((long)ci * H + y) * W + x;
@@ -138,34 +144,37 @@ void conv2d_fwd_kernel(
     }
     __syncthreads();
     const int ks_count = min(k_hi - k0, KC) / 32;
-    // waves whose whole co band is beyond Cout only help stage (the 32->1
-    // location-head conv would otherwise waste 3/4 of the block's MFMAs)
-    if (n0 + band < Cout) {
-      for (int ks = 0; ks < ks_count; ++ks) {
-        int co_a = n0 + band + l16;     // A row: this wave's co band
+    for (int ks = 0; ks < ks_count; ++ks) {
+      bf16x8c bi[4];
+      for (int nt = 0; nt < 4; ++nt)
+        bi[nt] = clds8(lds, cswz(nt * 16 + l16, ks * 32 + lq * 8));
+      for (int h = 0; h < 2; ++h) {
+        // waves/halves whose whole co band is beyond Cout skip compute
+        if (n0 + h * 64 + band >= Cout) continue;
+        int co_a = n0 + h * 64 + band + l16;
         bf16x8c a = (co_a < Cout)
             ? *(const bf16x8c*)(wp + (long)co_a * Kpad + k0 + ks * 32 + lq * 8)
             : (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
-        for (int nt = 0; nt < 4; ++nt) {
-          bf16x8c bi = clds8(lds, cswz(nt * 16 + l16, ks * 32 + lq * 8));
-          acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bi, acc[nt],
-                                                            0, 0, 0);
-        }
+        for (int nt = 0; nt < 4; ++nt)
+          acc[h][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, bi[nt], acc[h][nt], 0, 0, 0);
       }
     }
   }
   // epilogue: acc rows = co (band + lq*4 + r), cols = px (nt*16 + l16):
   // 16-lane groups store 32 contiguous bytes
-  for (int r = 0; r < 4; ++r) {
-    int co = n0 + band + lq * 4 + r;
-    if (co >= Cout) continue;
-    float bv = bias ? bias[co] : 0.f;
-    for (int nt = 0; nt < 4; ++nt) {
-      int p = p0 + nt * 16 + l16;
-      if (p >= HW) continue;
-      float v = acc[nt][r] + bv;
-      if (relu) v = fmaxf(v, 0.f);
-      out[(b * Cout + co) * HW + p] = __float2bfloat16(v);
+  for (int h = 0; h < 2; ++h) {
+    for (int r = 0; r < 4; ++r) {
+      int co = n0 + h * 64 + band + lq * 4 + r;
+      if (co >= Cout) continue;
+      float bv = bias ? bias[co] : 0.f;
+      for (int nt = 0; nt < 4; ++nt) {
+        int p = p0 + nt * 16 + l16;
+        if (p >= HW) continue;
+        float v = acc[h][nt][r] + bv;
+        if (relu) v = fmaxf(v, 0.f);
+        out[(b * Cout + co) * HW + p] = __float2bfloat16(v);
+      }
     }
   }
 }
