@@ -88,6 +88,10 @@ class Conv2dHIP(nn.Conv2d):
             kh, kw = self.kernel_size
             return _Conv2dFn.apply(x.contiguous(), self.weight, self.bias,
                                    kh, kw, kh // 2, kw // 2)
+        if x.is_cuda and os.environ.get('DISTAR_AMD_CONV_DEBUG') == '1':
+            print(f'[Conv2dHIP fallback] shape={tuple(x.shape)} dtype={x.dtype} '
+                  f'k={self.kernel_size} s={self.stride} p={self.padding} '
+                  f'd={self.dilation} g={self.groups}', flush=True)
         return super().forward(x)
 
 
