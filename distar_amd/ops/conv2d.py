@@ -89,9 +89,11 @@ class Conv2dHIP(nn.Conv2d):
             return _Conv2dFn.apply(x.contiguous(), self.weight, self.bias,
                                    kh, kw, kh // 2, kw // 2)
         if x.is_cuda and os.environ.get('DISTAR_AMD_CONV_DEBUG') == '1':
+            import sys
             print(f'[Conv2dHIP fallback] shape={tuple(x.shape)} dtype={x.dtype} '
                   f'k={self.kernel_size} s={self.stride} p={self.padding} '
-                  f'd={self.dilation} g={self.groups}', flush=True)
+                  f'd={self.dilation} g={self.groups}', file=sys.stderr,
+                  flush=True)
         return super().forward(x)
 
 
