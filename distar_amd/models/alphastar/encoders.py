@@ -453,8 +453,10 @@ class ValueEncoder(nn.Module):
         b, c, h, w = x['own_units_spatial'].shape
         scatter_map = scatter_connection((b, h, w), project_embedding, entity_location,
                                          self.scatter_dim, self.scatter_type)
-        spatial_x = torch.cat([scatter_map, x['own_units_spatial'].float(),
-                               x['enemy_units_spatial'].float()], dim=1)
+        spatial_x = torch.cat([scatter_map,
+                               x['own_units_spatial'].to(scatter_map.dtype),
+                               x['enemy_units_spatial'].to(scatter_map.dtype)],
+                              dim=1)
         spatial_x = self.project(spatial_x)
         spatial_x = self.downsample(spatial_x)
         for i in range(self.resblock_num):

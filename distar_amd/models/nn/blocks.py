@@ -228,7 +228,10 @@ class GatedResBlock(nn.Module):
         residual = x
         x = self.conv1(x)
         x = self.conv2(x)
-        x = torch.tanh(x * torch.sigmoid(self.GateWeightG(noise_map))) * self.UpdateSP
+        # keep the compute dtype: the fp32 UpdateSP parameter would promote
+        # the whole chain (and every downstream conv) to fp32 under autocast
+        x = torch.tanh(x * torch.sigmoid(self.GateWeightG(noise_map))) \
+            * self.UpdateSP.to(x.dtype)
         return self.act(x + residual)
 
 
