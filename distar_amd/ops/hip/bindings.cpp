@@ -597,6 +597,9 @@ extern "C" __global__ void conv2d_fwd_kernel(
 extern "C" __global__ void conv2d_small_fwd_kernel(
     const __hip_bfloat16*, const __hip_bfloat16*, const float*,
     __hip_bfloat16*, int, int, int, int, int, int, int, int, int, int, int);
+extern "C" __global__ void conv2d_wgrad_small_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, float*,
+    int, int, int, int, int, int, int, int, int, int);
 extern "C" __global__ void conv2d_wgrad_kernel(
     const __hip_bfloat16*, const __hip_bfloat16*, float*,
     int, int, int, int, int, int, int, int, int, int, int);
@@ -654,6 +657,26 @@ torch::Tensor conv2d_wgrad(torch::Tensor input, torch::Tensor dout,
                           input.options().dtype(torch::kFloat32));
   // pick images-per-block so total blocks lands in a healthy range
   int64_t K_real = Cin * KH * KW;
+  if (Cout <= 4) {
+    // tiny-Cout direct wgrad: dW (Cout, Cin, KH, KW) fp32, one input read
+    // per block, per-thread register partials (see conv2d.hip)
+    auto dwd = torch::zeros({Cout, Cin, KH, KW},
+                            input.options().dtype(torch::kFloat32));
+    int ipb2 = (int)std::max<int64_t>(1, B / std::max<int64_t>(1, 4096 / Cin));
+    dim3 g2((unsigned)Cin, (unsigned)((B + ipb2 - 1) / ipb2));
+    auto stream2 = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(conv2d_wgrad_small_kernel, g2, dim3(256), 0,
+                       stream2.stream(), bfp(input), bfp(dout),
+                       dwd.data_ptr<float>(), (int)B, (int)Cin, (int)Cout,
+                       (int)H, (int)W, (int)KH, (int)KW, (int)padH,
+                       (int)padW, ipb2);
+    // pack into the (Kpad, Cout) layout the python wrapper slices
+    auto dwp = torch::zeros({Kpad, Cout},
+                            input.options().dtype(torch::kFloat32));
+    dwp.narrow(0, 0, K_real).copy_(
+        dwd.permute({1, 2, 3, 0}).reshape({K_real, Cout}));
+    return dwp;
+  }
   int kt = (int)((std::min<int64_t>(Kpad, (K_real + 31) / 32 * 32) + 63) / 64);
   long want_z = 32768 / std::max(1, kt);
   int ipb = (int)std::max<long>(1, (B + want_z - 1) / std::max<long>(1, want_z));

@@ -193,6 +193,7 @@ void conv2d_wgrad_kernel(
   const int K_real = Cin * KH * KW;
   if (k_base >= K_real) return;                 // pure-padding k tile
   const int n_tiles_co = (Cout + 15) / 16;      // <= 8 (Cout <= 128)
+  const int wrecip = (1048576 + W - 1) / W;
 
   extern __shared__ char lds[];                 // A^T tile [64 k][64 px] 8 KB
   const int lane = threadIdx.x & 63;
@@ -218,11 +219,18 @@ void conv2d_wgrad_kernel(
         int k = k_base + kk;
         __hip_bfloat16 vals[8] = {};
         if (k < K_real) {
-          int ci = k / (KH * KW);
-          int off = k % (KH * KW);
-          int dy = off / KW - padH, dx = off % KW - padW;
+          int ci, dy, dx;
+          if (KH * KW == 1) {
+            ci = k; dy = -padH; dx = -padW;
+          } else {                         // 3x3: mul-shift div by 9 / 3
+            ci = (k * 7282) >> 16;
+            int off = k - ci * 9;
+            dy = ((off * 21846) >> 16) - padH;
+            dx = off - ((off * 21846) >> 16) * 3 - padW;
+          }
           int pbase = p0 + g * 8;
-          int y0 = pbase / W, x0 = pbase % W;  // the only division per group
+          int y0 = (int)(((long)pbase * wrecip) >> 20);
+          int x0 = pbase - y0 * W;
           int y = y0 + dy;
           int x = x0 + dx;
           const __hip_bfloat16* src = inb + ((long)ci * H + y) * W + x;
