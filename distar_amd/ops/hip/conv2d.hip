@@ -378,3 +378,56 @@ extern "C" __global__ void conv2d_small_fwd_kernel(
     out[i] = __float2bfloat16(acc);
   }
 }
+
+
+// Direct wgrad for small Cout (<= 4): dW[co,ci,dy,dx] over all pixels.
+// The MFMA wgrad wastes >= 16x of its N-tile there, and its im2col
+// staging re-reads the input 9x; here each input element is read once
+// per block and all KH*KW*Cout partials ride in registers.
+extern "C" __global__ void conv2d_wgrad_small_kernel(
+    const __hip_bfloat16* __restrict__ input,   // (B, Cin, H, W)
+    const __hip_bfloat16* __restrict__ dout,    // (B, Cout, H, W)
+    float* __restrict__ dw,                     // (Cout, Cin, KH, KW) fp32
+    int B, int Cin, int Cout, int H, int W,
+    int KH, int KW, int padH, int padW, int ipb) {
+  const long HW = (long)H * W;
+  const int ci = blockIdx.x;
+  const int b_base = blockIdx.y * ipb;
+  const int kwin = KH * KW;
+  float acc[4 * 9] = {};                        // [co][off]
+  for (int bi = 0; bi < ipb && b_base + bi < B; ++bi) {
+    const long b = b_base + bi;
+    const __hip_bfloat16* cib = input + (b * Cin + ci) * HW;
+    const __hip_bfloat16* dob = dout + b * Cout * HW;
+    for (long p = threadIdx.x; p < HW; p += blockDim.x) {
+      int y = p / W, x = p - (p / W) * W;
+      float inv = __bfloat162float(cib[p]);
+      if (inv == 0.f) continue;
+      for (int co = 0; co < Cout; ++co) {
+        const __hip_bfloat16* dco = dob + (long)co * HW;
+        for (int off = 0; off < kwin; ++off) {
+          // dW[co][ci][dy][dx] += in[y][x] * dout[y-dy+padH][x-dx+padW]
+          int dy = off / KW - padH, dx = off - (off / KW) * KW - padW;
+          int yy = y - dy, xx = x - dx;
+          if (yy >= 0 && yy < H && xx >= 0 && xx < W)
+            acc[co * kwin + off] += inv * __bfloat162float(dco[(long)yy * W + xx]);
+        }
+      }
+    }
+  }
+  __shared__ float red[4];
+  for (int co = 0; co < Cout; ++co)
+    for (int off = 0; off < kwin; ++off) {
+      float v = acc[co * kwin + off];
+      for (int o = 1; o < 64; o <<= 1) v += __shfl_xor(v, o, 64);
+      int lane = threadIdx.x & 63, wv = threadIdx.x >> 6;
+      if (lane == 0) red[wv] = v;
+      __syncthreads();
+      if (threadIdx.x == 0) {
+        float t = 0.f;
+        for (int w = 0; w < (int)(blockDim.x >> 6); ++w) t += red[w];
+        atomicAdd(&dw[((long)co * Cin + ci) * kwin + off], t);
+      }
+      __syncthreads();
+    }
+}
