@@ -228,3 +228,21 @@ def test_scalar_logger_writes_tb(tmp_path):
     sl.close()
     tb_dir = os.path.join(tmp_path, 'tb', 't')
     assert any(f.startswith('events.out') for f in os.listdir(tb_dir))
+
+
+def test_hip_extension_loads_and_exports():
+    """The in-tree _hip_ops.so must import on CPU too (undefined kernel
+    symbols surface at load time, long before a GPU box sees them)."""
+    from distar_amd.ops import hip_ext
+    ext = hip_ext._load()
+    if ext is None:
+        import pytest
+        pytest.skip('extension not built in this checkout')
+    for sym in ('entity_attn_fwd', 'entity_attn_bwd', 'conv2d_fwd',
+                'conv2d_wgrad', 'maxpool2x2_fwd', 'maxpool2x2_bwd',
+                'scatter_add_map', 'residual_ln_fwd', 'residual_ln_bwd',
+                'multi_norm_sq', 'multi_clip', 'masked_ce_fwd',
+                'entropy_fwd', 'kl_fwd', 'lnlstm_forward', 'su_sample',
+                'upsample2x', 'vtrace_scan', 'lambda_return_scan',
+                'entity_embed', 'mfma_selftest'):
+        assert hasattr(ext, sym), sym
