@@ -637,7 +637,7 @@ torch::Tensor conv2d_fwd(torch::Tensor input, torch::Tensor wp,
     return out;
   }
   dim3 grid((HW + 63) / 64, (unsigned)((Cout + 127) / 128), (unsigned)B);
-  hipLaunchKernelGGL(conv2d_fwd_kernel, grid, dim3(256), 2 * 64 * 256,
+  hipLaunchKernelGGL(conv2d_fwd_kernel, grid, dim3(256), 64 * 256,
                      stream.stream(), bfp(input), bfp(wp), bp, bfp_mut(out),
                      (int)B, (int)Cin, (int)Cout, (int)H, (int)W,
                      (int)KH, (int)KW, (int)padH, (int)padW, (int)Kpad,
@@ -682,7 +682,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor input, torch::Tensor dout,
   int ipb = (int)std::max<long>(1, (B + want_z - 1) / std::max<long>(1, want_z));
   dim3 grid(kt, 1, (unsigned)((B + ipb - 1) / ipb));
   auto stream = c10::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(conv2d_wgrad_kernel, grid, dim3(256), 2 * 64 * 256,
+  hipLaunchKernelGGL(conv2d_wgrad_kernel, grid, dim3(256), 64 * 256,
                      stream.stream(), bfp(input), bfp(dout),
                      dwp.data_ptr<float>(),
                      (int)B, (int)Cin, (int)Cout, (int)H, (int)W,

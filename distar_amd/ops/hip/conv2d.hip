@@ -95,18 +95,16 @@ void conv2d_fwd_kernel(
   // the staging's integer divisions + column-major LDS writes made the
   // kernel 69:1 VALU:MFMA with 32-way write conflicts)
   const int kwin = KH * KW;
-  const int n_chunks = (k_hi + KC - 1) / KC;
-  char* bufA = lds;
-  char* bufB = lds + 64 * 256;
-
-  // k-major staging with kk FASTEST across threads: the 8-pixel global
-  // run stays one vector load, and a wave's 64 LDS writes span all 32
-  // banks (kk>>3 varies per lane); ci/off/y0 via reciprocal mul-shifts
-  auto stage_chunk = [&](char* buf, int k0) {
+  for (int k0 = 0; k0 < k_hi; k0 += KC) {
     const int krows = min(k_hi - k0, KC);
+    __syncthreads();
+    // k-major staging with kk FASTEST across threads: the 8-pixel global
+    // run stays one vector load, and a wave's 64 LDS writes span all 32
+    // banks (kk>>3 varies per lane; px-major order had 16-way conflicts,
+    // px-major single-writes lost the vectorized global reads — PMC r2l/m)
     const int krecip = (65536 + krows - 1) / krows;
     for (int task = tid; task < krows * 8; task += 256) {
-      int g = (task * krecip) >> 16;      // pixel group
+      int g = (task * krecip) >> 16;      // pixel group (fixed per wave)
       int kk = task - g * krows;          // k row (fastest across lanes)
       int k = k0 + kk;
       __hip_bfloat16 vals[8] = {};
@@ -142,24 +140,14 @@ void conv2d_fwd_kernel(
         }
       }
       for (int j = 0; j < 8; ++j)
-        *(__hip_bfloat16*)(buf + cswz(g * 8 + j, kk)) = vals[j];
+        *(__hip_bfloat16*)(lds + cswz(g * 8 + j, kk)) = vals[j];
     }
-  };
-
-  // 2-phase pipeline: stage chunk c+1 while the MFMAs consume chunk c —
-  // one barrier per chunk (guide T3-minimum pattern)
-  stage_chunk(bufA, 0);
-  __syncthreads();
-  for (int c = 0; c < n_chunks; ++c) {
-    const int k0 = c * KC;
-    char* buf = (c & 1) ? bufB : bufA;
-    if (c + 1 < n_chunks)
-      stage_chunk((c & 1) ? bufA : bufB, k0 + KC);
+    __syncthreads();
     const int ks_count = min(k_hi - k0, KC) / 32;
     for (int ks = 0; ks < ks_count; ++ks) {
       bf16x8c bi[4];
       for (int nt = 0; nt < 4; ++nt)
-        bi[nt] = clds8(buf, cswz(nt * 16 + l16, ks * 32 + lq * 8));
+        bi[nt] = clds8(lds, cswz(nt * 16 + l16, ks * 32 + lq * 8));
       for (int h = 0; h < 2; ++h) {
         // waves/halves whose whole co band is beyond Cout skip compute
         if (n0 + h * 64 + band >= Cout) continue;
@@ -172,7 +160,6 @@ void conv2d_fwd_kernel(
               a, bi[nt], acc[h][nt], 0, 0, 0);
       }
     }
-    __syncthreads();
   }
   // epilogue: acc rows = co (band + lq*4 + r), cols = px (nt*16 + l16):
   // 16-lane groups store 32 contiguous bytes
@@ -219,88 +206,75 @@ void conv2d_wgrad_kernel(
   f32x4c acc[8];                        // co tiles (<= 128 cout)
   for (int nt = 0; nt < 8; ++nt) acc[nt] = (f32x4c){0, 0, 0, 0};
 
-  // flattened (image, pixel-chunk) sequence, double-buffered staging:
-  // stage chunk c+1 while the MFMAs consume chunk c (one barrier/chunk)
-  const int chunks_per_img = (HW + KC - 1) / KC;
-  const int n_imgs = min(ipb, (int)(B - b_base));
-  const int n_chunks = n_imgs * chunks_per_img;
-  const int crecip = (65536 + chunks_per_img - 1) / chunks_per_img;
-  char* bufA = lds;
-  char* bufB = lds + 64 * 256;
-
-  auto stage_chunk = [&](char* buf, int c) {
-    const int bi = (c * crecip) >> 16;
-    const int p0 = (c - bi * chunks_per_img) * KC;
-    const __hip_bfloat16* inb = input + (b_base + (long)bi) * Cin * HW;
-    for (int task = tid; task < CTILE * (KC / 8); task += 256) {
-      int kk = task >> 4, g = task & 15;
-      int k = k_base + kk;
-      __hip_bfloat16 vals[8] = {};
-      if (k < K_real) {
-        int ci, dy, dx;
-        if (KH * KW == 1) {
-          ci = k; dy = -padH; dx = -padW;
-        } else {                         // 3x3: mul-shift div by 9 / 3
-          ci = (k * 7282) >> 16;
-          int off = k - ci * 9;
-          dy = ((off * 21846) >> 16) - padH;
-          dx = off - ((off * 21846) >> 16) * 3 - padW;
-        }
-        int pbase = p0 + g * 8;
-        int y0 = (int)(((long)pbase * wrecip) >> 20);
-        int x0 = pbase - y0 * W;
-        int y = y0 + dy;
-        int x = x0 + dx;
-        const __hip_bfloat16* src = inb + ((long)ci * H + y) * W + x;
-        if (pbase + 7 < HW && x0 + 7 < W && y >= 0 && y < H &&
-            x >= 0 && x + 7 < W) {
-          __builtin_memcpy(vals, src, 16);
-        } else {
-          const __hip_bfloat16* cib = inb + (long)ci * H * W;
-          for (int j = 0; j < 8; ++j) {
-            if (pbase + j < HW) {
-              int yj = y0 + dy, xj = x0 + dx;
-              if (yj >= 0 && yj < H && xj >= 0 && xj < W)
-                vals[j] = cib[(long)yj * W + xj];
-            }
-            if (++x0 == W) { x0 = 0; ++y0; }
+  for (int bi = 0; bi < ipb && b_base + bi < B; ++bi) {
+    const long b = b_base + bi;
+    const __hip_bfloat16* inb = input + b * Cin * HW;
+    const __hip_bfloat16* dob = dout + b * Cout * HW;
+    for (int p0 = 0; p0 < HW; p0 += KC) {
+      __syncthreads();
+      // stage im2col^T chunk: rows = k (ci/dy/dx once per row), cols =
+      // px in groups of 8 -> one 16B ds_write per group
+      for (int task = tid; task < CTILE * (KC / 8); task += 256) {
+        int kk = task >> 4, g = task & 15;
+        int k = k_base + kk;
+        __hip_bfloat16 vals[8] = {};
+        if (k < K_real) {
+          int ci, dy, dx;
+          if (KH * KW == 1) {
+            ci = k; dy = -padH; dx = -padW;
+          } else {                         // 3x3: mul-shift div by 9 / 3
+            ci = (k * 7282) >> 16;
+            int off = k - ci * 9;
+            dy = ((off * 21846) >> 16) - padH;
+            dx = off - ((off * 21846) >> 16) * 3 - padW;
           }
-        }
-      }
-      // cols g*8..g*8+7 share one swizzled granule: single 16B write
-      __builtin_memcpy(buf + cswz(kk, g * 8), vals, 16);
-    }
-  };
-
-  stage_chunk(bufA, 0);
-  __syncthreads();
-  for (int c = 0; c < n_chunks; ++c) {
-    char* buf = (c & 1) ? bufB : bufA;
-    if (c + 1 < n_chunks)
-      stage_chunk((c & 1) ? bufA : bufB, c + 1);
-    const int bi = (c * crecip) >> 16;
-    const int p0c = (c - bi * chunks_per_img) * KC;
-    const __hip_bfloat16* dob = dout + (b_base + (long)bi) * Cout * HW;
-    for (int ks = 0; ks < 4; ++ks) {
-      bf16x8c a = clds8(buf, cswz(band + l16, ks * 32 + lq * 8));
-      for (int nt = 0; nt < n_tiles_co; ++nt) {
-        int co = nt * 16 + l16;
-        bf16x8c bdo = (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
-        if (co < Cout) {
-          int p = p0c + ks * 32 + lq * 8;
-          if (p + 8 <= HW) {
-            __builtin_memcpy(&bdo, dob + (long)co * HW + p, 16);
+          int pbase = p0 + g * 8;
+          int y0 = (int)(((long)pbase * wrecip) >> 20);
+          int x0 = pbase - y0 * W;
+          int y = y0 + dy;
+          int x = x0 + dx;
+          const __hip_bfloat16* src = inb + ((long)ci * H + y) * W + x;
+          if (pbase + 7 < HW && x0 + 7 < W && y >= 0 && y < H &&
+              x >= 0 && x + 7 < W) {
+            __builtin_memcpy(vals, src, 16);
           } else {
-            for (int j = 0; j < 8; ++j)
-              if (p + j < HW)
-                bdo[j] = ((const __bf16*)dob)[(long)co * HW + p + j];
+            const __hip_bfloat16* cib = inb + (long)ci * H * W;
+            for (int j = 0; j < 8; ++j) {
+              if (pbase + j < HW) {
+                int yj = y0 + dy, xj = x0 + dx;
+                if (yj >= 0 && yj < H && xj >= 0 && xj < W)
+                  vals[j] = cib[(long)yj * W + xj];
+              }
+              if (++x0 == W) { x0 = 0; ++y0; }  // incremental, no div/mod
+            }
           }
         }
-        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bdo, acc[nt],
-                                                          0, 0, 0);
+        // cols g*8..g*8+7 share one swizzled granule: single 16B write
+        __builtin_memcpy(lds + cswz(kk, g * 8), vals, 16);
+      }
+      __syncthreads();
+      for (int ks = 0; ks < 4; ++ks) {
+        bf16x8c a = clds8(lds, cswz(band + l16, ks * 32 + lq * 8));
+        for (int nt = 0; nt < n_tiles_co; ++nt) {
+          int co = nt * 16 + l16;
+          bf16x8c bdo = (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
+          if (co < Cout) {
+            int p = p0 + ks * 32 + lq * 8;
+            if (p + 8 <= HW) {
+              // HW*2B is not always 16B-aligned per-row: memcpy lets the
+              // compiler emit the widest legal loads
+              __builtin_memcpy(&bdo, dob + (long)co * HW + p, 16);
+            } else {
+              for (int j = 0; j < 8; ++j)
+                if (p + j < HW)
+                  bdo[j] = ((const __bf16*)dob)[(long)co * HW + p + j];
+            }
+          }
+          acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bdo, acc[nt],
+                                                            0, 0, 0);
+        }
       }
     }
-    __syncthreads();
   }
   // accumulate into dwp (Kpad, Cout) fp32
   for (int nt = 0; nt < n_tiles_co; ++nt) {
