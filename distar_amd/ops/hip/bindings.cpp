@@ -11,6 +11,11 @@ extern "C" __global__ void vtrace_kernel(
     const float*, float*, int, int);
 
 using bf16_t = __hip_bfloat16;
+extern "C" __global__ void lnlstm_forward_coop_kernel(
+    const float*, const float*, const float*, const bf16_t*,
+    const float*, const float*, const float*, const float*,
+    float*, float*, float*, float*, bf16_t*, float*, float*, float*,
+    int, int, int);
 extern "C" __global__ void lnlstm_forward_kernel(
     const float*, const float*, const float*, const bf16_t*,
     const float*, const float*, const float*, const float*,
@@ -143,6 +148,49 @@ std::vector<torch::Tensor> lnlstm_forward(
   auto cellraw = torch::empty({T, B, H}, opt);
   size_t lds = sizeof(float) * ((H + 1) / 2 + G + H + kNT);
   auto stream = c10::hip::getCurrentHIPStream();
+  if (B <= 64 && H % 32 == 0) {
+    // small-batch learner core: spread each timestep's h @ W_hh^T across
+    // the chip as an MFMA matmul with grid-wide syncs (the per-row kernel
+    // would use only B of 256 CUs).  Same outputs; same backward.
+    int64_t Mt = (B + 15) / 16, Nt = G / 16;
+    int blocks = (int)((Mt * Nt + 3) / 4);
+    auto bopt = igates.options().dtype(torch::kBFloat16);
+    auto h_bf = torch::zeros({Mt * 16, H}, bopt);
+    h_bf.narrow(0, 0, B).copy_(h0);
+    auto o_ws = torch::empty({B, H}, igates.options());
+    auto hstats = torch::zeros({2, B, 2}, igates.options());
+    auto cstats = torch::zeros({2, B, 2}, igates.options());
+    h_all[0].copy_(h0);
+    c_all[0].copy_(c0);
+    const float* igp = igates.data_ptr<float>();
+    const float* h0p = h0.data_ptr<float>();
+    const float* c0p = c0.data_ptr<float>();
+    const __hip_bfloat16* wp2 =
+        reinterpret_cast<const __hip_bfloat16*>(w_hh_bf16.data_ptr());
+    const float* lhw = lnh_w.data_ptr<float>();
+    const float* lhb = lnh_b.data_ptr<float>();
+    const float* lcw = lnc_w.data_ptr<float>();
+    const float* lcb = lnc_b.data_ptr<float>();
+    float* hap = h_all.data_ptr<float>();
+    float* cap = c_all.data_ptr<float>();
+    float* hgp = hgates_raw.data_ptr<float>();
+    float* crp = cellraw.data_ptr<float>();
+    __hip_bfloat16* hbp =
+        reinterpret_cast<__hip_bfloat16*>(h_bf.data_ptr());
+    float* owp = o_ws.data_ptr<float>();
+    float* hsp = hstats.data_ptr<float>();
+    float* csp = cstats.data_ptr<float>();
+    int Ti = (int)T, Bi = (int)B, Hi = (int)H;
+    void* args[] = {&igp, &h0p, &c0p, &wp2, &lhw, &lhb, &lcw, &lcb,
+                    &hap, &cap, &hgp, &crp, &hbp, &owp, &hsp, &csp,
+                    &Ti, &Bi, &Hi};
+    hipError_t err = hipLaunchCooperativeKernel(
+        (const void*)lnlstm_forward_coop_kernel, dim3(blocks), dim3(256),
+        args, 0, stream.stream());
+    TORCH_CHECK(err == hipSuccess, "coop lnlstm launch: ",
+                hipGetErrorString(err));
+    return {h_all, c_all, hgates_raw, cellraw};
+  }
   hipLaunchKernelGGL(lnlstm_forward_kernel, dim3(B), dim3(kNT), lds,
                      stream.stream(),
                      igates.data_ptr<float>(), h0.data_ptr<float>(),

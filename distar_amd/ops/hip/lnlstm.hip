@@ -330,3 +330,134 @@ extern "C" __global__ void lnlstm_backward_kernel(
     atomicAdd(&dlnc_b[k], dbc_loc);
   }
 }
+
+
+// ---------------------------------------------------------------------------
+// Cooperative-launch forward for SMALL BATCH (the learner core: B=16..64).
+// The per-row kernel above runs only B workgroups — at B=32 that is 32 of
+// 256 CUs.  This variant spreads ONE timestep's h @ W_hh^T across the whole
+// chip as an MFMA matmul (M=B, N=4H, K=H), with grid-wide syncs between the
+// matmul / gate / cell phases (3 per step) and double-buffered LN stats.
+// Outputs are identical to lnlstm_forward_kernel, so the existing backward
+// consumes them unchanged.
+#include <hip/hip_cooperative_groups.h>
+
+typedef __bf16 lbf16x8 __attribute__((ext_vector_type(8)));
+typedef float lf32x4 __attribute__((ext_vector_type(4)));
+
+extern "C" __global__ void lnlstm_forward_coop_kernel(
+    const float* __restrict__ igates,     // (T, B, 4H) post-LN_i
+    const float* __restrict__ h0,
+    const float* __restrict__ c0,
+    const bf16* __restrict__ w_hh,        // (4H, H) bf16 row-major
+    const float* __restrict__ lnh_w, const float* __restrict__ lnh_b,
+    const float* __restrict__ lnc_w, const float* __restrict__ lnc_b,
+    float* __restrict__ h_all,            // (T+1, B, H)
+    float* __restrict__ c_all,
+    float* __restrict__ hgates_raw,       // (T, B, 4H)
+    float* __restrict__ cellraw,          // (T, B, H)
+    bf16* __restrict__ h_bf,              // ws (Mt*16, H) bf16, h0-filled
+    float* __restrict__ o_ws,             // ws (B, H)
+    float* __restrict__ hstats,           // ws (2, B, 2) zeroed
+    float* __restrict__ cstats,           // ws (2, B, 2) zeroed
+    int T, int B, int H) {
+  namespace cg = cooperative_groups;
+  cg::grid_group grid = cg::this_grid();
+  const int G = 4 * H;
+  const int Mt = (B + 15) / 16;
+  const int Nt = G / 16;
+  const int lane = threadIdx.x & 63;
+  const int l16 = lane & 15;
+  const int lq = lane >> 4;
+  const int slot = (int)(blockIdx.x * 4 + (threadIdx.x >> 6));   // wave slot
+  const int mt = slot % Mt;
+  const int nt = slot / Mt;               // may exceed Nt for pad waves
+  const long total_bh = (long)B * H;
+  const long tid_g = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long nthreads = (long)gridDim.x * blockDim.x;
+
+  for (int t = 0; t < T; ++t) {
+    const int sl = t & 1;
+    // ---- P1: hg_raw = h_bf @ W_hh^T (MFMA) + LN_h partials
+    if (nt < Nt) {
+      lf32x4 acc = (lf32x4){0, 0, 0, 0};
+      for (int ks = 0; ks < H / 32; ++ks) {
+        lbf16x8 a, b;
+        __builtin_memcpy(&a, h_bf + (long)(mt * 16 + l16) * H + ks * 32
+                         + lq * 8, 16);
+        __builtin_memcpy(&b, w_hh + (long)(nt * 16 + l16) * H + ks * 32
+                         + lq * 8, 16);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+      }
+      // acc: col = gate (nt*16 + l16), row = b (mt*16 + lq*4 + r)
+      for (int r = 0; r < 4; ++r) {
+        int b_i = mt * 16 + lq * 4 + r;
+        int g_i = nt * 16 + l16;
+        if (b_i < B) {
+          float v = acc[r];
+          hgates_raw[((long)t * B + b_i) * G + g_i] = v;
+          // per-(wave, b) partial over this wave's 16 gates
+          float s = v, sq = v * v;
+          for (int o = 1; o < 16; o <<= 1) {
+            s += __shfl_xor(s, o, 64);
+            sq += __shfl_xor(sq, o, 64);
+          }
+          if (l16 == 0) {
+            atomicAdd(&hstats[(sl * B + b_i) * 2 + 0], s);
+            atomicAdd(&hstats[(sl * B + b_i) * 2 + 1], sq);
+          }
+        }
+      }
+    }
+    grid.sync();
+    // ---- P2: gates + cell-raw + LN_c partials; zero next hstats slot
+    for (long i = tid_g; i < total_bh; i += nthreads) {
+      int b_i = i / H, h_i = i - (long)(i / H) * H;
+      float mean = hstats[(sl * B + b_i) * 2 + 0] / G;
+      float var = hstats[(sl * B + b_i) * 2 + 1] / G - mean * mean;
+      float rstd = rsqrtf(var + LN_EPS);
+      const float* hg = hgates_raw + ((long)t * B + b_i) * G;
+      const float* ig = igates + ((long)t * B + b_i) * G;
+      float gate[4];
+      for (int k = 0; k < 4; ++k) {
+        int g_i = k * H + h_i;
+        gate[k] = ig[g_i] + (hg[g_i] - mean) * rstd * lnh_w[g_i] + lnh_b[g_i];
+      }
+      float iv = 1.f / (1.f + __expf(-gate[0]));
+      float fv = 1.f / (1.f + __expf(-gate[1]));
+      float gv = tanhf(gate[2]);
+      float ov = 1.f / (1.f + __expf(-gate[3]));
+      float cprev = c_all[((long)t * B + b_i) * H + h_i];
+      float craw = fv * cprev + iv * gv;
+      cellraw[((long)t * B + b_i) * H + h_i] = craw;
+      o_ws[i] = ov;
+      float s = craw, sq = craw * craw;
+      atomicAdd(&cstats[(sl * B + b_i) * 2 + 0], s);
+      atomicAdd(&cstats[(sl * B + b_i) * 2 + 1], sq);
+      // zero the other hstats slot for t+1 (its last reader was P2(t-1))
+      if (h_i == 0) {
+        hstats[((1 - sl) * B + b_i) * 2 + 0] = 0.f;
+        hstats[((1 - sl) * B + b_i) * 2 + 1] = 0.f;
+      }
+    }
+    grid.sync();
+    // ---- P3: LN_c + h; zero next cstats slot
+    for (long i = tid_g; i < total_bh; i += nthreads) {
+      int b_i = i / H, h_i = i - (long)(i / H) * H;
+      float cmean = cstats[(sl * B + b_i) * 2 + 0] / H;
+      float cvar = cstats[(sl * B + b_i) * 2 + 1] / H - cmean * cmean;
+      float crstd = rsqrtf(cvar + LN_EPS);
+      float craw = cellraw[((long)t * B + b_i) * H + h_i];
+      float c_new = (craw - cmean) * crstd * lnc_w[h_i] + lnc_b[h_i];
+      float h_new = o_ws[i] * tanhf(c_new);
+      c_all[((long)(t + 1) * B + b_i) * H + h_i] = c_new;
+      h_all[((long)(t + 1) * B + b_i) * H + h_i] = h_new;
+      h_bf[(long)b_i * H + h_i] = __float2bfloat16(h_new);
+      if (h_i == 0) {
+        cstats[((1 - sl) * B + b_i) * 2 + 0] = 0.f;
+        cstats[((1 - sl) * B + b_i) * 2 + 1] = 0.f;
+      }
+    }
+    grid.sync();
+  }
+}
