@@ -148,10 +148,15 @@ std::vector<torch::Tensor> lnlstm_forward(
   auto cellraw = torch::empty({T, B, H}, opt);
   size_t lds = sizeof(float) * ((H + 1) / 2 + G + H + kNT);
   auto stream = c10::hip::getCurrentHIPStream();
-  if (B <= 64 && H % 32 == 0) {
-    // small-batch learner core: spread each timestep's h @ W_hh^T across
-    // the chip as an MFMA matmul with grid-wide syncs (the per-row kernel
-    // would use only B of 256 CUs).  Same outputs; same backward.
+  static const bool use_coop = []() {
+    const char* v = getenv("DISTAR_AMD_LSTM_COOP");
+    return v && v[0] == '1';
+  }();
+  if (use_coop && B <= 64 && H % 32 == 0) {
+    // small-batch cooperative variant: spreads each timestep's h @ W_hh^T
+    // across the chip as an MFMA matmul.  MEASURED NULL RESULT (r2w):
+    // grid.sync() costs ~50us on gfx950, and 3 syncs x 64 steps eat the
+    // occupancy win — kept behind DISTAR_AMD_LSTM_COOP=1 as evidence.
     int64_t Mt = (B + 15) / 16, Nt = G / 16;
     int blocks = (int)((Mt * Nt + 3) / 4);
     auto bopt = igates.options().dtype(torch::kBFloat16);
