@@ -152,6 +152,12 @@ void entity_attn_fwd_kernel(
   for (int r = 0; r < 4; ++r) { m_row[r] = -INFINITY; l_row[r] = 0.f; }
   for (int t = 0; t < 8; ++t) acc_o[t] = (f32x4){0, 0, 0, 0};
 
+  // the wave's Q fragments are loop-invariant: load once into registers
+  bf16x8 qfrag[4];
+  __syncthreads();
+  for (int ks = 0; ks < 4; ++ks)
+    qfrag[ks] = lds_read8(Qs, swz_off(band + l16, ks * 32 + lq * 8));
+
   for (int kt = 0; kt < n_tiles; ++kt) {
     const int kv0 = kt * TILE;
     __syncthreads();
@@ -163,7 +169,7 @@ void entity_attn_fwd_kernel(
     f32x4 acc_s[4];
     for (int nt = 0; nt < 4; ++nt) acc_s[nt] = (f32x4){0, 0, 0, 0};
     for (int ks = 0; ks < 4; ++ks) {
-      bf16x8 a = lds_read8(Qs, swz_off(band + l16, ks * 32 + lq * 8));
+      bf16x8 a = qfrag[ks];
       for (int nt = 0; nt < 4; ++nt) {
         bf16x8 bfr = lds_read8(Ks, swz_off(nt * 16 + l16, ks * 32 + lq * 8));
         acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr,
@@ -308,6 +314,8 @@ void entity_attn_bwd_kv_kernel(
 
   stage_tile128(Ks, kg, n_stride, kv0, TILE, N);
   stage_tile128(Vs, vg, n_stride, kv0, TILE, N);   // V row-major (dP' B-op)
+  // (K/V fragment hoisting here spills — 16 extra VGPRs tip this kernel
+  // past 256 with its two 8-tile accumulators; LDS re-reads are cheaper)
 
   f32x4 acc_dvT[8], acc_dk[8];
   for (int t = 0; t < 8; ++t) {
@@ -481,6 +489,13 @@ void entity_attn_bwd_q_kernel(
     lse_s[i] = (q0 + i < N) ? lse_g[q0 + i] : 0.f;
     drow_s[i] = (q0 + i < N) ? drow_g[q0 + i] : 0.f;
   }
+  // loop-invariant A-operand fragments (this wave's q band) -> registers
+  __syncthreads();
+  bf16x8 qfrag[4], dofrag[4];
+  for (int ks = 0; ks < 4; ++ks) {
+    qfrag[ks] = lds_read8(Qs, swz_off(band + l16, ks * 32 + lq * 8));
+    dofrag[ks] = lds_read8(dOs, swz_off(band + l16, ks * 32 + lq * 8));
+  }
 
   f32x4 acc_dq[8];
   for (int t = 0; t < 8; ++t) acc_dq[t] = (f32x4){0, 0, 0, 0};
@@ -496,7 +511,7 @@ void entity_attn_bwd_q_kernel(
     f32x4 acc_s[4];
     for (int nt = 0; nt < 4; ++nt) acc_s[nt] = (f32x4){0, 0, 0, 0};
     for (int ks = 0; ks < 4; ++ks) {
-      bf16x8 a = lds_read8(Qs, swz_off(band + l16, ks * 32 + lq * 8));
+      bf16x8 a = qfrag[ks];
       for (int nt = 0; nt < 4; ++nt) {
         bf16x8 bk = lds_read8(Ks, swz_off(nt * 16 + l16, ks * 32 + lq * 8));
         acc_s[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bk,
@@ -507,7 +522,7 @@ void entity_attn_bwd_q_kernel(
     f32x4 acc_dp[4];
     for (int nt = 0; nt < 4; ++nt) acc_dp[nt] = (f32x4){0, 0, 0, 0};
     for (int ks = 0; ks < 4; ++ks) {
-      bf16x8 a = lds_read8(dOs, swz_off(band + l16, ks * 32 + lq * 8));
+      bf16x8 a = dofrag[ks];
       for (int nt = 0; nt < 4; ++nt) {
         bf16x8 bv = lds_read8(Vs, swz_off(nt * 16 + l16, ks * 32 + lq * 8));
         acc_dp[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bv,
