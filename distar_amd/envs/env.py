@@ -190,6 +190,15 @@ class SC2Env:
                     done = True
                     rewards[i] = outcome
         if done:
+            # game over: EVERY agent gets its final observation + outcome,
+            # due or not (the skip schedule no longer applies)
+            for i, ctrl in enumerate(self._controllers):
+                if i not in obs:
+                    o = self._wrap_obs(i, ctrl.observe())
+                    obs[i] = o
+                    outcome = ctrl.outcome(o)
+                    if outcome is not None:
+                        rewards[i] = outcome
             for i in obs:
                 rewards.setdefault(i, -sum(rewards.values()))
             if self._save_replay_episodes and \

@@ -259,6 +259,7 @@ class RemoteController:
         if resp.join_game.HasField('error') and resp.join_game.error:
             raise ProtocolError(f'join_game failed: {resp.join_game.error} '
                                 f'{resp.join_game.error_details}')
+        self._player_id = resp.join_game.player_id
         return resp.join_game.player_id
 
     def restart_game(self):
@@ -346,14 +347,17 @@ class RemoteController:
         req.available_maps.SetInParent()
         return self._request(req).available_maps
 
-    @staticmethod
-    def outcome(obs):
-        """player_result -> +1/-1/0, None while the game is running."""
+    def outcome(self, obs):
+        """player_result -> +1/-1/0 for THIS controller's player (the list
+        carries every player's result), None while the game is running."""
         results = getattr(obs['raw_obs'], 'player_result', None)
         if not results:
             return None
+        pid = getattr(self, '_player_id', None)
+        raw = next((r.result for r in results if r.player_id == pid),
+                   results[0].result)
         mapping = {1: 1, 2: -1, 3: 0}     # Victory / Defeat / Tie
-        return mapping.get(results[0].result, 0)
+        return mapping.get(raw, 0)
 
     # -------------------------------------------------------------- replay
     def save_replay(self, replay_dir):
