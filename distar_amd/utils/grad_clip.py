@@ -24,6 +24,8 @@ class GradClip:
         self.step = 0
         self._ema = None
         self._state = {}
+        self._chunks = None          # K14 chunk-view cache (same grads)
+        self._chunks_of = None
 
     def apply(self, parameters):
         params = [p for p in parameters if p.grad is not None]
@@ -44,11 +46,25 @@ class GradClip:
                     and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'):
                 # K14 hand-written multi-tensor path: one norm kernel + one
                 # clip kernel with a DEVICE-resident norm (the scale never
-                # round-trips to the host inside the step)
+                # round-trips to the host inside the step).  Big tensors are
+                # split into 64k-element chunks — the kernel runs one block
+                # per chunk, and whole-tensor chunks left only ~300 blocks
+                # on a 256-CU chip (measured 5 ms for a 0.2 ms reduction).
                 from ..ops import hip_ext
                 ops = hip_ext.require()
-                norm_sq = ops.multi_norm_sq(grads)
-                ops.multi_clip(grads, norm_sq, float(self.threshold), 1e-6)
+                chunks = self._chunks
+                if chunks is None or self._chunks_of is not grads[0]:
+                    chunks = []
+                    for g in grads:
+                        flat = g.view(-1)
+                        n = flat.numel()
+                        for off in range(0, n, 65536):
+                            chunks.append(flat.narrow(0, off,
+                                                      min(65536, n - off)))
+                    self._chunks = chunks
+                    self._chunks_of = grads[0]
+                norm_sq = ops.multi_norm_sq(chunks)
+                ops.multi_clip(chunks, norm_sq, float(self.threshold), 1e-6)
                 return float(norm_sq.sqrt())
             total = torch.nn.utils.clip_grad_norm_(params, self.threshold,
                                                    norm_type=self.norm_type,
