@@ -63,12 +63,16 @@ class _Conv2dFn(torch.autograd.Function):
             wpf = _pack_weight_flipped(weight, KH, KW)
             dx = ops.conv2d_fwd(dy, wpf, None, Cin, KH, KW,
                                 KH - 1 - padH, KW - 1 - padW, False)
+        want_bias = ctx.has_bias and ctx.needs_input_grad[2]
         if ctx.needs_input_grad[1]:
             K = Cin * KH * KW
-            dwp = ops.conv2d_wgrad(x, dy, KH, KW, padH, padW, _kpad(K))
+            dwp, dbias = ops.conv2d_wgrad(x, dy, KH, KW, padH, padW,
+                                          _kpad(K), want_bias)
             dw = dwp[:K].view(Cin, KH, KW, Cout).permute(3, 0, 1, 2) \
                 .contiguous().to(weight.dtype)
-        if ctx.has_bias and ctx.needs_input_grad[2]:
+            if want_bias and dbias is not None and dbias.numel():
+                db = dbias.to(weight.dtype)    # fused into the wgrad kernel
+        if want_bias and db is None:
             db = dy.sum(dim=(0, 2, 3)).to(weight.dtype)
         return dx, dw, db, None, None, None, None
 

@@ -654,7 +654,7 @@ extern "C" __global__ void conv2d_wgrad_small_kernel(
     const __hip_bfloat16*, const __hip_bfloat16*, float*,
     int, int, int, int, int, int, int, int, int, int);
 extern "C" __global__ void conv2d_wgrad_kernel(
-    const __hip_bfloat16*, const __hip_bfloat16*, float*,
+    const __hip_bfloat16*, const __hip_bfloat16*, float*, float*,
     int, int, int, int, int, int, int, int, int, int, int);
 extern "C" __global__ void maxpool2x2_fwd_kernel(
     const __hip_bfloat16*, __hip_bfloat16*, unsigned char*, long, int, int);
@@ -698,9 +698,10 @@ torch::Tensor conv2d_fwd(torch::Tensor input, torch::Tensor wp,
   return out;
 }
 
-torch::Tensor conv2d_wgrad(torch::Tensor input, torch::Tensor dout,
-                           int64_t KH, int64_t KW, int64_t padH, int64_t padW,
-                           int64_t Kpad) {
+std::vector<torch::Tensor> conv2d_wgrad(
+    torch::Tensor input, torch::Tensor dout,
+    int64_t KH, int64_t KW, int64_t padH, int64_t padW,
+    int64_t Kpad, bool want_bias) {
   TORCH_CHECK(input.is_cuda() && input.is_contiguous());
   TORCH_CHECK(dout.is_contiguous() && dout.scalar_type() == torch::kBFloat16);
   int64_t B = input.size(0), Cin = input.size(1);
@@ -728,19 +729,28 @@ torch::Tensor conv2d_wgrad(torch::Tensor input, torch::Tensor dout,
                             input.options().dtype(torch::kFloat32));
     dwp.narrow(0, 0, K_real).copy_(
         dwd.permute({1, 2, 3, 0}).reshape({K_real, Cout}));
-    return dwp;
+    auto dbs = want_bias ? dout.sum(torch::IntArrayRef{0, 2, 3},
+                                    false, torch::kFloat32)
+                         : torch::Tensor();
+    return {dwp, dbs};
   }
   int kt = (int)((std::min<int64_t>(Kpad, (K_real + 31) / 32 * 32) + 63) / 64);
   long want_z = 32768 / std::max(1, kt);
   int ipb = (int)std::max<long>(1, (B + want_z - 1) / std::max<long>(1, want_z));
   dim3 grid(kt, 1, (unsigned)((B + ipb - 1) / ipb));
+  torch::Tensor dbias;
+  float* dbp = nullptr;
+  if (want_bias) {
+    dbias = torch::zeros({Cout}, input.options().dtype(torch::kFloat32));
+    dbp = dbias.data_ptr<float>();
+  }
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(conv2d_wgrad_kernel, grid, dim3(256), 64 * 256,
                      stream.stream(), bfp(input), bfp(dout),
-                     dwp.data_ptr<float>(),
+                     dwp.data_ptr<float>(), dbp,
                      (int)B, (int)Cin, (int)Cout, (int)H, (int)W,
                      (int)KH, (int)KW, (int)padH, (int)padW, (int)Kpad, ipb);
-  return dwp;
+  return {dwp, dbias};
 }
 
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor input) {

@@ -185,6 +185,7 @@ void conv2d_wgrad_kernel(
     const __hip_bfloat16* __restrict__ input,   // (B, Cin, H, W)
     const __hip_bfloat16* __restrict__ dout,    // (B, Cout, H, W)
     float* __restrict__ dwp,                    // (Kpad, Cout) fp32
+    float* __restrict__ dbias,                  // (Cout) fp32 or nullptr
     int B, int Cin, int Cout, int H, int W,
     int KH, int KW, int padH, int padW, int Kpad, int ipb) {
   const int HW = H * W;
@@ -205,6 +206,11 @@ void conv2d_wgrad_kernel(
 
   f32x4c acc[8];                        // co tiles (<= 128 cout)
   for (int nt = 0; nt < 8; ++nt) acc[nt] = (f32x4c){0, 0, 0, 0};
+  // bias grad rides along in the k-tile-0 blocks (wave 0's B fragments
+  // cover each dout element of the image range exactly once) — replaces
+  // a separate full-tensor reduce per conv
+  const bool do_db = dbias && blockIdx.x == 0 && wave == 0;
+  float acc_db[8] = {};
 
   for (int bi = 0; bi < ipb && b_base + bi < B; ++bi) {
     const long b = b_base + bi;
@@ -272,8 +278,23 @@ void conv2d_wgrad_kernel(
           }
           acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bdo, acc[nt],
                                                             0, 0, 0);
+          if (do_db) {
+            float sdb = 0.f;
+            for (int j = 0; j < 8; ++j) sdb += __bfloat162float(bdo[j]);
+            acc_db[nt] += sdb;
+          }
         }
       }
+    }
+  }
+  if (do_db) {
+    for (int nt = 0; nt < n_tiles_co; ++nt) {
+      float v = acc_db[nt];
+      v += __shfl_xor(v, 16, 64);     // combine the 4 lq groups per co
+      v += __shfl_xor(v, 32, 64);
+      int co = nt * 16 + l16;
+      if (lq == 0 && co < Cout)
+        atomicAdd(&dbias[co], v);
     }
   }
   // accumulate into dwp (Kpad, Cout) fp32
