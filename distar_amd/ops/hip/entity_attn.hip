@@ -33,7 +33,6 @@
 
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
-typedef unsigned int uint32_t_;
 
 #define D_DIM 128            // head dim (fixed: entity transformer)
 #define TILE 64              // q-rows / kv-rows per block
