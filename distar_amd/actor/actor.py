@@ -130,6 +130,15 @@ class Actor:
                 self._ckpt_helper.load(spath, agent.successive_model,
                                        strict=False,
                                        logger_prints=self._logger.info)
+            # direct (non-batch-server) inference with use_cuda: the agent
+            # moves its inputs to cuda, so the models must live there too
+            # (batch-inference mode moves them in _start_batch_inference)
+            if self._cfg.use_cuda and torch.cuda.is_available() and \
+                    not self._cfg.get('gpu_batch_inference', False):
+                for m in ('model', 'teacher_model', 'successive_model'):
+                    mod = getattr(agent, m, None)
+                    if mod is not None:
+                        mod.to('cuda')
             self._agents.append(agent)
         return job
 

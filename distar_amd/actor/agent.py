@@ -445,6 +445,8 @@ class Agent:
             teacher_input = default_collate_with_dim([teacher_obs])
             teacher_input['hidden_state'] = [(h.unsqueeze(0), c.unsqueeze(0))
                                              for h, c in self._teacher_hidden_state]
+            if self._use_cuda:
+                teacher_input = to_device(teacher_input, 'cuda')
             teacher = self.teacher_model or self.model
             with torch.no_grad():
                 teacher_output = teacher.compute_teacher_logit(**teacher_input)
@@ -456,6 +458,8 @@ class Agent:
             succ_input = default_collate_with_dim([teacher_obs])
             succ_input['hidden_state'] = [(h.unsqueeze(0), c.unsqueeze(0))
                                           for h, c in self._successive_hidden_state]
+            if self._use_cuda:
+                succ_input = to_device(succ_input, 'cuda')
             with torch.no_grad():
                 successive_output = self.successive_model.compute_teacher_logit(
                     **succ_input)
