@@ -653,6 +653,12 @@ extern "C" __global__ void conv2d_small_fwd_kernel(
 extern "C" __global__ void conv2d_wgrad_small_kernel(
     const __hip_bfloat16*, const __hip_bfloat16*, float*,
     int, int, int, int, int, int, int, int, int, int);
+extern "C" __global__ void conv2d_stencil_c1_fwd_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+    __hip_bfloat16*, int, int, int, int, int, int);
+extern "C" __global__ void conv2d_stencil_c1_wgrad_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, float*, float*,
+    int, int, int, int);
 extern "C" __global__ void conv2d_wgrad_kernel(
     const __hip_bfloat16*, const __hip_bfloat16*, float*, float*,
     int, int, int, int, int, int, int, int, int, int, int);
@@ -677,6 +683,17 @@ torch::Tensor conv2d_fwd(torch::Tensor input, torch::Tensor wp,
   if (bias.has_value()) bp = bias->data_ptr<float>();
   int HW = (int)(H * W);
   auto stream = c10::hip::getCurrentHIPStream();
+  if (Cout == 1 && KH == 3 && KW == 3 && padH == 1 && padW == 1 &&
+      Cin % 16 == 0 && Cin <= 256 && B < 65536) {
+    // LDS-staged stencil: input tile read once, 3x3 window out of LDS
+    dim3 sg((unsigned)((W + 63) / 64), (unsigned)((H + 15) / 16),
+            (unsigned)B);
+    hipLaunchKernelGGL(conv2d_stencil_c1_fwd_kernel, sg, dim3(256), 0,
+                       stream.stream(), bfp(input), bfp(wp), bp,
+                       bfp_mut(out), (int)B, (int)Cin, (int)H, (int)W,
+                       (int)Kpad, relu ? 1 : 0);
+    return out;
+  }
   if (Cout < 16) {
     // memory-bound tiny-channel conv (e.g. the 32->1 location head):
     // direct VALU kernel, one thread per output pixel
@@ -712,26 +729,43 @@ std::vector<torch::Tensor> conv2d_wgrad(
   // pick images-per-block so total blocks lands in a healthy range
   int64_t K_real = Cin * KH * KW;
   if (Cout <= 4) {
-    // tiny-Cout direct wgrad: dW (Cout, Cin, KH, KW) fp32, one input read
-    // per block, per-thread register partials (see conv2d.hip)
     auto dwd = torch::zeros({Cout, Cin, KH, KW},
                             input.options().dtype(torch::kFloat32));
-    int ipb2 = (int)std::max<int64_t>(1, B / std::max<int64_t>(1, 4096 / Cin));
-    dim3 g2((unsigned)Cin, (unsigned)((B + ipb2 - 1) / ipb2));
     auto stream2 = c10::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL(conv2d_wgrad_small_kernel, g2, dim3(256), 0,
-                       stream2.stream(), bfp(input), bfp(dout),
-                       dwd.data_ptr<float>(), (int)B, (int)Cin, (int)Cout,
-                       (int)H, (int)W, (int)KH, (int)KW, (int)padH,
-                       (int)padW, ipb2);
+    torch::Tensor dbs;
+    if (Cout == 1 && KH == 3 && KW == 3 && padH == 1 && padW == 1 &&
+        Cin <= 32) {
+      // LDS-staged stencil wgrad with fused dbias (LocationHead 32->1)
+      float* dbp2 = nullptr;
+      if (want_bias) {
+        dbs = torch::zeros({1}, input.options().dtype(torch::kFloat32));
+        dbp2 = dbs.data_ptr<float>();
+      }
+      long tiles = B * ((W + 63) / 64) * ((H + 7) / 8);
+      int g = (int)std::min<long>(tiles, 4096);
+      hipLaunchKernelGGL(conv2d_stencil_c1_wgrad_kernel, dim3(g), dim3(256),
+                         0, stream2.stream(), bfp(input), bfp(dout),
+                         dwd.data_ptr<float>(), dbp2,
+                         (int)B, (int)Cin, (int)H, (int)W);
+    } else {
+      // tiny-Cout direct wgrad: dW (Cout, Cin, KH, KW) fp32, one input
+      // read per block, per-thread register partials (see conv2d.hip)
+      int ipb2 = (int)std::max<int64_t>(
+          1, B / std::max<int64_t>(1, 4096 / Cin));
+      dim3 g2((unsigned)Cin, (unsigned)((B + ipb2 - 1) / ipb2));
+      hipLaunchKernelGGL(conv2d_wgrad_small_kernel, g2, dim3(256), 0,
+                         stream2.stream(), bfp(input), bfp(dout),
+                         dwd.data_ptr<float>(), (int)B, (int)Cin, (int)Cout,
+                         (int)H, (int)W, (int)KH, (int)KW, (int)padH,
+                         (int)padW, ipb2);
+      if (want_bias)
+        dbs = dout.sum(torch::IntArrayRef{0, 2, 3}, false, torch::kFloat32);
+    }
     // pack into the (Kpad, Cout) layout the python wrapper slices
     auto dwp = torch::zeros({Kpad, Cout},
                             input.options().dtype(torch::kFloat32));
     dwp.narrow(0, 0, K_real).copy_(
         dwd.permute({1, 2, 3, 0}).reshape({K_real, Cout}));
-    auto dbs = want_bias ? dout.sum(torch::IntArrayRef{0, 2, 3},
-                                    false, torch::kFloat32)
-                         : torch::Tensor();
     return {dwp, dbs};
   }
   int kt = (int)((std::min<int64_t>(Kpad, (K_real + 31) / 32 * 32) + 63) / 64);
