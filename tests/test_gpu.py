@@ -529,7 +529,9 @@ def test_fused_residual_ln_matches_eager():
     (32, 64, (76, 80), 3),       # downsample convs
     (128, 128, (19, 20), 3),     # ResBlocks
     (132, 128, (19, 20), 1),     # location head 1x1
-    (32, 1, (76, 80), 3),        # tiny-Cout direct path (location head out)
+    (32, 1, (76, 80), 3),        # stencil path, ragged 16-col tail
+    (32, 1, (152, 160), 3),      # stencil path at the real location shape
+    (24, 2, (76, 80), 3),        # generic small-Cout fallback kernels
 ])
 def test_conv2d_hip_matches_eager(cin, cout, hw, kh):
     """K4 MFMA implicit-GEMM conv vs fp32 F.conv2d (fwd + both bwds)."""
