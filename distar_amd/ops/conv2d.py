@@ -77,6 +77,63 @@ class _Conv2dFn(torch.autograd.Function):
         return dx, dw, db, None, None, None, None
 
 
+class _ConvSmallHWFn(torch.autograd.Function):
+    """Small-spatial convs (the 19x20 ResBlock/GatedResBlock stack):
+    materialized im2col + batched hipBLASLt GEMMs.  At HW ~ 380 the
+    in-kernel gather of the implicit-GEMM kernels runs ~8x off the HBM
+    floor (profiles/r02_notes.md r2dd attribution: 86 ms/step across the
+    17 such convs); here data movement is two dedicated HIP kernels
+    (im2col / gather-col2im, no atomics) and every contraction is a plain
+    library GEMM.  The col tensor is saved for wgrad (~1.8 GB per 3x3
+    conv at the SL batch - sized for MI355X's 288 GB)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, KH, KW, padH, padW):
+        ops = hip_ext.require()
+        B, Cin, H, W = x.shape
+        Cout = weight.shape[0]
+        w2 = weight.detach().to(torch.bfloat16).reshape(Cout, -1)
+        col = x.reshape(B, Cin, H * W) if KH == 1 else ops.im2col3x3(x)
+        out = torch.matmul(w2, col)              # (Co,K) @ (B,K,P) batched
+        if bias is not None:
+            out += bias.detach().to(out.dtype).view(1, -1, 1)
+        ctx.save_for_backward(x, weight, col)
+        ctx.dims = (KH, KW, padH, padW)
+        ctx.has_bias = bias is not None
+        return out.view(B, Cout, H, W)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ops = hip_ext.require()
+        x, weight, col = ctx.saved_tensors
+        B, Cin, H, W = x.shape
+        Cout = weight.shape[0]
+        KH = ctx.dims[0]
+        dyf = dy.contiguous().to(torch.bfloat16).reshape(B, Cout, H * W)
+        w2 = weight.detach().to(torch.bfloat16).reshape(Cout, -1)
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dcol = torch.matmul(w2.t(), dyf)     # (B, K, P)
+            dx = dcol.view(B, Cin, H, W) if KH == 1 \
+                else ops.col2im3x3(dcol.contiguous(), H, W)
+        if ctx.needs_input_grad[1]:
+            # per-image partials in bf16, image sum in fp32: the 2048-image
+            # sum averages the per-partial rounding down ~sqrt(B)
+            dwb = torch.bmm(dyf, col.transpose(1, 2))
+            dw = dwb.sum(0, dtype=torch.float32).view_as(weight) \
+                .to(weight.dtype)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = dyf.sum(dim=(0, 2)).to(weight.dtype)
+        return dx, dw, db, None, None, None, None
+
+
+def _small_hw_limit():
+    try:
+        return int(os.environ.get('DISTAR_AMD_CONV_SMALLHW', '512'))
+    except ValueError:
+        return 512
+
+
 class Conv2dHIP(nn.Conv2d):
     def _use_hip(self, x):
         kh, kw = self.kernel_size
@@ -90,6 +147,10 @@ class Conv2dHIP(nn.Conv2d):
     def forward(self, x):
         if self._use_hip(x):
             kh, kw = self.kernel_size
+            if x.shape[2] * x.shape[3] <= _small_hw_limit():
+                return _ConvSmallHWFn.apply(x.contiguous(), self.weight,
+                                            self.bias, kh, kw, kh // 2,
+                                            kw // 2)
             return _Conv2dFn.apply(x.contiguous(), self.weight, self.bias,
                                    kh, kw, kh // 2, kw // 2)
         if x.is_cuda and os.environ.get('DISTAR_AMD_CONV_DEBUG') == '1':

@@ -653,6 +653,10 @@ extern "C" __global__ void conv2d_small_fwd_kernel(
 extern "C" __global__ void conv2d_wgrad_small_kernel(
     const __hip_bfloat16*, const __hip_bfloat16*, float*,
     int, int, int, int, int, int, int, int, int, int);
+extern "C" __global__ void im2col_3x3_kernel(
+    const __hip_bfloat16*, __hip_bfloat16*, long, int, int);
+extern "C" __global__ void col2im_3x3_kernel(
+    const __hip_bfloat16*, __hip_bfloat16*, long, int, int);
 extern "C" __global__ void conv2d_stencil_c1_fwd_kernel(
     const __hip_bfloat16*, const __hip_bfloat16*, const float*,
     __hip_bfloat16*, int, int, int, int, int, int);
@@ -787,6 +791,31 @@ std::vector<torch::Tensor> conv2d_wgrad(
   return {dwp, dbias};
 }
 
+torch::Tensor im2col3x3(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  int64_t B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  auto col = torch::empty({B, C * 9, H * W}, x.options());
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(im2col_3x3_kernel, dim3((unsigned)(B * C), 9),
+                     dim3(256), 0, stream.stream(), bfp(x), bfp_mut(col),
+                     (long)(B * C), (int)H, (int)W);
+  return col;
+}
+
+torch::Tensor col2im3x3(torch::Tensor dcol, int64_t H, int64_t W) {
+  TORCH_CHECK(dcol.is_cuda() && dcol.is_contiguous() && dcol.dim() == 3);
+  TORCH_CHECK(dcol.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(dcol.size(1) % 9 == 0 && dcol.size(2) == H * W);
+  int64_t B = dcol.size(0), C = dcol.size(1) / 9;
+  auto dx = torch::empty({B, C, H, W}, dcol.options());
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(col2im_3x3_kernel, dim3((unsigned)(B * C)),
+                     dim3(256), 0, stream.stream(), bfp(dcol), bfp_mut(dx),
+                     (long)(B * C), (int)H, (int)W);
+  return dx;
+}
+
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor input) {
   TORCH_CHECK(input.is_cuda() && input.is_contiguous() && input.dim() == 4);
   TORCH_CHECK(input.scalar_type() == torch::kBFloat16);
@@ -900,6 +929,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &conv2d_fwd,
         "K4 NCHW bf16 MFMA implicit-GEMM conv (stride 1), fused bias/relu");
   m.def("conv2d_wgrad", &conv2d_wgrad, "K4 conv weight gradient");
+  m.def("im2col3x3", &im2col3x3, "3x3 pad-1 im2col (B,C,HW)->(B,C*9,HW)");
+  m.def("col2im3x3", &col2im3x3, "3x3 pad-1 col2im gather (no atomics)");
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd, "2x2 maxpool fwd + argmax");
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd, "2x2 maxpool gather backward");
   m.def("entropy_fwd", &entropy_fwd, "fused rowwise entropy forward");

@@ -621,3 +621,60 @@ void conv2d_stencil_c1_wgrad_kernel(
     if (ci == 0 && dbias) atomicAdd(dbias, accb);
   }
 }
+
+
+// ---------------------------------------------------------------------------
+// Small-HW conv path (the 19x20 ResBlock/GatedResBlock stack, HW = 380):
+// at 20-wide rows the in-kernel im2col gather of the MFMA kernels
+// degenerates (runs <= 20, halo logic per element), measured ~8x off the
+// HBM floor.  For these shapes conv is materialized im2col -> batched
+// hipBLASLt GEMM (a plain library GEMM per the design rules) with these
+// two hand-written movement kernels; col2im is the gather formulation
+// (each dx element sums its 9 tap sources - no atomics).
+
+// col[b][ci*9 + (dy*3+dx)][p] = x[b][ci][y+dy-1][x+dx-1]  (3x3, pad 1)
+extern "C" __global__ __launch_bounds__(256)
+void im2col_3x3_kernel(
+    const __hip_bfloat16* __restrict__ x,   // (B, C, H, W)
+    __hip_bfloat16* __restrict__ col,       // (B, C*9, H*W)
+    long BC, int H, int W) {                // BC = B*C planes
+  const int HW = H * W;
+  // one block per (plane, tap): writes one contiguous (H*W) run of col
+  const long plane = blockIdx.x;
+  const int tap = blockIdx.y;               // 0..8
+  if (plane >= BC) return;
+  const int dy = tap / 3 - 1, dx = tap % 3 - 1;
+  const __hip_bfloat16* xp = x + plane * (long)HW;
+  __hip_bfloat16* cp = col + (plane * 9 + tap) * (long)HW;
+  const __hip_bfloat16 z = __float2bfloat16(0.f);
+  for (int p = threadIdx.x; p < HW; p += 256) {
+    int y = p / W, xx = p - y * W;
+    int sy = y + dy, sx = xx + dx;
+    cp[p] = (sy >= 0 && sy < H && sx >= 0 && sx < W) ? xp[sy * W + sx] : z;
+  }
+}
+
+// dx[b][ci][y][x] = sum_tap dcol[b][ci*9+tap][(y-dy)(x-dx)]
+extern "C" __global__ __launch_bounds__(256)
+void col2im_3x3_kernel(
+    const __hip_bfloat16* __restrict__ dcol,  // (B, C*9, H*W)
+    __hip_bfloat16* __restrict__ dx,          // (B, C, H, W)
+    long BC, int H, int W) {
+  const int HW = H * W;
+  const long plane = blockIdx.x;
+  if (plane >= BC) return;
+  const __hip_bfloat16* cp = dcol + plane * 9 * (long)HW;
+  __hip_bfloat16* xp = dx + plane * (long)HW;
+  for (int p = threadIdx.x; p < HW; p += 256) {
+    int y = p / W, xx = p - y * W;
+    float acc = 0.f;
+    #pragma unroll
+    for (int tap = 0; tap < 9; ++tap) {
+      int dy = tap / 3 - 1, dx_ = tap % 3 - 1;
+      int sy = y - dy, sx = xx - dx_;
+      if (sy >= 0 && sy < H && sx >= 0 && sx < W)
+        acc += __bfloat162float(cp[tap * (long)HW + sy * W + sx]);
+    }
+    xp[p] = __float2bfloat16(acc);
+  }
+}
