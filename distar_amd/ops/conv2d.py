@@ -94,13 +94,23 @@ class _ConvSmallHWFn(torch.autograd.Function):
         Cout = weight.shape[0]
         w2 = weight.detach().to(torch.bfloat16).reshape(Cout, -1)
         col = x.reshape(B, Cin, H * W) if KH == 1 else ops.im2col3x3(x)
-        out = torch.matmul(w2, col)              # (Co,K) @ (B,K,P) batched
+        # matmul into a preallocated base tensor: returning a view of the
+        # matmul result breaks the in-place relu that follows in
+        # conv2d_block (autograd forbids in-place on custom-Function views)
+        out = torch.empty((B, Cout, H, W), device=x.device,
+                          dtype=torch.bfloat16)
+        torch.matmul(w2, col, out=out.view(B, Cout, H * W))
         if bias is not None:
-            out += bias.detach().to(out.dtype).view(1, -1, 1)
-        ctx.save_for_backward(x, weight, col)
+            out += bias.detach().to(out.dtype).view(1, -1, 1, 1)
+        # keep col for wgrad when it is small (~1.8 GB per 19x20 conv);
+        # recompute it in backward for larger-HW shapes
+        keep = KH == 1 or col.numel() * 2 <= _col_save_bytes()
+        ctx.save_for_backward(x, weight,
+                              col if keep else torch.empty(0, device=x.device))
+        ctx.kept_col = keep
         ctx.dims = (KH, KW, padH, padW)
         ctx.has_bias = bias is not None
-        return out.view(B, Cout, H, W)
+        return out
 
     @staticmethod
     def backward(ctx, dy):
@@ -117,6 +127,8 @@ class _ConvSmallHWFn(torch.autograd.Function):
             dx = dcol.view(B, Cin, H, W) if KH == 1 \
                 else ops.col2im3x3(dcol.contiguous(), H, W)
         if ctx.needs_input_grad[1]:
+            if not ctx.kept_col:
+                col = ops.im2col3x3(x)
             # per-image partials in bf16, image sum in fp32: the 2048-image
             # sum averages the per-partial rounding down ~sqrt(B)
             dwb = torch.bmm(dyf, col.transpose(1, 2))
@@ -132,6 +144,13 @@ def _small_hw_limit():
         return int(os.environ.get('DISTAR_AMD_CONV_SMALLHW', '512'))
     except ValueError:
         return 512
+
+
+def _col_save_bytes():
+    try:
+        return int(os.environ.get('DISTAR_AMD_CONV_COL_SAVE_MB', '2048')) << 20
+    except ValueError:
+        return 2048 << 20
 
 
 class Conv2dHIP(nn.Conv2d):

@@ -79,6 +79,12 @@ extern "C" __global__ void upsample2x_fwd_f32(const float*, float*, int, int, in
 extern "C" __global__ void upsample2x_fwd_bf16(const __hip_bfloat16*, __hip_bfloat16*, int, int, int);
 extern "C" __global__ void upsample2x_bwd_f32(const float*, float*, int, int, int);
 extern "C" __global__ void upsample2x_bwd_bf16(const __hip_bfloat16*, __hip_bfloat16*, int, int, int);
+extern "C" __global__ void upsample2x_bwd_bf16_fast(const __hip_bfloat16*, __hip_bfloat16*, int, int, int);
+extern "C" __global__ void maxpool2x2_fwd_vec_kernel(
+    const __hip_bfloat16*, __hip_bfloat16*, unsigned char*, long, int, int);
+extern "C" __global__ void maxpool2x2_bwd_vec_kernel(
+    const __hip_bfloat16*, const unsigned char*, __hip_bfloat16*,
+    long, int, int);
 
 torch::Tensor upsample2x(torch::Tensor input) {
   TORCH_CHECK(input.is_cuda() && input.is_contiguous() && input.dim() == 4);
@@ -118,8 +124,11 @@ torch::Tensor upsample2x_backward(torch::Tensor gout) {
                        stream.stream(), gout.data_ptr<float>(),
                        gin.data_ptr<float>(), (int)(N * C), (int)H, (int)W);
   } else if (gout.scalar_type() == torch::kBFloat16) {
-    hipLaunchKernelGGL(upsample2x_bwd_bf16, dim3(blocks), dim3(threads), 0,
-                       stream.stream(),
+    // fast path: constant 4x4 tap weights, generic per-px at borders
+    long groups = N * C * H * (W / 4);
+    int gblocks = (int)std::min<long>((groups + threads - 1) / threads, 8192);
+    hipLaunchKernelGGL(upsample2x_bwd_bf16_fast, dim3(gblocks), dim3(threads),
+                       0, stream.stream(),
                        reinterpret_cast<const __hip_bfloat16*>(gout.data_ptr()),
                        reinterpret_cast<__hip_bfloat16*>(gin.data_ptr()),
                        (int)(N * C), (int)H, (int)W);
@@ -827,10 +836,19 @@ std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor input) {
   long total = out.numel();
   int blocks = (int)std::min<long>((total + 255) / 256, 4096);
   auto stream = c10::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(maxpool2x2_fwd_kernel, dim3(blocks), dim3(256), 0,
-                     stream.stream(), bfp(input), bfp_mut(out),
-                     idx.data_ptr<unsigned char>(), (long)(B * C),
-                     (int)H, (int)W);
+  if (W % 16 == 0) {
+    long groups = B * C * (H / 2) * (W / 16);
+    int gb = (int)std::min<long>((groups + 255) / 256, 4096);
+    hipLaunchKernelGGL(maxpool2x2_fwd_vec_kernel, dim3(gb), dim3(256), 0,
+                       stream.stream(), bfp(input), bfp_mut(out),
+                       idx.data_ptr<unsigned char>(), (long)(B * C),
+                       (int)H, (int)W);
+  } else {
+    hipLaunchKernelGGL(maxpool2x2_fwd_kernel, dim3(blocks), dim3(256), 0,
+                       stream.stream(), bfp(input), bfp_mut(out),
+                       idx.data_ptr<unsigned char>(), (long)(B * C),
+                       (int)H, (int)W);
+  }
   return {out, idx};
 }
 
@@ -842,10 +860,19 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor dout, torch::Tensor idx,
   long total = din.numel();
   int blocks = (int)std::min<long>((total + 255) / 256, 4096);
   auto stream = c10::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(maxpool2x2_bwd_kernel, dim3(blocks), dim3(256), 0,
-                     stream.stream(), bfp(dout),
-                     idx.data_ptr<unsigned char>(), bfp_mut(din),
-                     (long)(B * C), (int)H, (int)W);
+  if (W % 8 == 0) {
+    long groups = B * C * H * (W / 8);
+    int gb = (int)std::min<long>((groups + 255) / 256, 4096);
+    hipLaunchKernelGGL(maxpool2x2_bwd_vec_kernel, dim3(gb), dim3(256), 0,
+                       stream.stream(), bfp(dout),
+                       idx.data_ptr<unsigned char>(), bfp_mut(din),
+                       (long)(B * C), (int)H, (int)W);
+  } else {
+    hipLaunchKernelGGL(maxpool2x2_bwd_kernel, dim3(blocks), dim3(256), 0,
+                       stream.stream(), bfp(dout),
+                       idx.data_ptr<unsigned char>(), bfp_mut(din),
+                       (long)(B * C), (int)H, (int)W);
+  }
   return din;
 }
 
