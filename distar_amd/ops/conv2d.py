@@ -166,7 +166,11 @@ class Conv2dHIP(nn.Conv2d):
     def forward(self, x):
         if self._use_hip(x):
             kh, kw = self.kernel_size
-            if x.shape[2] * x.shape[3] <= _small_hw_limit():
+            # 1x1 convs are batched GEMMs outright (col == x, no im2col);
+            # 3x3 below the HW threshold go through materialized im2col
+            if (kh == 1 and os.environ.get('DISTAR_AMD_CONV_1X1GEMM',
+                                           '1') == '1') \
+                    or x.shape[2] * x.shape[3] <= _small_hw_limit():
                 return _ConvSmallHWFn.apply(x.contiguous(), self.weight,
                                             self.bias, kh, kw, kh // 2,
                                             kw // 2)
@@ -210,6 +214,10 @@ def conv2d(x, weight, bias=None, padding=0):
             and pad == (kh // 2, kw // 2)
             and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'
             and os.environ.get('DISTAR_AMD_CONV') != '0'):
+        if (kh == 1 and os.environ.get('DISTAR_AMD_CONV_1X1GEMM', '1') == '1') \
+                or x.shape[2] * x.shape[3] <= _small_hw_limit():
+            return _ConvSmallHWFn.apply(x.contiguous(), weight, bias, kh, kw,
+                                        kh // 2, kw // 2)
         return _Conv2dFn.apply(x.contiguous(), weight, bias, kh, kw,
                                kh // 2, kw // 2)
     return F.conv2d(x, weight, bias, padding=padding)
