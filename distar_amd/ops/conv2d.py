@@ -78,14 +78,14 @@ class _Conv2dFn(torch.autograd.Function):
 
 
 class _ConvSmallHWFn(torch.autograd.Function):
-    """Small-spatial convs (the 19x20 ResBlock/GatedResBlock stack):
-    materialized im2col + batched hipBLASLt GEMMs.  At HW ~ 380 the
-    in-kernel gather of the implicit-GEMM kernels runs ~8x off the HBM
-    floor (profiles/r02_notes.md r2dd attribution: 86 ms/step across the
-    17 such convs); here data movement is two dedicated HIP kernels
-    (im2col / gather-col2im, no atomics) and every contraction is a plain
-    library GEMM.  The col tensor is saved for wgrad (~1.8 GB per 3x3
-    conv at the SL batch - sized for MI355X's 288 GB)."""
+    """Materialized im2col + batched hipBLASLt GEMM conv (data movement =
+    two dedicated HIP kernels: im2col / gather-col2im, no atomics).
+
+    MEASURED NULL RESULT, kept env-gated (DISTAR_AMD_CONV_SMALLHW /
+    DISTAR_AMD_CONV_1X1GEMM): the same-box A/B in profiles/r02_notes.md
+    (r2gg) has the implicit-GEMM kernels at 382.4 ms/step vs 410.0 with
+    the 19x20 stack on this path — the col round-trip plus 2048-image
+    tiny batched GEMMs lose to the in-kernel gather that stays in L2."""
 
     @staticmethod
     def forward(ctx, x, weight, bias, KH, KW, padH, padW):
@@ -140,10 +140,14 @@ class _ConvSmallHWFn(torch.autograd.Function):
 
 
 def _small_hw_limit():
+    # default 0 = always implicit-GEMM: the materialized-col path measured
+    # SLOWER at every threshold (r2ff/r2gg same-box A/B: off 382.4 ms,
+    # 19x20-only 410.0, +38x40 421.6, +76x80 454.3 — the 2048-image tiny
+    # batched GEMMs and the col round-trip lose to the in-kernel gather)
     try:
-        return int(os.environ.get('DISTAR_AMD_CONV_SMALLHW', '512'))
+        return int(os.environ.get('DISTAR_AMD_CONV_SMALLHW', '0'))
     except ValueError:
-        return 512
+        return 0
 
 
 def _col_save_bytes():
@@ -169,7 +173,7 @@ class Conv2dHIP(nn.Conv2d):
             # 1x1 convs are batched GEMMs outright (col == x, no im2col);
             # 3x3 below the HW threshold go through materialized im2col
             if (kh == 1 and os.environ.get('DISTAR_AMD_CONV_1X1GEMM',
-                                           '1') == '1') \
+                                           '0') == '1') \
                     or x.shape[2] * x.shape[3] <= _small_hw_limit():
                 return _ConvSmallHWFn.apply(x.contiguous(), self.weight,
                                             self.bias, kh, kw, kh // 2,
@@ -214,7 +218,7 @@ def conv2d(x, weight, bias=None, padding=0):
             and pad == (kh // 2, kw // 2)
             and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'
             and os.environ.get('DISTAR_AMD_CONV') != '0'):
-        if (kh == 1 and os.environ.get('DISTAR_AMD_CONV_1X1GEMM', '1') == '1') \
+        if (kh == 1 and os.environ.get('DISTAR_AMD_CONV_1X1GEMM', '0') == '1') \
                 or x.shape[2] * x.shape[3] <= _small_hw_limit():
             return _ConvSmallHWFn.apply(x.contiguous(), weight, bias, kh, kw,
                                         kh // 2, kw // 2)
