@@ -698,10 +698,10 @@ torch::Tensor conv2d_fwd(torch::Tensor input, torch::Tensor wp,
   auto stream = c10::hip::getCurrentHIPStream();
   if (Cout == 1 && KH == 3 && KW == 3 && padH == 1 && padW == 1 &&
       Cin % 16 == 0 && Cin <= 256 && B < 65536) {
-    // LDS-staged stencil: input tile read once, 3x3 window out of LDS
-    dim3 sg((unsigned)((W + 63) / 64), (unsigned)((H + 15) / 16),
-            (unsigned)B);
-    hipLaunchKernelGGL(conv2d_stencil_c1_fwd_kernel, sg, dim3(256), 0,
+    // direct-load stencil: row items, vector loads, register partials
+    long items = B * H * ((W + 15) / 16);
+    int sgb = (int)std::min<long>((items + 255) / 256, 8192);
+    hipLaunchKernelGGL(conv2d_stencil_c1_fwd_kernel, dim3(sgb), dim3(256), 0,
                        stream.stream(), bfp(input), bfp(wp), bp,
                        bfp_mut(out), (int)B, (int)Cin, (int)H, (int)W,
                        (int)Kpad, relu ? 1 : 0);
@@ -747,15 +747,15 @@ std::vector<torch::Tensor> conv2d_wgrad(
     auto stream2 = c10::hip::getCurrentHIPStream();
     torch::Tensor dbs;
     if (Cout == 1 && KH == 3 && KW == 3 && padH == 1 && padW == 1 &&
-        Cin <= 32) {
-      // LDS-staged stencil wgrad with fused dbias (LocationHead 32->1)
+        Cin <= 32 && (Cin & (Cin - 1)) == 0) {
+      // direct-load stencil wgrad with fused dbias (LocationHead 32->1)
       float* dbp2 = nullptr;
       if (want_bias) {
         dbs = torch::zeros({1}, input.options().dtype(torch::kFloat32));
         dbp2 = dbs.data_ptr<float>();
       }
-      long tiles = B * ((W + 63) / 64) * ((H + 7) / 8);
-      int g = (int)std::min<long>(tiles, 4096);
+      long items = B * H * Cin;
+      int g = (int)std::min<long>((items + 255) / 256, 8192);
       hipLaunchKernelGGL(conv2d_stencil_c1_wgrad_kernel, dim3(g), dim3(256),
                          0, stream2.stream(), bfp(input), bfp(dout),
                          dwd.data_ptr<float>(), dbp2,

@@ -456,87 +456,79 @@ extern "C" __global__ void conv2d_wgrad_small_kernel(
 
 // ---------------------------------------------------------------------------
 // Cout=1 3x3 s1p1 stencil pair (the LocationHead 32->1 output conv at
-// (B*T, 32, 152, 160) — the only <16-channel conv in the model).  The
-// generic small kernels above re-read the input from global once per tap
-// (~9x traffic, scalar 2B loads); these stage each input tile through LDS
-// once and run the 3x3 window out of LDS, which puts both kernels at the
-// HBM floor (~3.3 GB per pass) instead of ~30x over it.
-
-#define STH 16                    // stencil fwd: output rows per block
-#define STW 64                    // output cols per block
-#define SCC 16                    // ci staged per chunk
-
-extern "C" __global__ __launch_bounds__(256, 3)
+// (B*T, 32, 152, 160) — the only <16-channel conv in the model).
+//
+// v2: direct-load row accumulators.  v1 staged tiles through LDS with one
+// scalar global load per element consumed behind a barrier; PMC (r2jj)
+// showed both kernels 10-20x over their VALU instruction floor — pure
+// latency serialization.  Here each thread owns a row item, issues its
+// row reads as independent 16-40 B vector loads (the compiler keeps a
+// ci-iteration's 9+ loads in flight), keeps all partials in registers,
+// and never synchronizes.
+extern "C" __global__ __launch_bounds__(256, 4)
 void conv2d_stencil_c1_fwd_kernel(
     const __hip_bfloat16* __restrict__ input,   // (B, Cin, H, W)
     const __hip_bfloat16* __restrict__ wp,      // (1, Kpad), k = ci*9+tap
-    const float* __restrict__ bias,             // (1) or nullptr
+    const float* __restrict__ bias,
     __hip_bfloat16* __restrict__ out,           // (B, 1, H, W)
     int B, int Cin, int H, int W, int Kpad, int relu) {
-  __shared__ __hip_bfloat16 xs[SCC * 18 * 66];  // [ci][row+2halo][col+2halo]
   __shared__ __hip_bfloat16 ws[256 * 9];        // [ci][tap], Cin <= 256
-  const int tid = threadIdx.x;
-  const long b = blockIdx.z;
-  const int y0 = blockIdx.y * STH;
-  const int x0 = blockIdx.x * STW;
-  for (int e = tid; e < Cin * 9; e += 256) ws[e] = wp[e];
-  const int prow = tid >> 4;                    // 16 rows
-  const int pcol = (tid & 15) * 4;              // 4 adjacent px per thread
-  float acc[4] = {};
-  const int nchunk = Cin >> 4;                  // Cin % 16 == 0 (host checks)
-  for (int ch = 0; ch < nchunk; ++ch) {
-    __syncthreads();
-    for (int e = tid; e < SCC * 18 * 66; e += 256) {
-      int q = (int)(((long)e * 63551) >> 22);   // e / 66
-      int c = e - q * 66;
-      int ci = (int)(((unsigned)q * 58255) >> 20);  // q / 18
-      int rr = q - ci * 18;
-      int gy = y0 + rr - 1, gx = x0 + c - 1;
-      __hip_bfloat16 v = __float2bfloat16(0.f);
-      if (gy >= 0 && gy < H && gx >= 0 && gx < W)
-        v = input[((b * Cin + ch * SCC + ci) * (long)H + gy) * W + gx];
-      xs[e] = v;
-    }
-    __syncthreads();
-    for (int ci = 0; ci < SCC; ++ci) {
+  for (int e = threadIdx.x; e < Cin * 9; e += 256) ws[e] = wp[e];
+  __syncthreads();
+  const int CW = (W + 15) >> 4;                 // 16-px chunks per row
+  const long items = (long)B * H * CW;          // item = (b, y, chunk)
+  const float bb = bias ? bias[0] : 0.f;
+  for (long i = (long)blockIdx.x * 256 + threadIdx.x; i < items;
+       i += (long)gridDim.x * 256) {
+    const int c = (int)(i % CW);
+    const long q = i / CW;
+    const int y = (int)(q % H);
+    const long b = q / H;
+    const int x0 = c * 16;
+    const bool interior = (x0 >= 16) && (x0 + 18 <= W);
+    float acc[16] = {};
+    for (int ci = 0; ci < Cin; ++ci) {
       float wv[9];
       #pragma unroll
-      for (int t9 = 0; t9 < 9; ++t9)
-        wv[t9] = __bfloat162float(ws[(ch * SCC + ci) * 9 + t9]);
-      const __hip_bfloat16* xrow = &xs[ci * 18 * 66];
+      for (int t = 0; t < 9; ++t) wv[t] = __bfloat162float(ws[ci * 9 + t]);
+      const __hip_bfloat16* plane = input + ((b * Cin + ci) * (long)H) * W;
       #pragma unroll
-      for (int r3 = 0; r3 < 3; ++r3) {
-        // output row prow needs lds rows prow..prow+2; cols pcol..pcol+5
-        union { unsigned u[3]; __hip_bfloat16 h[6]; } raw;
-        const unsigned* sp = (const unsigned*)(xrow + (prow + r3) * 66 + pcol);
-        raw.u[0] = sp[0]; raw.u[1] = sp[1]; raw.u[2] = sp[2];
-        float v[6];
-        #pragma unroll
-        for (int j = 0; j < 6; ++j) v[j] = __bfloat162float(raw.h[j]);
+      for (int r = 0; r < 3; ++r) {
+        const int gy = y + r - 1;
+        if (gy < 0 || gy >= H) continue;
+        const __hip_bfloat16* row = plane + (long)gy * W;
+        float f[18];
+        if (interior) {
+          __hip_bfloat16 v[20];                 // cols x0-2 .. x0+17, 40 B
+          __builtin_memcpy(v, row + x0 - 2, 40);
+          #pragma unroll
+          for (int j = 0; j < 18; ++j) f[j] = __bfloat162float(v[j + 1]);
+        } else {
+          #pragma unroll
+          for (int j = 0; j < 18; ++j) {
+            const int xx = x0 - 1 + j;
+            f[j] = (xx >= 0 && xx < W) ? __bfloat162float(row[xx]) : 0.f;
+          }
+        }
         #pragma unroll
         for (int dx = 0; dx < 3; ++dx)
           #pragma unroll
-          for (int j = 0; j < 4; ++j)
-            acc[j] += wv[r3 * 3 + dx] * v[j + dx];
+          for (int j = 0; j < 16; ++j)
+            acc[j] += wv[r * 3 + dx] * f[j + dx];
       }
     }
-  }
-  const float bb = bias ? bias[0] : 0.f;
-  const int gy = y0 + prow;
-  if (gy < H) {
-    __hip_bfloat16* orow = out + (b * (long)H + gy) * W + x0 + pcol;
-    if (x0 + pcol + 3 < W) {
-      bf16x8c pack;                              // 4 used, 8B store
-      __hip_bfloat16 hv[4];
+    __hip_bfloat16* orow = out + (b * (long)H + y) * W + x0;
+    if (x0 + 16 <= W) {
+      __hip_bfloat16 hv[16];
       #pragma unroll
-      for (int j = 0; j < 4; ++j) {
+      for (int j = 0; j < 16; ++j) {
         float o = acc[j] + bb;
         if (relu) o = fmaxf(o, 0.f);
         hv[j] = __float2bfloat16(o);
       }
-      __builtin_memcpy(orow, hv, 8);
+      __builtin_memcpy(orow, hv, 32);
     } else {
-      for (int j = 0; j < 4 && x0 + pcol + j < W; ++j) {
+      for (int j = 0; j < 16 && x0 + j < W; ++j) {
         float o = acc[j] + bb;
         if (relu) o = fmaxf(o, 0.f);
         orow[j] = __float2bfloat16(o);
@@ -545,83 +537,96 @@ void conv2d_stencil_c1_fwd_kernel(
   }
 }
 
-// wgrad: dW[0,ci,ty,tx] = sum_px dout[px] * in[ci, px + (ty-1, tx-1)],
-// dbias = sum dout.  Persistent accumulators: each block walks a strided
-// slice of the (B x row-slab x col-tile) tiles, thread (ci, strip) keeps
-// its 9 tap partials in registers the whole kernel, one atomicAdd set at
-// the end (tiles >> blocks, so dW contention is per-block not per-tile).
-#define GTH 8                     // input rows per tile
-#define GTW 64                    // input cols per tile
-
+// wgrad: item = (b, y, ci); thread ci is constant across the grid-stride
+// walk (host guarantees Cin is a power of two <= 32, so 256 % Cin == 0),
+// all 9 tap partials + the bias sum ride in registers for the whole
+// kernel, then one shfl/LDS/global-atomic reduction per block.
 extern "C" __global__ __launch_bounds__(256, 4)
 void conv2d_stencil_c1_wgrad_kernel(
-    const __hip_bfloat16* __restrict__ input,   // (B, Cin, H, W), Cin <= 32
+    const __hip_bfloat16* __restrict__ input,   // (B, Cin, H, W)
     const __hip_bfloat16* __restrict__ dout,    // (B, 1, H, W)
     float* __restrict__ dw,                     // (1, Cin, 3, 3) fp32
     float* __restrict__ dbias,                  // (1) fp32 or nullptr
     int B, int Cin, int H, int W) {
-  __shared__ __hip_bfloat16 xs[32 * GTH * GTW]; // [ci][row][col], no halo
-  __shared__ __hip_bfloat16 dys[10 * 66];       // dout tile + 1-halo
+  __shared__ float reds[32 * 9];
+  __shared__ float redb;
   const int tid = threadIdx.x;
-  const int ci = tid >> 3;                      // 0..31
-  const int strip = tid & 7;                    // strip == tile row
+  const int ci = tid & (Cin - 1);
   float acc[9] = {};
   float accb = 0.f;
-  const int ntx = (W + GTW - 1) / GTW, nty = (H + GTH - 1) / GTH;
-  const long tiles = (long)B * ntx * nty;
-  for (long t = blockIdx.x; t < tiles; t += gridDim.x) {
-    long q = t / ntx;
-    const int tx = (int)(t - q * ntx);
-    const long b = q / nty;
-    const int ty = (int)(q - b * nty);
-    const int y0 = ty * GTH, x0 = tx * GTW;
-    __syncthreads();
-    for (int e = tid; e < Cin * GTH * GTW; e += 256) {
-      int c = e & 63, r = (e >> 6) & 7, cc = e >> 9;
-      int gy = y0 + r, gx = x0 + c;
-      xs[e] = (gy < H && gx < W)
-          ? input[((b * Cin + cc) * (long)H + gy) * W + gx]
-          : __float2bfloat16(0.f);
-    }
-    for (int e = tid; e < 10 * 66; e += 256) {
-      int r = (int)(((long)e * 63551) >> 22);   // e / 66
-      int c = e - r * 66;
-      int gy = y0 + r - 1, gx = x0 + c - 1;
-      dys[e] = (gy >= 0 && gy < H && gx >= 0 && gx < W)
-          ? dout[(b * (long)H + gy) * W + gx]
-          : __float2bfloat16(0.f);
-    }
-    __syncthreads();
-    if (ci < Cin) {
-      const __hip_bfloat16* xrow = &xs[(ci * GTH + strip) * GTW];
-      const __hip_bfloat16* drow = &dys[0];
-      for (int ii = 0; ii < GTW; ++ii) {
-        int i = (ii + strip * 8 + ci * 17) & 63;  // stagger LDS banks
-        float xf = __bfloat162float(xrow[i]);
+  const long items = (long)B * H * Cin;         // item = (b, y, ci)
+  for (long i = (long)blockIdx.x * 256 + tid; i < items;
+       i += (long)gridDim.x * 256) {
+    const long q = i / Cin;                     // i % Cin == ci
+    const int y = (int)(q % H);
+    const long b = q / H;
+    const __hip_bfloat16* xrow = input + ((b * Cin + ci) * (long)H + y) * W;
+    const __hip_bfloat16* dbase = dout + b * (long)H * W;
+    for (int x0 = 0; x0 < W; x0 += 16) {
+      float xf[16];
+      if (x0 + 16 <= W) {
+        __hip_bfloat16 v[16];
+        __builtin_memcpy(v, xrow + x0, 32);
         #pragma unroll
-        for (int ty3 = 0; ty3 < 3; ++ty3)
+        for (int j = 0; j < 16; ++j) xf[j] = __bfloat162float(v[j]);
+      } else {
+        #pragma unroll
+        for (int j = 0; j < 16; ++j)
+          xf[j] = (x0 + j < W) ? __bfloat162float(xrow[x0 + j]) : 0.f;
+      }
+      const bool interior = (x0 >= 16) && (x0 + 18 <= W);
+      // dW[ty,tx] += x[y][x] * dy[y+1-ty][x+1-tx]
+      #pragma unroll
+      for (int ty = 0; ty < 3; ++ty) {
+        const int gy = y + 1 - ty;
+        if (gy < 0 || gy >= H) continue;
+        const __hip_bfloat16* drow = dbase + (long)gy * W;
+        float f[18];                            // dy cols x0-1 .. x0+16
+        if (interior) {
+          __hip_bfloat16 v[20];
+          __builtin_memcpy(v, drow + x0 - 2, 40);
           #pragma unroll
-          for (int tx3 = 0; tx3 < 3; ++tx3)
-            acc[ty3 * 3 + tx3] += xf *
-                __bfloat162float(drow[(strip + 2 - ty3) * 66 + (i + 2 - tx3)]);
-        if (ci == 0)
-          accb += __bfloat162float(drow[(strip + 1) * 66 + (i + 1)]);
+          for (int j = 0; j < 18; ++j) f[j] = __bfloat162float(v[j + 1]);
+        } else {
+          #pragma unroll
+          for (int j = 0; j < 18; ++j) {
+            const int xx = x0 - 1 + j;
+            f[j] = (xx >= 0 && xx < W) ? __bfloat162float(drow[xx]) : 0.f;
+          }
+        }
+        #pragma unroll
+        for (int tx = 0; tx < 3; ++tx)
+          #pragma unroll
+          for (int j = 0; j < 16; ++j)
+            acc[ty * 3 + tx] += xf[j] * f[j + 2 - tx];
+        if (ty == 1 && ci == 0) {
+          #pragma unroll
+          for (int j = 0; j < 16; ++j) accb += f[j + 1];
+        }
       }
     }
   }
+  // lanes l and l^Cin .. share ci; fold strips above Cin
   #pragma unroll
-  for (int o = 1; o < 8; o <<= 1) {             // reduce the 8 strips
-    #pragma unroll
-    for (int k9 = 0; k9 < 9; ++k9) acc[k9] += __shfl_xor(acc[k9], o, 64);
-    accb += __shfl_xor(accb, o, 64);
+  for (int o = 32; o >= 1; o >>= 1) {
+    if (o >= Cin) {
+      #pragma unroll
+      for (int k = 0; k < 9; ++k) acc[k] += __shfl_xor(acc[k], o, 64);
+      accb += __shfl_xor(accb, o, 64);
+    }
   }
-  if (strip == 0 && ci < Cin) {
+  if (tid < Cin * 9) reds[tid] = 0.f;
+  if (tid == 0) redb = 0.f;
+  __syncthreads();
+  if ((tid & 63) < Cin) {                       // lane ci of each wave
     #pragma unroll
-    for (int k9 = 0; k9 < 9; ++k9) atomicAdd(&dw[ci * 9 + k9], acc[k9]);
-    if (ci == 0 && dbias) atomicAdd(dbias, accb);
+    for (int k = 0; k < 9; ++k) atomicAdd(&reds[ci * 9 + k], acc[k]);
+    if (ci == 0) atomicAdd(&redb, accb);
   }
+  __syncthreads();
+  if (tid < Cin * 9) atomicAdd(&dw[tid], reds[tid]);
+  if (tid == 0 && dbias) atomicAdd(dbias, redb);
 }
-
 
 // ---------------------------------------------------------------------------
 // Small-HW conv path (the 19x20 ResBlock/GatedResBlock stack, HW = 380):
