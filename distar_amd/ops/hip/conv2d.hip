@@ -127,6 +127,25 @@ void conv2d_fwd_kernel(
         if (pbase + 7 < HW && x0 + 7 < W && y >= 0 && y < H &&
             x >= 0 && x + 7 < W) {
           __builtin_memcpy(vals, src, 16);
+        } else if (kwin == 1 && pbase + 8 <= HW) {
+          // 1x1 im2col is x itself: contiguous ACROSS row boundaries —
+          // the row-crossing check above is irrelevant here
+          __builtin_memcpy(vals, inb + (long)ci * HW + pbase, 16);
+        } else if (W >= 8) {
+          // branchless row-crossing gather: a px-run of 8 crosses at most
+          // one row when W >= 8, so each of the 8 loads is independent
+          // (the previous sequential ++x0 walk serialized their latency)
+          const __hip_bfloat16* cib = inb + (long)ci * H * W;
+          #pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int wrap = (x0 + j >= W) ? 1 : 0;
+            const int yj = y0 + wrap + dy;
+            const int xj = x0 + j - wrap * W + dx;
+            const bool ok = (pbase + j < HW) && (yj >= 0) && (yj < H) &&
+                            (xj >= 0) && (xj < W);
+            const __hip_bfloat16 t = cib[ok ? ((long)yj * W + xj) : 0];
+            if (ok) vals[j] = t;
+          }
         } else {
           const __hip_bfloat16* cib = inb + (long)ci * H * W;
           for (int j = 0; j < 8; ++j) {
@@ -243,6 +262,23 @@ void conv2d_wgrad_kernel(
           if (pbase + 7 < HW && x0 + 7 < W && y >= 0 && y < H &&
               x >= 0 && x + 7 < W) {
             __builtin_memcpy(vals, src, 16);
+          } else if (KH * KW == 1 && pbase + 8 <= HW) {
+            // 1x1 im2col is x itself: contiguous across row boundaries
+            __builtin_memcpy(vals, inb + (long)ci * HW + pbase, 16);
+          } else if (W >= 8) {
+            // branchless: <= 1 row crossing per 8-run when W >= 8; the 8
+            // loads are independent (the ++x0 walk serialized latency)
+            const __hip_bfloat16* cib = inb + (long)ci * H * W;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              const int wrap = (x0 + j >= W) ? 1 : 0;
+              const int yj = y0 + wrap + dy;
+              const int xj = x0 + j - wrap * W + dx;
+              const bool ok = (pbase + j < HW) && (yj >= 0) && (yj < H) &&
+                              (xj >= 0) && (xj < W);
+              const __hip_bfloat16 t = cib[ok ? ((long)yj * W + xj) : 0];
+              if (ok) vals[j] = t;
+            }
           } else {
             const __hip_bfloat16* cib = inb + (long)ci * H * W;
             for (int j = 0; j < 8; ++j) {
@@ -615,8 +651,8 @@ void conv2d_stencil_c1_wgrad_kernel(
       accb += __shfl_xor(accb, o, 64);
     }
   }
-  if (tid < Cin * 9) reds[tid] = 0.f;
-  if (tid == 0) redb = 0.f;
+  for (int e = tid; e < Cin * 9; e += 256) reds[e] = 0.f;   // Cin*9 can be
+  if (tid == 0) redb = 0.f;                                 // > blockDim
   __syncthreads();
   if ((tid & 63) < Cin) {                       // lane ci of each wave
     #pragma unroll
@@ -624,7 +660,7 @@ void conv2d_stencil_c1_wgrad_kernel(
     if (ci == 0) atomicAdd(&redb, accb);
   }
   __syncthreads();
-  if (tid < Cin * 9) atomicAdd(&dw[tid], reds[tid]);
+  for (int e = tid; e < Cin * 9; e += 256) atomicAdd(&dw[e], reds[e]);
   if (tid == 0 && dbias) atomicAdd(dbias, redb);
 }
 
