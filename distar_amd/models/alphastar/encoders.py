@@ -217,6 +217,8 @@ class SpatialEncoder(nn.Module):
                        w[:, n_const:].contiguous(), bias=conv.bias)
         for layer in list(self.project)[1:]:        # norm/act of the block
             out = layer(out)
+        if getattr(conv, 'fuse_relu', False):       # act folded into the
+            out = torch.relu(out)                   # conv it bypassed here
         map_skip = []
         for i in range(len(self.downsample)):
             map_skip.append(out)

@@ -139,11 +139,17 @@ def conv2d_block(in_channels, out_channels, kernel_size, stride=1, padding=0,
                      padding=padding, dilation=dilation, groups=groups)
     _weight_init(conv.weight, init_type, activation)
     block.append(conv)
-    if norm_type is not None and norm_type != 'none':
+    has_norm = norm_type is not None and norm_type != 'none'
+    if has_norm:
         block.append(build_normalization(norm_type, dim=2)(out_channels))
     act = build_activation(activation) if isinstance(activation, str) else activation
     if act is not None:
-        block.append(act)
+        if isinstance(act, nn.ReLU) and not has_norm:
+            # fold the relu into the conv kernel epilogue (saves a full
+            # read+write pass; the backward mask reuses the saved output)
+            conv.fuse_relu = True
+        else:
+            block.append(act)
     return sequential_pack(block)
 
 

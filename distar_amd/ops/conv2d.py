@@ -40,13 +40,17 @@ def _pack_weight_flipped(w, KH, KW):
 
 class _Conv2dFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, KH, KW, padH, padW):
+    def forward(ctx, x, weight, bias, KH, KW, padH, padW, relu=False):
         ops = hip_ext.require()
         wp = _pack_weight(weight, KH, KW)
         b32 = bias.detach().float().contiguous() if bias is not None else None
         out = ops.conv2d_fwd(x, wp, b32, weight.shape[0], KH, KW, padH, padW,
-                             False)
-        ctx.save_for_backward(x, weight)
+                             relu)
+        if relu:
+            ctx.save_for_backward(x, weight, out)   # out>0 is the relu mask
+        else:
+            ctx.save_for_backward(x, weight)
+        ctx.relu = relu
         ctx.dims = (KH, KW, padH, padW)
         ctx.has_bias = bias is not None
         return out
@@ -54,7 +58,11 @@ class _Conv2dFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         ops = hip_ext.require()
-        x, weight = ctx.saved_tensors
+        if ctx.relu:
+            x, weight, out = ctx.saved_tensors
+            dy = torch.ops.aten.threshold_backward(dy, out, 0)
+        else:
+            x, weight = ctx.saved_tensors
         KH, KW, padH, padW = ctx.dims
         dy = dy.contiguous().to(torch.bfloat16)
         Cout, Cin = weight.shape[0], weight.shape[1]
@@ -74,7 +82,7 @@ class _Conv2dFn(torch.autograd.Function):
                 db = dbias.to(weight.dtype)    # fused into the wgrad kernel
         if want_bias and db is None:
             db = dy.sum(dim=(0, 2, 3)).to(weight.dtype)
-        return dx, dw, db, None, None, None, None
+        return dx, dw, db, None, None, None, None, None
 
 
 class _ConvSmallHWFn(torch.autograd.Function):
@@ -167,6 +175,8 @@ class Conv2dHIP(nn.Conv2d):
                 and os.environ.get('DISTAR_AMD_DISABLE_HIP') != '1'
                 and os.environ.get('DISTAR_AMD_CONV') != '0')
 
+    fuse_relu = False             # set by conv2d_block when act folds in
+
     def forward(self, x):
         if self._use_hip(x):
             kh, kw = self.kernel_size
@@ -175,18 +185,20 @@ class Conv2dHIP(nn.Conv2d):
             if (kh == 1 and os.environ.get('DISTAR_AMD_CONV_1X1GEMM',
                                            '0') == '1') \
                     or x.shape[2] * x.shape[3] <= _small_hw_limit():
-                return _ConvSmallHWFn.apply(x.contiguous(), self.weight,
-                                            self.bias, kh, kw, kh // 2,
-                                            kw // 2)
+                out = _ConvSmallHWFn.apply(x.contiguous(), self.weight,
+                                           self.bias, kh, kw, kh // 2,
+                                           kw // 2)
+                return F.relu(out) if self.fuse_relu else out
             return _Conv2dFn.apply(x.contiguous(), self.weight, self.bias,
-                                   kh, kw, kh // 2, kw // 2)
+                                   kh, kw, kh // 2, kw // 2, self.fuse_relu)
         if x.is_cuda and os.environ.get('DISTAR_AMD_CONV_DEBUG') == '1':
             import sys
             print(f'[Conv2dHIP fallback] shape={tuple(x.shape)} dtype={x.dtype} '
                   f'k={self.kernel_size} s={self.stride} p={self.padding} '
                   f'd={self.dilation} g={self.groups}', file=sys.stderr,
                   flush=True)
-        return super().forward(x)
+        out = super().forward(x)
+        return F.relu(out) if self.fuse_relu else out
 
 
 class _MaxPool2x2Fn(torch.autograd.Function):

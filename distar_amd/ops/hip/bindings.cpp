@@ -77,6 +77,7 @@ torch::Tensor vtrace_scan(torch::Tensor clipped_rhos, torch::Tensor clipped_cs,
 
 extern "C" __global__ void upsample2x_fwd_f32(const float*, float*, int, int, int);
 extern "C" __global__ void upsample2x_fwd_bf16(const __hip_bfloat16*, __hip_bfloat16*, int, int, int);
+extern "C" __global__ void upsample2x_fwd_bf16_fast(const __hip_bfloat16*, __hip_bfloat16*, int, int, int);
 extern "C" __global__ void upsample2x_bwd_f32(const float*, float*, int, int, int);
 extern "C" __global__ void upsample2x_bwd_bf16(const __hip_bfloat16*, __hip_bfloat16*, int, int, int);
 extern "C" __global__ void upsample2x_bwd_bf16_fast(const __hip_bfloat16*, __hip_bfloat16*, int, int, int);
@@ -99,7 +100,9 @@ torch::Tensor upsample2x(torch::Tensor input) {
                        stream.stream(), input.data_ptr<float>(),
                        out.data_ptr<float>(), (int)(N * C), (int)H, (int)W);
   } else if (input.scalar_type() == torch::kBFloat16) {
-    hipLaunchKernelGGL(upsample2x_fwd_bf16, dim3(blocks), dim3(threads), 0,
+    long groups = N * C * H * 2 * (W * 2 / 8);
+    int gb = (int)std::min<long>((groups + threads - 1) / threads, 8192);
+    hipLaunchKernelGGL(upsample2x_fwd_bf16_fast, dim3(gb), dim3(threads), 0,
                        stream.stream(),
                        reinterpret_cast<const __hip_bfloat16*>(input.data_ptr()),
                        reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),

@@ -111,6 +111,87 @@ __device__ void upsample2x_bwd(const T* __restrict__ gout, T* __restrict__ gin,
   }
 }
 
+// Fast fwd: scale-2 align_corners=false weights are constant by parity
+// (out[2s] = .25 in[s-1] + .75 in[s]; out[2s+1] = .75 in[s] + .25 in[s+1],
+// separable in y) — the generic kernel burned a taps2x() per output px.
+// Interior items first, border items after (uniform waves, as in bwd).
+extern "C" __global__ void upsample2x_fwd_bf16_fast(
+    const bf16* __restrict__ in, bf16* __restrict__ out,
+    int NC, int H, int W) {
+  const int H2 = H * 2, W2 = W * 2;
+  const int WG = W2 / 8;                        // 8 out px = 4 src pairs
+  // interior: out rows 2..H2-3 (src rows fully valid), out col groups
+  // whose src cols s-1..s+4 stay in range: groups 1..WG-2
+  const int IW = WG - 2, IH = H2 - 4;
+  const long total_i = (IW > 0 && IH > 0) ? (long)NC * IH * IW : 0;
+  const int bc = 4 * WG + 2 * (H2 - 4);         // border groups per plane
+  const long total = total_i + (long)NC * bc;
+  for (long g = (long)blockIdx.x * blockDim.x + threadIdx.x; g < total;
+       g += (long)gridDim.x * blockDim.x) {
+    int y, xg;
+    long nc;
+    bool interior;
+    if (g < total_i) {
+      xg = ((int)(g % IW) + 1) * 8;
+      y = (int)((g / IW) % IH) + 2;
+      nc = g / ((long)IW * IH);
+      interior = true;
+    } else {
+      const long gb = g - total_i;
+      nc = gb / bc;
+      const int r = (int)(gb - nc * bc);
+      if (r < 2 * WG) { y = r / WG; xg = (r % WG) * 8; }
+      else if (r < 4 * WG) { y = H2 - 2 + (r - 2 * WG) / WG;
+                             xg = ((r - 2 * WG) % WG) * 8; }
+      else { const int r2 = r - 4 * WG; y = 2 + (r2 >> 1);
+             xg = (r2 & 1) ? (WG - 1) * 8 : 0; }
+      interior = false;
+    }
+    const bf16* base = in + nc * (long)H * W;
+    if (interior) {
+      // src rows y0,y1 for out row y; weights by parity
+      const int s0 = (y - 1) >> 1;
+      const float wy0 = (y & 1) ? 0.75f : 0.25f;   // weight of row s0
+      const int sx = (xg - 1) >> 1;                // first src col needed
+      bf16 r0[6], r1[6];
+      __builtin_memcpy(r0, base + (long)s0 * W + sx, 12);
+      __builtin_memcpy(r1, base + (long)(s0 + 1) * W + sx, 12);
+      float f[6];
+      #pragma unroll
+      for (int t = 0; t < 6; ++t)
+        f[t] = wy0 * __bfloat162float(r0[t]) +
+               (1.f - wy0) * __bfloat162float(r1[t]);
+      bf16 ov[8];
+      #pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        // out cols xg+2j (even), xg+2j+1 (odd); src col of even = s
+        // f index: src col (xg+2j-1)>>1 ... relative to sx
+        const int fe = ((xg + 2 * j - 1) >> 1) - sx;  // = j if xg odd-base
+        ov[2 * j] = __float2bfloat16(0.25f * f[fe] + 0.75f * f[fe + 1]);
+        ov[2 * j + 1] = __float2bfloat16(0.75f * f[fe + 1] +
+                                         0.25f * f[fe + 2]);
+      }
+      __builtin_memcpy(out + (nc * (long)H2 + y) * W2 + xg, ov, 16);
+    } else {
+      bf16 vals[8];
+      int y0, y1;
+      float wy0;
+      taps2x(y, H, y0, y1, wy0);
+      const bf16* r0 = base + (long)y0 * W;
+      const bf16* r1 = base + (long)y1 * W;
+      for (int j = 0; j < 8; ++j) {
+        int x0, x1;
+        float wx0;
+        taps2x(xg + j, W, x0, x1, wx0);
+        float v = wy0 * (wx0 * ld(r0 + x0) + (1.f - wx0) * ld(r0 + x1)) +
+                  (1.f - wy0) * (wx0 * ld(r1 + x0) + (1.f - wx0) * ld(r1 + x1));
+        vals[j] = __float2bfloat16(v);
+      }
+      __builtin_memcpy(out + (nc * (long)H2 + y) * W2 + xg, vals, 16);
+    }
+  }
+}
+
 extern "C" __global__ void upsample2x_fwd_f32(const float* in, float* out,
                                               int NC, int H, int W) {
   upsample2x_fwd<float>(in, out, NC, H, W);

@@ -603,3 +603,38 @@ def test_multi_tensor_norm_clip_matches_eager():
     scale = min(1.0, 1.0 / (float(expected_norm) + 1e-6))
     for g, r in zip(grads, ref):
         torch.testing.assert_close(g, r * scale, rtol=1e-5, atol=1e-6)
+
+
+@pytest.mark.gpu
+def test_conv2d_fused_relu_matches_eager():
+    """conv2d_block folds ReLU into the conv epilogue (fwd) and applies
+    the mask via threshold_backward on the saved output (bwd)."""
+    import torch.nn.functional as F
+    from distar_amd.models.nn.blocks import conv2d_block
+    torch.manual_seed(11)
+    blk = conv2d_block(32, 64, 3, 1, 1, activation='relu').cuda()
+    conv = blk[0]
+    assert getattr(conv, 'fuse_relu', False) and len(blk) == 1
+    x0 = (torch.randn(2, 32, 76, 80, device='cuda') * 0.5).bfloat16()
+    dout = (torch.randn(2, 64, 76, 80, device='cuda') * 0.5).bfloat16()
+
+    x_h = x0.detach().clone().requires_grad_(True)
+    out_h = blk(x_h)
+    out_h.backward(dout)
+    gw, gb = conv.weight.grad.clone(), conv.bias.grad.clone()
+    conv.weight.grad = None
+    conv.bias.grad = None
+
+    x_e = x0.detach().clone().float().requires_grad_(True)
+    out_e = F.relu(F.conv2d(x_e, conv.weight, conv.bias, padding=1))
+    out_e.backward(dout.float())
+
+    torch.testing.assert_close(out_h.float(), out_e.detach(), rtol=3e-2,
+                               atol=3e-2)
+    torch.testing.assert_close(x_h.grad.float(), x_e.grad, rtol=5e-2,
+                               atol=5e-2)
+    sw = float(conv.weight.grad.abs().max())
+    torch.testing.assert_close(gw, conv.weight.grad, rtol=3e-2,
+                               atol=0.03 * sw)
+    torch.testing.assert_close(gb, conv.bias.grad, rtol=3e-2,
+                               atol=0.03 * float(conv.bias.grad.abs().max()))
