@@ -625,16 +625,20 @@ def test_conv2d_fused_relu_matches_eager():
     conv.weight.grad = None
     conv.bias.grad = None
 
-    x_e = x0.detach().clone().float().requires_grad_(True)
-    out_e = F.relu(F.conv2d(x_e, conv.weight, conv.bias, padding=1))
-    out_e.backward(dout.float())
-
-    torch.testing.assert_close(out_h.float(), out_e.detach(), rtol=3e-2,
-                               atol=3e-2)
-    torch.testing.assert_close(x_h.grad.float(), x_e.grad, rtol=5e-2,
+    out_e = F.relu(F.conv2d(x0.float(), conv.weight, conv.bias, padding=1))
+    torch.testing.assert_close(out_h.float(), out_e, rtol=3e-2, atol=3e-2)
+    # grads: elements with pre-relu output ~0 get their mask flipped by
+    # bf16-vs-fp32 accumulation-order noise, so the reference must use the
+    # KERNEL's mask (otherwise ~1% of x.grad differs by the full |dy|)
+    dy_m = dout.float() * (out_h.detach().float() > 0)
+    ref_dx = torch.nn.grad.conv2d_input(x0.shape, conv.weight, dy_m,
+                                        padding=1)
+    ref_dw = torch.nn.grad.conv2d_weight(x0.float(), conv.weight.shape,
+                                         dy_m, padding=1)
+    ref_db = dy_m.sum(dim=(0, 2, 3))
+    torch.testing.assert_close(x_h.grad.float(), ref_dx, rtol=5e-2,
                                atol=5e-2)
-    sw = float(conv.weight.grad.abs().max())
-    torch.testing.assert_close(gw, conv.weight.grad, rtol=3e-2,
-                               atol=0.03 * sw)
-    torch.testing.assert_close(gb, conv.bias.grad, rtol=3e-2,
-                               atol=0.03 * float(conv.bias.grad.abs().max()))
+    torch.testing.assert_close(gw, ref_dw, rtol=3e-2,
+                               atol=0.03 * float(ref_dw.abs().max()))
+    torch.testing.assert_close(gb, ref_db, rtol=3e-2,
+                               atol=0.03 * float(ref_db.abs().max()))
