@@ -787,6 +787,11 @@ std::vector<torch::Tensor> conv2d_wgrad(
   int kt = (int)((std::min<int64_t>(Kpad, (K_real + 31) / 32 * 32) + 63) / 64);
   long want_z = 32768 / std::max(1, kt);
   int ipb = (int)std::max<long>(1, (B + want_z - 1) / std::max<long>(1, want_z));
+  // small images: batch enough px-chunks per block that the per-block
+  // dW atomic flush amortizes (at 19x20 the old heuristic gave 1024
+  // image-blocks per k-tile = 1024 serialized atomic adds per dW cell)
+  long px_chunks = (H * W + 127) / 128;
+  ipb = std::max<long>(ipb, std::min<int64_t>(B, 24 / std::max<long>(1, px_chunks)));
   dim3 grid(kt, 1, (unsigned)((B + ipb - 1) / ipb));
   torch::Tensor dbias;
   float* dbp = nullptr;
