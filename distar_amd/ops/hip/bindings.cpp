@@ -665,6 +665,12 @@ extern "C" __global__ void conv2d_small_fwd_kernel(
 extern "C" __global__ void conv2d_wgrad_small_kernel(
     const __hip_bfloat16*, const __hip_bfloat16*, float*,
     int, int, int, int, int, int, int, int, int, int);
+extern "C" __global__ void conv2d_fwd_smallhw_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+    __hip_bfloat16*, int, int, int, int, int, int, int);
+extern "C" __global__ void conv2d_wgrad_smallhw_kernel(
+    const __hip_bfloat16*, const __hip_bfloat16*, float*, float*,
+    int, int, int, int, int, int, int);
 extern "C" __global__ void im2col_3x3_kernel(
     const __hip_bfloat16*, __hip_bfloat16*, long, int, int);
 extern "C" __global__ void col2im_3x3_kernel(
@@ -723,6 +729,16 @@ torch::Tensor conv2d_fwd(torch::Tensor input, torch::Tensor wp,
     return out;
   }
   dim3 grid((HW + 63) / 64, (unsigned)((Cout + 127) / 128), (unsigned)B);
+  if (KH == 3 && KW == 3 && padH == 1 && padW == 1 &&
+      (H + 2) * (W + 2) <= 484 && W >= 8) {
+    // windowed small-image path: padded window in LDS, branchless
+    // LDS-local im2col build (see conv2d.hip)
+    hipLaunchKernelGGL(conv2d_fwd_smallhw_kernel, grid, dim3(256), 64 * 256,
+                       stream.stream(), bfp(input), bfp(wp), bp,
+                       bfp_mut(out), (int)B, (int)Cin, (int)Cout,
+                       (int)H, (int)W, (int)Kpad, relu ? 1 : 0);
+    return out;
+  }
   hipLaunchKernelGGL(conv2d_fwd_kernel, grid, dim3(256), 64 * 256,
                      stream.stream(), bfp(input), bfp(wp), bp, bfp_mut(out),
                      (int)B, (int)Cin, (int)Cout, (int)H, (int)W,
@@ -800,6 +816,15 @@ std::vector<torch::Tensor> conv2d_wgrad(
     dbp = dbias.data_ptr<float>();
   }
   auto stream = c10::hip::getCurrentHIPStream();
+  if (KH == 3 && KW == 3 && padH == 1 && padW == 1 &&
+      (H + 2) * (W + 2) <= 484 && W >= 8) {
+    hipLaunchKernelGGL(conv2d_wgrad_smallhw_kernel, grid, dim3(256),
+                       64 * 256, stream.stream(), bfp(input), bfp(dout),
+                       dwp.data_ptr<float>(), dbp,
+                       (int)B, (int)Cin, (int)Cout, (int)H, (int)W,
+                       (int)Kpad, ipb);
+    return {dwp, dbias};
+  }
   hipLaunchKernelGGL(conv2d_wgrad_kernel, grid, dim3(256), 64 * 256,
                      stream.stream(), bfp(input), bfp(dout),
                      dwp.data_ptr<float>(), dbp,

@@ -790,3 +790,234 @@ extern "C" __global__ void maxpool2x2_bwd_vec_kernel(
     __builtin_memcpy(din + (n * H + y) * (long)W + xg, ov, 16);
   }
 }
+
+
+// ---------------------------------------------------------------------------
+// Windowed small-image variants (3x3 s1p1, (H+2)*(W+2) <= 484 — the 19x20
+// ResBlock/GatedResBlock stack).  The generic kernels gather im2col
+// elements straight from global with per-element decode + bounds; at
+// W = 20 that staging is instruction- and latency-bound (PMC r2jj: the
+// wgrad ran at 12k VALU + 11k SALU per wave).  Here each block first
+// copies the PADDED input window of the <= 15 ci it needs into LDS
+// (coalesced, zero-filled halo), then builds im2col tiles from LDS with
+// NO bounds checks (padding makes every read valid) and one decode per
+// (ci,dy) TRIPLE (dx=0,1,2 are shifted reads of the same row).
+
+#define SWHP 22                    // max padded H
+#define SWWP 22                    // max padded W
+
+// stage the padded window of ci [ci0, ci0+nci) of image b
+__device__ __forceinline__ void stage_window(
+    __hip_bfloat16* win, const __hip_bfloat16* __restrict__ input,
+    long b, int Cin, int ci0, int nci, int H, int W, int tid) {
+  const int Hp = H + 2, Wp = W + 2;
+  const int nwin = nci * Hp * Wp;
+  const __hip_bfloat16 z = __float2bfloat16(0.f);
+  for (int e = tid; e < nwin; e += 256) {
+    const int c = e % Wp;
+    const int t = e / Wp;
+    const int r = t % Hp;
+    const int ic = t / Hp;
+    const int gy = r - 1, gx = c - 1;
+    win[e] = (gy >= 0 && gy < H && gx >= 0 && gx < W)
+        ? input[((b * Cin + ci0 + ic) * (long)H + gy) * W + gx] : z;
+  }
+}
+
+// build one [k][px] (transpose=0: [px][k]) im2col chunk tile from the
+// window: k rows [k_base, k_base+krows), px [p0, p0+npx)
+__device__ __forceinline__ void build_tile_from_window(
+    char* lds, const __hip_bfloat16* win, int ci0,
+    int k_base, int krows, int p0, int npx, int K_real,
+    int H, int W, int wrecip, int transpose, int tid) {
+  const int Hp = H + 2, Wp = W + 2;
+  const int HW = H * W;
+  const int t0 = k_base / 3;                    // triple = ci*3 + dy
+  const int t1 = (min(k_base + krows, K_real) + 2) / 3;
+  const int ntrip = t1 - t0;
+  const int ngrp = npx / 8;
+  const __hip_bfloat16 z = __float2bfloat16(0.f);
+  for (int task = tid; task < ntrip * ngrp; task += 256) {
+    const int g = task / ntrip;
+    const int tr = task - g * ntrip + t0;
+    const int ci = (tr * 21846) >> 16;          // tr / 3 (tr < 3*128*3)
+    const int dyy = tr - ci * 3;
+    const __hip_bfloat16* w0 = win + (ci - ci0) * Hp * Wp;
+    const int pbase = p0 + g * 8;
+    const int y0p = (int)(((long)pbase * wrecip) >> 20);
+    const int x0p = pbase - y0p * W;
+    __hip_bfloat16 v3[3][8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int wrap = (x0p + j >= W) ? 1 : 0;
+      const int yj = y0p + wrap;
+      const int xj = x0p + j - wrap * W;
+      const bool ok = (pbase + j < HW);
+      const int off = ok ? (yj + dyy) * Wp + xj : 0;
+      #pragma unroll
+      for (int dx = 0; dx < 3; ++dx)
+        v3[dx][j] = ok ? w0[off + dx] : z;
+    }
+    #pragma unroll
+    for (int dx = 0; dx < 3; ++dx) {
+      const int k = tr * 3 + dx;
+      const int kk = k - k_base;
+      if (kk < 0 || kk >= krows || k >= K_real) continue;
+      if (transpose) {                          // [k][px] (wgrad A^T)
+        __builtin_memcpy(lds + cswz(kk, g * 8), v3[dx], 16);
+      } else {                                  // [px][k] (fwd A)
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+          *(__hip_bfloat16*)(lds + cswz(g * 8 + j, kk)) = v3[dx][j];
+      }
+    }
+  }
+}
+
+extern "C" __global__ __launch_bounds__(256, 4)
+void conv2d_fwd_smallhw_kernel(
+    const __hip_bfloat16* __restrict__ input,   // (B, Cin, H, W)
+    const __hip_bfloat16* __restrict__ wp,      // (Cout, Kpad)
+    const float* __restrict__ bias,
+    __hip_bfloat16* __restrict__ out,           // (B, Cout, H, W)
+    int B, int Cin, int Cout, int H, int W, int Kpad, int relu) {
+  const int HW = H * W;
+  const int p0 = blockIdx.x * CTILE;
+  const int n0 = blockIdx.y * 128;
+  const long b = blockIdx.z;
+  if (p0 >= HW || n0 >= Cout) return;
+  const int K_real = Cin * 9;
+  const int wrecip = (1048576 + W - 1) / W;
+  extern __shared__ char lds[];                 // [64px][128k] 16 KB
+  __shared__ __hip_bfloat16 win[15 * SWHP * SWWP];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int l16 = lane & 15;
+  const int lq = lane >> 4;
+  const int band = wave * 16;
+  const int tid = threadIdx.x;
+  f32x4c acc[2][4];
+  for (int h = 0; h < 2; ++h)
+    for (int nt = 0; nt < 4; ++nt) acc[h][nt] = (f32x4c){0, 0, 0, 0};
+  const int k_hi = (K_real + 31) / 32 * 32;
+  for (int k0 = 0; k0 < k_hi; k0 += KC) {
+    const int krows = min(k_hi - k0, KC);
+    const int ci0 = k0 / 9;
+    const int ci1 = min((min(k0 + krows, K_real) - 1) / 9, Cin - 1);
+    __syncthreads();
+    stage_window(win, input, b, Cin, ci0, ci1 - ci0 + 1, H, W, tid);
+    __syncthreads();
+    build_tile_from_window(lds, win, ci0, k0, krows, p0, CTILE,
+                           K_real, H, W, wrecip, 0, tid);
+    __syncthreads();
+    const int ks_count = krows / 32;
+    for (int ks = 0; ks < ks_count; ++ks) {
+      bf16x8c bi[4];
+      for (int nt = 0; nt < 4; ++nt)
+        bi[nt] = clds8(lds, cswz(nt * 16 + l16, ks * 32 + lq * 8));
+      for (int h = 0; h < 2; ++h) {
+        if (n0 + h * 64 + band >= Cout) continue;
+        int co_a = n0 + h * 64 + band + l16;
+        bf16x8c a = (co_a < Cout)
+            ? *(const bf16x8c*)(wp + (long)co_a * Kpad + k0 + ks * 32 + lq * 8)
+            : (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
+        for (int nt = 0; nt < 4; ++nt)
+          acc[h][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, bi[nt], acc[h][nt], 0, 0, 0);
+      }
+    }
+  }
+  for (int h = 0; h < 2; ++h) {
+    for (int r = 0; r < 4; ++r) {
+      int co = n0 + h * 64 + band + lq * 4 + r;
+      if (co >= Cout) continue;
+      float bv = bias ? bias[co] : 0.f;
+      for (int nt = 0; nt < 4; ++nt) {
+        int p = p0 + nt * 16 + l16;
+        if (p >= HW) continue;
+        float v = acc[h][nt][r] + bv;
+        if (relu) v = fmaxf(v, 0.f);
+        out[(b * Cout + co) * HW + p] = __float2bfloat16(v);
+      }
+    }
+  }
+}
+
+extern "C" __global__ __launch_bounds__(256, 2)
+void conv2d_wgrad_smallhw_kernel(
+    const __hip_bfloat16* __restrict__ input,   // (B, Cin, H, W)
+    const __hip_bfloat16* __restrict__ dout,    // (B, Cout, H, W)
+    float* __restrict__ dwp,                    // (Kpad, Cout) fp32
+    float* __restrict__ dbias,                  // (Cout) fp32 or nullptr
+    int B, int Cin, int Cout, int H, int W, int Kpad, int ipb) {
+  const int HW = H * W;
+  const int k_base = blockIdx.x * CTILE;        // 64 k-rows of dW
+  const int b_base = blockIdx.z * ipb;
+  const int K_real = Cin * 9;
+  if (k_base >= K_real) return;
+  const int n_tiles_co = (Cout + 15) / 16;
+  const int wrecip = (1048576 + W - 1) / W;
+  extern __shared__ char lds[];                 // [64k][128px] 16 KB
+  __shared__ __hip_bfloat16 win[9 * SWHP * SWWP];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int l16 = lane & 15;
+  const int lq = lane >> 4;
+  const int band = wave * 16;
+  const int tid = threadIdx.x;
+  f32x4c acc[8];
+  for (int nt = 0; nt < 8; ++nt) acc[nt] = (f32x4c){0, 0, 0, 0};
+  const bool do_db = dbias && blockIdx.x == 0 && wave == 0;
+  float acc_db[8] = {};
+  const int ci0 = k_base / 9;
+  const int ci1 = min((min(k_base + CTILE, K_real) - 1) / 9, Cin - 1);
+  for (int bi = 0; bi < ipb && b_base + bi < B; ++bi) {
+    const long b = b_base + bi;
+    const __hip_bfloat16* dob = dout + b * Cout * (long)HW;
+    __syncthreads();
+    stage_window(win, input, b, Cin, ci0, ci1 - ci0 + 1, H, W, tid);
+    for (int p0 = 0; p0 < HW; p0 += KC) {
+      __syncthreads();
+      build_tile_from_window(lds, win, ci0, k_base, CTILE, p0, KC,
+                             K_real, H, W, wrecip, 1, tid);
+      __syncthreads();
+      for (int ks = 0; ks < 4; ++ks) {
+        bf16x8c a = clds8(lds, cswz(band + l16, ks * 32 + lq * 8));
+        for (int nt = 0; nt < n_tiles_co; ++nt) {
+          int co = nt * 16 + l16;
+          bf16x8c bdo = (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
+          if (co < Cout) {
+            int p = p0 + ks * 32 + lq * 8;
+            if (p + 8 <= HW) {
+              __builtin_memcpy(&bdo, dob + (long)co * HW + p, 16);
+            } else {
+              for (int j = 0; j < 8; ++j)
+                if (p + j < HW)
+                  bdo[j] = ((const __bf16*)dob)[(long)co * HW + p + j];
+            }
+          }
+          acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bdo, acc[nt],
+                                                            0, 0, 0);
+          if (do_db) {
+            float sdb = 0.f;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) sdb += __bfloat162float(bdo[j]);
+            acc_db[nt] += sdb;
+          }
+        }
+      }
+    }
+  }
+  for (int nt = 0; nt < n_tiles_co; ++nt) {
+    for (int r = 0; r < 4; ++r) {
+      int k = k_base + band + lq * 4 + r;
+      int co = nt * 16 + l16;
+      if (k < Kpad && co < Cout && k < K_real)
+        atomicAdd(&dwp[(long)k * Cout + co], acc[nt][r]);
+    }
+    if (do_db) {
+      int co = nt * 16 + l16;
+      if (co < Cout) atomicAdd(&dbias[co], acc_db[nt]);
+    }
+  }
+}
