@@ -729,16 +729,10 @@ torch::Tensor conv2d_fwd(torch::Tensor input, torch::Tensor wp,
     return out;
   }
   dim3 grid((HW + 63) / 64, (unsigned)((Cout + 127) / 128), (unsigned)B);
-  if (KH == 3 && KW == 3 && padH == 1 && padW == 1 &&
-      (H + 2) * (W + 2) <= 484 && W >= 8) {
-    // windowed small-image path: padded window in LDS, branchless
-    // LDS-local im2col build (see conv2d.hip)
-    hipLaunchKernelGGL(conv2d_fwd_smallhw_kernel, grid, dim3(256), 64 * 256,
-                       stream.stream(), bfp(input), bfp(wp), bp,
-                       bfp_mut(out), (int)B, (int)Cin, (int)Cout,
-                       (int)H, (int)W, (int)Kpad, relu ? 1 : 0);
-    return out;
-  }
+  // (a windowed fwd variant was measured SLOWER: each block owns one
+  // 64-px tile, so the per-chunk window restage gets no cross-tile reuse
+  // — see profiles/r02_notes.md; the wgrad variant below keeps it because
+  // its px loop is INSIDE the block)
   hipLaunchKernelGGL(conv2d_fwd_kernel, grid, dim3(256), 64 * 256,
                      stream.stream(), bfp(input), bfp(wp), bp, bfp_mut(out),
                      (int)B, (int)Cin, (int)Cout, (int)H, (int)W,
