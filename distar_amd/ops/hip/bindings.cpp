@@ -813,13 +813,13 @@ std::vector<torch::Tensor> conv2d_wgrad(
   if (KH == 3 && KW == 3 && padH == 1 && padW == 1 &&
       (H + 2) * (W + 2) <= 484 && W >= 8) {
     hipLaunchKernelGGL(conv2d_wgrad_smallhw_kernel, grid, dim3(256),
-                       (64 + 128) * 256, stream.stream(), bfp(input), bfp(dout),
+                       64 * 256, stream.stream(), bfp(input), bfp(dout),
                        dwp.data_ptr<float>(), dbp,
                        (int)B, (int)Cin, (int)Cout, (int)H, (int)W,
                        (int)Kpad, ipb);
     return {dwp, dbias};
   }
-  hipLaunchKernelGGL(conv2d_wgrad_kernel, grid, dim3(256), (64 + 128) * 256,
+  hipLaunchKernelGGL(conv2d_wgrad_kernel, grid, dim3(256), 64 * 256,
                      stream.stream(), bfp(input), bfp(dout),
                      dwp.data_ptr<float>(), dbp,
                      (int)B, (int)Cin, (int)Cout, (int)H, (int)W,

@@ -294,33 +294,24 @@ void conv2d_wgrad_kernel(
         // cols g*8..g*8+7 share one swizzled granule: single 16B write
         __builtin_memcpy(lds + cswz(kk, g * 8), vals, 16);
       }
-      // stage the dout chunk ONCE per block: every wave previously
-      // re-streamed the same [Cout x 128px] slab from global (4x redundant
-      // reads — ~14 GB per 19x20 call, the measured wgrad bound)
-      {
-        const int co_rows = n_tiles_co * 16;
-        for (int task = tid; task < co_rows * (KC / 8); task += 256) {
-          int co = task >> 4, g = task & 15;
-          int p = p0 + g * 8;
-          __hip_bfloat16 dv[8] = {};
-          if (co < Cout) {
-            if (p + 8 <= HW) {
-              __builtin_memcpy(dv, dob + (long)co * HW + p, 16);
-            } else {
-              for (int j = 0; j < 8; ++j)
-                if (p + j < HW)
-                  dv[j] = dob[(long)co * HW + p + j];
-            }
-          }
-          __builtin_memcpy(lds + 16384 + cswz(co, g * 8), dv, 16);
-        }
-      }
       __syncthreads();
       for (int ks = 0; ks < 4; ++ks) {
         bf16x8c a = clds8(lds, cswz(band + l16, ks * 32 + lq * 8));
         for (int nt = 0; nt < n_tiles_co; ++nt) {
           int co = nt * 16 + l16;
-          bf16x8c bdo = clds8(lds + 16384, cswz(co, ks * 32 + lq * 8));
+          bf16x8c bdo = (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
+          if (co < Cout) {
+            int p = p0 + ks * 32 + lq * 8;
+            if (p + 8 <= HW) {
+              // HW*2B is not always 16B-aligned per-row: memcpy lets the
+              // compiler emit the widest legal loads
+              __builtin_memcpy(&bdo, dob + (long)co * HW + p, 16);
+            } else {
+              for (int j = 0; j < 8; ++j)
+                if (p + j < HW)
+                  bdo[j] = ((const __bf16*)dob)[(long)co * HW + p + j];
+            }
+          }
           acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bdo, acc[nt],
                                                             0, 0, 0);
           if (do_db) {
@@ -989,30 +980,22 @@ void conv2d_wgrad_smallhw_kernel(
       __syncthreads();
       build_tile_from_window(lds, win, ci0, k_base, CTILE, p0, KC,
                              K_real, H, W, wrecip, 1, tid);
-      {
-        const int co_rows = n_tiles_co * 16;   // dout chunk staged once
-        for (int task = tid; task < co_rows * (KC / 8); task += 256) {
-          int co = task >> 4, g = task & 15;
-          int p = p0 + g * 8;
-          __hip_bfloat16 dv[8] = {};
-          if (co < Cout) {
-            if (p + 8 <= HW) {
-              __builtin_memcpy(dv, dob + (long)co * HW + p, 16);
-            } else {
-              for (int j = 0; j < 8; ++j)
-                if (p + j < HW)
-                  dv[j] = dob[(long)co * HW + p + j];
-            }
-          }
-          __builtin_memcpy(lds + 16384 + cswz(co, g * 8), dv, 16);
-        }
-      }
       __syncthreads();
       for (int ks = 0; ks < 4; ++ks) {
         bf16x8c a = clds8(lds, cswz(band + l16, ks * 32 + lq * 8));
         for (int nt = 0; nt < n_tiles_co; ++nt) {
           int co = nt * 16 + l16;
-          bf16x8c bdo = clds8(lds + 16384, cswz(co, ks * 32 + lq * 8));
+          bf16x8c bdo = (bf16x8c){0, 0, 0, 0, 0, 0, 0, 0};
+          if (co < Cout) {
+            int p = p0 + ks * 32 + lq * 8;
+            if (p + 8 <= HW) {
+              __builtin_memcpy(&bdo, dob + (long)co * HW + p, 16);
+            } else {
+              for (int j = 0; j < 8; ++j)
+                if (p + j < HW)
+                  bdo[j] = ((const __bf16*)dob)[(long)co * HW + p + j];
+            }
+          }
           acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bdo, acc[nt],
                                                             0, 0, 0);
           if (do_db) {
