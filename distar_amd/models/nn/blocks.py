@@ -91,16 +91,24 @@ def sequential_pack(layers):
 def fc_block(in_channels, out_channels, init_type='xavier', activation=None,
              norm_type=None, use_dropout=False, dropout_probability=0.5):
     """Linear [+norm] [+act] [+dropout]; Linear lives at index 0."""
-    block = [nn.Linear(in_channels, out_channels)]
-    _weight_init(block[0].weight, init_type, activation)
-    if norm_type is not None and norm_type != 'none':
-        block.append(build_normalization(norm_type, dim=1)(out_channels))
     act = build_activation(activation, inplace=False) \
         if isinstance(activation, str) else activation
     if isinstance(act, nn.ReLU) and act.inplace:
         act = nn.ReLU(inplace=False)   # see build_activation docstring
-    if act is not None:
-        block.append(act)
+    has_norm = norm_type is not None and norm_type != 'none'
+    if isinstance(act, nn.ReLU) and not has_norm:
+        # relu runs inside the hipBLASLt GEMM epilogue (ops/linear_relu.py);
+        # same state-dict keys (the Linear stays at index 0)
+        from ...ops.linear_relu import FusedLinearReLU
+        block = [FusedLinearReLU(in_channels, out_channels)]
+        _weight_init(block[0].weight, init_type, activation)
+    else:
+        block = [nn.Linear(in_channels, out_channels)]
+        _weight_init(block[0].weight, init_type, activation)
+        if has_norm:
+            block.append(build_normalization(norm_type, dim=1)(out_channels))
+        if act is not None:
+            block.append(act)
     if use_dropout:
         block.append(nn.Dropout(dropout_probability))
     return sequential_pack(block)

@@ -642,3 +642,35 @@ def test_conv2d_fused_relu_matches_eager():
                                atol=0.03 * float(ref_dw.abs().max()))
     torch.testing.assert_close(gb, ref_db, rtol=3e-2,
                                atol=0.03 * float(ref_db.abs().max()))
+
+
+@pytest.mark.gpu
+def test_fused_linear_relu_matches_eager():
+    """fc_block's [Linear, ReLU] runs through the hipBLASLt relu epilogue
+    with a hand-supplied backward (threshold mask on the saved output)."""
+    from distar_amd.models.nn.blocks import fc_block
+    from distar_amd.ops.linear_relu import FusedLinearReLU
+    torch.manual_seed(3)
+    blk = fc_block(256, 512, activation='relu').cuda()
+    assert isinstance(blk[0], FusedLinearReLU)
+    lin = blk[0]
+    x0 = (torch.randn(6, 33, 256, device='cuda') * 0.5).bfloat16()
+    dout = (torch.randn(6, 33, 512, device='cuda') * 0.5).bfloat16()
+
+    x_h = x0.detach().clone().requires_grad_(True)
+    out_h = blk(x_h)
+    out_h.backward(dout)
+    gw, gb = lin.weight.grad.clone(), lin.bias.grad.clone()
+
+    out_e = torch.relu(torch.nn.functional.linear(
+        x0.float(), lin.weight, lin.bias))
+    torch.testing.assert_close(out_h.float(), out_e, rtol=3e-2, atol=3e-2)
+    dy_m = dout.float() * (out_h.detach().float() > 0)   # kernel's mask
+    ref_dx = dy_m @ lin.weight
+    ref_dw = dy_m.reshape(-1, 512).t() @ x0.float().reshape(-1, 256)
+    ref_db = dy_m.sum(dim=(0, 1))
+    torch.testing.assert_close(x_h.grad.float(), ref_dx, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(gw, ref_dw, rtol=3e-2,
+                               atol=0.03 * float(ref_dw.abs().max()))
+    torch.testing.assert_close(gb, ref_db, rtol=3e-2,
+                               atol=0.03 * float(ref_db.abs().max()))
