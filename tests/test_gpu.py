@@ -674,3 +674,39 @@ def test_fused_linear_relu_matches_eager():
                                atol=0.03 * float(ref_dw.abs().max()))
     torch.testing.assert_close(gb, ref_db, rtol=3e-2,
                                atol=0.03 * float(ref_db.abs().max()))
+
+
+@pytest.mark.gpu
+def test_film_location_head_autocast():
+    """FiLM-conditioned LocationHead (the alternative league pipeline's
+    location path) under cuda autocast vs the CPU fp32 reference —
+    closes the SURVEY/roadmap gap of the FiLM path having no GPU test."""
+    import copy
+    from distar_amd.models.alphastar.heads import LocationHead
+    from distar_amd.utils.config import Config, read_config
+    import os as _os
+    cfg = read_config(_os.path.join(
+        _os.path.dirname(__file__), '..', 'distar_amd', 'models',
+        'alphastar', 'actor_critic_default_config.yaml'))
+    whole = Config(copy.deepcopy(dict(cfg)))
+    whole.model.policy.head.location_head.film = True
+    whole.model.policy.head.location_head.gate = False
+    torch.manual_seed(7)
+    head_cpu = LocationHead(whole).float()
+    head_gpu = copy.deepcopy(head_cpu).cuda()
+    B = 3
+    emb = torch.randn(B, 1024)
+    skips = [torch.randn(B, 128, 19, 20) * 0.3 for _ in range(7)]
+    loc = torch.randint(0, 152 * 160, (B,))
+
+    logits_cpu, _ = head_cpu(emb, skips, loc)
+    with torch.autocast('cuda', dtype=torch.bfloat16):
+        logits_gpu, _ = head_gpu(emb.cuda(), [s.cuda() for s in skips],
+                                 loc.cuda())
+    assert torch.isfinite(logits_gpu).all()
+    # bf16 through 4 FiLM res stages + upsample chain: loose elementwise
+    torch.testing.assert_close(logits_gpu.float().cpu(), logits_cpu,
+                               rtol=8e-2, atol=8e-2)
+    logits_gpu.float().sum().backward()
+    assert all(torch.isfinite(p.grad).all() for p in head_gpu.parameters()
+               if p.grad is not None)
