@@ -68,9 +68,18 @@ class _Conv2dFn(torch.autograd.Function):
         Cout, Cin = weight.shape[0], weight.shape[1]
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            wpf = _pack_weight_flipped(weight, KH, KW)
-            dx = ops.conv2d_fwd(dy, wpf, None, Cin, KH, KW,
-                                KH - 1 - padH, KW - 1 - padW, False)
+            if KH == 1 and Cin < 16:
+                # tiny-Cin 1x1 backward-data is a plain batched GEMM; the
+                # direct VALU kernel ran 4.7 ms/step on the value encoder's
+                # 16->10 dx at (272, 152, 160) (profiles r2ww)
+                w2 = weight.detach().to(torch.bfloat16).reshape(Cout, Cin)
+                dx = torch.matmul(
+                    w2.t(), dy.reshape(dy.shape[0], Cout, -1)) \
+                    .reshape(x.shape).contiguous()
+            else:
+                wpf = _pack_weight_flipped(weight, KH, KW)
+                dx = ops.conv2d_fwd(dy, wpf, None, Cin, KH, KW,
+                                    KH - 1 - padH, KW - 1 - padW, False)
         want_bias = ctx.has_bias and ctx.needs_input_grad[2]
         if ctx.needs_input_grad[1]:
             K = Cin * KH * KW
