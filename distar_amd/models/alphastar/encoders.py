@@ -81,7 +81,8 @@ class ScalarEncoder(nn.Module):
             if k == 'time':
                 continue
             if item['arc'] == 'one_hot':
-                enc = nn.Embedding(item['num_embeddings'], item['embedding_dim'])
+                from ...ops.linear_relu import EmbeddingGEMM
+                enc = EmbeddingGEMM(item['num_embeddings'], item['embedding_dim'])
                 nn.init.xavier_uniform_(enc.weight)
                 self.encode_modules[k] = enc
                 self.one_hot_keys.append(k)
@@ -412,8 +413,9 @@ class ValueEncoder(nn.Module):
                 self.encode_modules[k] = fc_block(item['input_dim'], item['output_dim'],
                                                   activation=self.act)
             elif item['arc'] == 'one_hot':
-                self.encode_modules[k] = nn.Embedding(item['num_embeddings'],
-                                                      item['embedding_dim'])
+                from ...ops.linear_relu import EmbeddingGEMM
+                self.encode_modules[k] = EmbeddingGEMM(item['num_embeddings'],
+                                                       item['embedding_dim'])
         bo_cfg = self.cfg.modules.beginning_order
         self.encode_modules['beginning_order'] = BeginningBuildOrderEncoder(self.whole_cfg, bo_cfg)
         self.scatter_project = fc_block(self.cfg.scatter.scatter_input_dim,
