@@ -244,55 +244,37 @@ extern "C" __global__ void upsample2x_bwd_bf16_fast(
     int NC, int H, int W) {
   const int H2 = H * 2, W2 = W * 2;
   const int WG = W / 4;
-  // interior items first, border items after: a wave is then (almost)
-  // always uniformly fast-path or uniformly generic — the mixed layout
-  // serialized both paths in nearly every wave at these widths (r2jj)
-  const int IW = WG - 2, IH = H - 2;
-  const long total_i = (IW > 0 && IH > 0) ? (long)NC * IH * IW : 0;
-  const int bc = 2 * WG + 2 * (H - 2);          // border groups per plane
-  const long total = total_i + (long)NC * bc;
-  const float wy[4] = {0.25f, 0.75f, 0.75f, 0.25f};
-  for (long g = (long)blockIdx.x * blockDim.x + threadIdx.x; g < total;
+  const long total_g = (long)NC * H * WG;
+  for (long g = (long)blockIdx.x * blockDim.x + threadIdx.x; g < total_g;
        g += (long)gridDim.x * blockDim.x) {
-    int sy, sxg;
-    long nc;
-    if (g < total_i) {
-      sxg = (int)(g % IW) * 4 + 4;
-      sy = (int)((g / IW) % IH) + 1;
-      nc = g / ((long)IW * IH);
-      const bf16* base = gout + nc * (long)H2 * W2;
-      float acc[4] = {};
-      #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        bf16 v[12];
-        __builtin_memcpy(v, base + (long)(sy * 2 - 1 + i) * W2 + sxg * 2 - 2,
-                         24);
-        float f[12];
-        #pragma unroll
-        for (int t = 1; t < 12; ++t) f[t] = __bfloat162float(v[t]);
-        #pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[j] += wy[i] * (0.25f * (f[2 * j + 1] + f[2 * j + 4]) +
-                             0.75f * (f[2 * j + 2] + f[2 * j + 3]));
-      }
-      bf16 outv[4];
-      #pragma unroll
-      for (int j = 0; j < 4; ++j) outv[j] = __float2bfloat16(acc[j]);
-      __builtin_memcpy(gin + (nc * (long)H + sy) * W + sxg, outv, 8);
-    } else {
-      const long gb = g - total_i;
-      nc = gb / bc;
-      const int r = (int)(gb - nc * bc);
-      if (r < WG) { sy = 0; sxg = r * 4; }
-      else if (r < 2 * WG) { sy = H - 1; sxg = (r - WG) * 4; }
-      else {
-        const int r2 = r - 2 * WG;
-        sy = 1 + (r2 >> 1);
-        sxg = (r2 & 1) ? (WG - 1) * 4 : 0;
-      }
-      const bf16* base = gout + nc * (long)H2 * W2;
+    const int sxg = (int)(g % WG) * 4;
+    const int sy = (int)((g / WG) % H);
+    const long nc = g / ((long)WG * H);
+    const bf16* base = gout + nc * (long)H2 * W2;
+    if (sy == 0 || sy == H - 1 || sxg == 0 || sxg + 4 >= W) {
       for (int j = 0; j < 4; ++j)
         upsample2x_bwd_onepx(base, gin, nc, sy, sxg + j, H, W);
+      continue;
     }
+    // interior: dest rows 2sy-1..2sy+2, dest cols 2sxg-1..2sxg+8 — read 12
+    // (24 B, 4B-aligned since sxg is even) and use elements 1..10
+    const float wy[4] = {0.25f, 0.75f, 0.75f, 0.25f};
+    float acc[4] = {};
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      bf16 v[12];
+      __builtin_memcpy(v, base + (long)(sy * 2 - 1 + i) * W2 + sxg * 2 - 2, 24);
+      float f[12];
+      #pragma unroll
+      for (int t = 1; t < 12; ++t) f[t] = __bfloat162float(v[t]);
+      #pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[j] += wy[i] * (0.25f * (f[2 * j + 1] + f[2 * j + 4]) +
+                           0.75f * (f[2 * j + 2] + f[2 * j + 3]));
+    }
+    bf16 outv[4];
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) outv[j] = __float2bfloat16(acc[j]);
+    __builtin_memcpy(gin + (nc * (long)H + sy) * W + sxg, outv, 8);
   }
 }
