@@ -532,6 +532,8 @@ def test_fused_residual_ln_matches_eager():
     (32, 1, (76, 80), 3),        # stencil path, ragged 16-col tail
     (32, 1, (152, 160), 3),      # stencil path at the real location shape
     (24, 2, (76, 80), 3),        # generic small-Cout fallback kernels
+    (48, 32, (19, 20), 3),       # windowed wgrad, ragged k-tiles (K=432)
+    (16, 16, (16, 16), 3),       # windowed wgrad, 18x18 padded window
 ])
 def test_conv2d_hip_matches_eager(cin, cout, hw, kh):
     """K4 MFMA implicit-GEMM conv vs fp32 F.conv2d (fwd + both bwds)."""
