@@ -813,14 +813,27 @@ __device__ __forceinline__ void stage_window(
   const int Hp = H + 2, Wp = W + 2;
   const int nwin = nci * Hp * Wp;
   const __hip_bfloat16 z = __float2bfloat16(0.f);
-  for (int e = tid; e < nwin; e += 256) {
-    const int c = e % Wp;
-    const int t = e / Wp;
-    const int r = t % Hp;
-    const int ic = t / Hp;
-    const int gy = r - 1, gx = c - 1;
-    win[e] = (gy >= 0 && gy < H && gx >= 0 && gx < W)
-        ? input[((b * Cin + ci0 + ic) * (long)H + gy) * W + gx] : z;
+  // 4-batched: the load of iteration i must not serialize behind the
+  // LDS store of iteration i-1 (the stencil-v1 lesson, PMC r2jj)
+  for (int e0 = tid; e0 < nwin; e0 += 1024) {
+    __hip_bfloat16 v[4];
+    #pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int e = e0 + u * 256;
+      v[u] = z;
+      if (e < nwin) {
+        const int c = e % Wp;
+        const int t = e / Wp;
+        const int r = t % Hp;
+        const int ic = t / Hp;
+        const int gy = r - 1, gx = c - 1;
+        if (gy >= 0 && gy < H && gx >= 0 && gx < W)
+          v[u] = input[((b * Cin + ci0 + ic) * (long)H + gy) * W + gx];
+      }
+    }
+    #pragma unroll
+    for (int u = 0; u < 4; ++u)
+      if (e0 + u * 256 < nwin) win[e0 + u * 256] = v[u];
   }
 }
 
