@@ -49,6 +49,61 @@ __device__ __forceinline__ float im2col_elem(
   return __bfloat162float(inb[((long)ci * H + y) * W + x]);
 }
 
+
+// one staged im2col 8-px group for k row `k` of image `inb` (shared by
+// the fwd and wgrad staging loops)
+__device__ __forceinline__ void im2col_gather8(
+    __hip_bfloat16* vals, const __hip_bfloat16* __restrict__ inb,
+    int k, int pbase, int H, int W, int HW, int KH, int KW,
+    int padH, int padW, int K_real, int kwin, int wrecip) {
+  for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.f);
+  if (k >= K_real) return;
+  int ci, dy, dx;
+  if (kwin == 1) {
+    ci = k; dy = -padH; dx = -padW;
+  } else {                           // 3x3: mul-shift div by 9 / 3
+    ci = (k * 7282) >> 16;
+    int off = k - ci * 9;
+    dy = ((off * 21846) >> 16) - padH;
+    dx = off - ((off * 21846) >> 16) * 3 - padW;
+  }
+  int y0 = (int)(((long)pbase * wrecip) >> 20);
+  int x0 = pbase - y0 * W;
+  int y = y0 + dy;
+  int x = x0 + dx;
+  const __hip_bfloat16* src = inb + ((long)ci * H + y) * W + x;
+  if (pbase + 7 < HW && x0 + 7 < W && y >= 0 && y < H &&
+      x >= 0 && x + 7 < W) {
+    __builtin_memcpy(vals, src, 16);
+  } else if (kwin == 1 && pbase + 8 <= HW) {
+    // 1x1 im2col is x itself: contiguous across row boundaries
+    __builtin_memcpy(vals, inb + (long)ci * HW + pbase, 16);
+  } else if (W >= 8) {
+    // branchless: <= 1 row crossing per 8-run when W >= 8
+    const __hip_bfloat16* cib = inb + (long)ci * H * W;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int wrap = (x0 + j >= W) ? 1 : 0;
+      const int yj = y0 + wrap + dy;
+      const int xj = x0 + j - wrap * W + dx;
+      const bool ok = (pbase + j < HW) && (yj >= 0) && (yj < H) &&
+                      (xj >= 0) && (xj < W);
+      const __hip_bfloat16 t = cib[ok ? ((long)yj * W + xj) : 0];
+      if (ok) vals[j] = t;
+    }
+  } else {
+    const __hip_bfloat16* cib = inb + (long)ci * H * W;
+    for (int j = 0; j < 8; ++j) {
+      if (pbase + j < HW) {
+        int yj = y0 + dy, xj = x0 + dx;
+        if (yj >= 0 && yj < H && xj >= 0 && xj < W)
+          vals[j] = cib[(long)yj * W + xj];
+      }
+      if (++x0 == W) { x0 = 0; ++y0; }
+    }
+  }
+}
+
 extern "C" __global__ __launch_bounds__(256, 4)
 void conv2d_fwd_kernel(
     const __hip_bfloat16* __restrict__ input,   // (B, Cin, H, W)
@@ -103,63 +158,29 @@ void conv2d_fwd_kernel(
     // banks (kk>>3 varies per lane; px-major order had 16-way conflicts,
     // px-major single-writes lost the vectorized global reads — PMC r2l/m)
     const int krecip = (65536 + krows - 1) / krows;
-    for (int task = tid; task < krows * 8; task += 256) {
-      int g = (task * krecip) >> 16;      // pixel group (fixed per wave)
-      int kk = task - g * krows;          // k row (fastest across lanes)
-      int k = k0 + kk;
-      __hip_bfloat16 vals[8] = {};
-      if (k < K_real) {
-        int ci, dy, dx;
-        if (kwin == 1) {
-          ci = k; dy = -padH; dx = -padW;
-        } else {                           // 3x3: mul-shift div by 9 / 3
-          ci = (k * 7282) >> 16;
-          int off = k - ci * 9;
-          dy = ((off * 21846) >> 16) - padH;
-          dx = off - ((off * 21846) >> 16) * 3 - padW;
-        }
-        int pbase = p0 + g * 8;
-        int y0 = (int)(((long)pbase * wrecip) >> 20);
-        int x0 = pbase - y0 * W;
-        int y = y0 + dy;
-        int x = x0 + dx;
-        const __hip_bfloat16* src = inb + ((long)ci * H + y) * W + x;
-        if (pbase + 7 < HW && x0 + 7 < W && y >= 0 && y < H &&
-            x >= 0 && x + 7 < W) {
-          __builtin_memcpy(vals, src, 16);
-        } else if (kwin == 1 && pbase + 8 <= HW) {
-          // 1x1 im2col is x itself: contiguous ACROSS row boundaries —
-          // the row-crossing check above is irrelevant here
-          __builtin_memcpy(vals, inb + (long)ci * HW + pbase, 16);
-        } else if (W >= 8) {
-          // branchless row-crossing gather: a px-run of 8 crosses at most
-          // one row when W >= 8, so each of the 8 loads is independent
-          // (the previous sequential ++x0 walk serialized their latency)
-          const __hip_bfloat16* cib = inb + (long)ci * H * W;
-          #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int wrap = (x0 + j >= W) ? 1 : 0;
-            const int yj = y0 + wrap + dy;
-            const int xj = x0 + j - wrap * W + dx;
-            const bool ok = (pbase + j < HW) && (yj >= 0) && (yj < H) &&
-                            (xj >= 0) && (xj < W);
-            const __hip_bfloat16 t = cib[ok ? ((long)yj * W + xj) : 0];
-            if (ok) vals[j] = t;
-          }
-        } else {
-          const __hip_bfloat16* cib = inb + (long)ci * H * W;
-          for (int j = 0; j < 8; ++j) {
-            if (pbase + j < HW) {
-              int yj = y0 + dy, xj = x0 + dx;
-              if (yj >= 0 && yj < H && xj >= 0 && xj < W)
-                vals[j] = cib[(long)yj * W + xj];
-            }
-            if (++x0 == W) { x0 = 0; ++y0; }
-          }
+    // 2-task batches with the LDS writes hoisted: iteration i's gather
+    // must not serialize behind iteration i-1's LDS store (PMC r2jj)
+    for (int t0 = tid; t0 < krows * 8; t0 += 512) {
+      __hip_bfloat16 vals[2][8];
+      int gs[2], kks[2];
+      #pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        const int task = t0 + u * 256;
+        gs[u] = -1;
+        if (task < krows * 8) {
+          const int g = (task * krecip) >> 16;
+          const int kk = task - g * krows;
+          gs[u] = g; kks[u] = kk;
+          im2col_gather8(vals[u], inb, k0 + kk, p0 + g * 8, H, W, HW,
+                         KH, KW, padH, padW, K_real, kwin, wrecip);
         }
       }
-      for (int j = 0; j < 8; ++j)
-        *(__hip_bfloat16*)(lds + cswz(g * 8 + j, kk)) = vals[j];
+      #pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        if (gs[u] < 0) continue;
+        for (int j = 0; j < 8; ++j)
+          *(__hip_bfloat16*)(lds + cswz(gs[u] * 8 + j, kks[u])) = vals[u][j];
+      }
     }
     __syncthreads();
     const int ks_count = min(k_hi - k0, KC) / 32;
@@ -237,62 +258,25 @@ void conv2d_wgrad_kernel(
     const __hip_bfloat16* dob = dout + b * Cout * HW;
     for (int p0 = 0; p0 < HW; p0 += KC) {
       __syncthreads();
-      // stage im2col^T chunk: rows = k (ci/dy/dx once per row), cols =
-      // px in groups of 8 -> one 16B ds_write per group
-      for (int task = tid; task < CTILE * (KC / 8); task += 256) {
-        int kk = task >> 4, g = task & 15;
-        int k = k_base + kk;
-        __hip_bfloat16 vals[8] = {};
-        if (k < K_real) {
-          int ci, dy, dx;
-          if (KH * KW == 1) {
-            ci = k; dy = -padH; dx = -padW;
-          } else {                         // 3x3: mul-shift div by 9 / 3
-            ci = (k * 7282) >> 16;
-            int off = k - ci * 9;
-            dy = ((off * 21846) >> 16) - padH;
-            dx = off - ((off * 21846) >> 16) * 3 - padW;
-          }
-          int pbase = p0 + g * 8;
-          int y0 = (int)(((long)pbase * wrecip) >> 20);
-          int x0 = pbase - y0 * W;
-          int y = y0 + dy;
-          int x = x0 + dx;
-          const __hip_bfloat16* src = inb + ((long)ci * H + y) * W + x;
-          if (pbase + 7 < HW && x0 + 7 < W && y >= 0 && y < H &&
-              x >= 0 && x + 7 < W) {
-            __builtin_memcpy(vals, src, 16);
-          } else if (KH * KW == 1 && pbase + 8 <= HW) {
-            // 1x1 im2col is x itself: contiguous across row boundaries
-            __builtin_memcpy(vals, inb + (long)ci * HW + pbase, 16);
-          } else if (W >= 8) {
-            // branchless: <= 1 row crossing per 8-run when W >= 8; the 8
-            // loads are independent (the ++x0 walk serialized latency)
-            const __hip_bfloat16* cib = inb + (long)ci * H * W;
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              const int wrap = (x0 + j >= W) ? 1 : 0;
-              const int yj = y0 + wrap + dy;
-              const int xj = x0 + j - wrap * W + dx;
-              const bool ok = (pbase + j < HW) && (yj >= 0) && (yj < H) &&
-                              (xj >= 0) && (xj < W);
-              const __hip_bfloat16 t = cib[ok ? ((long)yj * W + xj) : 0];
-              if (ok) vals[j] = t;
-            }
-          } else {
-            const __hip_bfloat16* cib = inb + (long)ci * H * W;
-            for (int j = 0; j < 8; ++j) {
-              if (pbase + j < HW) {
-                int yj = y0 + dy, xj = x0 + dx;
-                if (yj >= 0 && yj < H && xj >= 0 && xj < W)
-                  vals[j] = cib[(long)yj * W + xj];
-              }
-              if (++x0 == W) { x0 = 0; ++y0; }  // incremental, no div/mod
-            }
+      // stage im2col^T chunk (2-task batches, writes hoisted — see fwd)
+      for (int t0 = tid; t0 < CTILE * (KC / 8); t0 += 512) {
+        __hip_bfloat16 vals[2][8];
+        int kks[2], gg[2];
+        #pragma unroll
+        for (int u = 0; u < 2; ++u) {
+          const int task = t0 + u * 256;
+          gg[u] = -1;
+          if (task < CTILE * (KC / 8)) {
+            const int kk = task >> 4, g = task & 15;
+            kks[u] = kk; gg[u] = g;
+            im2col_gather8(vals[u], inb, k_base + kk, p0 + g * 8, H, W, HW,
+                           KH, KW, padH, padW, K_real, KH * KW, wrecip);
           }
         }
-        // cols g*8..g*8+7 share one swizzled granule: single 16B write
-        __builtin_memcpy(lds + cswz(kk, g * 8), vals, 16);
+        #pragma unroll
+        for (int u = 0; u < 2; ++u)
+          if (gg[u] >= 0)
+            __builtin_memcpy(lds + cswz(kks[u], gg[u] * 8), vals[u], 16);
       }
       __syncthreads();
       for (int ks = 0; ks < 4; ++ks) {
