@@ -324,3 +324,43 @@ def test_league_resume_restores_payoff_and_ratings(league):
         assert abs(lg2.trueskill.mu['MP0'] - league.trueskill.mu['MP0']) < 1e-9
     finally:
         lg2.close()
+
+
+def test_warmup_ema_meter_phases():
+    """Uniform mean during warm-up, EMA after (reference MoveAverageMeter
+    semantics, log_helper.py:483-522)."""
+    from distar_amd.league.meters import WarmupEmaMeter
+    m = WarmupEmaMeter(decay=0.9, warm_up_size=3)
+    for v in (1.0, 2.0, 3.0):
+        m.update(v)
+    assert abs(m.val - 2.0) < 1e-9          # plain mean of the warm-up
+    m.update(12.0)                          # 0.9*2 + 0.1*12
+    assert abs(m.val - 3.0) < 1e-9
+
+
+def test_league_telemetry_stats():
+    """DistStat/CumStat/UnitNumStat aggregate and pickle (they ride in the
+    league resume file — reference {dist,cum,unit_num}_stat.py)."""
+    import pickle
+    from distar_amd.league.stats import CumStat, DistStat, UnitNumStat
+    d = DistStat(decay=0.5, warm_up_size=1)
+    d.update('zerg', {'dist/bo': 4.0, 'winloss': 1, 'player_id': 'MP0',
+                      'text': 'ignored'})
+    d.update('zerg', {'dist/bo': 8.0})
+    # warm_up 1 -> first sets 4, second EMA: 0.5*4 + 0.5*8 = 6
+    assert abs(d.stat_info_dict['zerg']['dist/bo'] - 6.0) < 1e-9
+    assert d.game_count['zerg'] == 2
+
+    c = CumStat(decay=0.5, warm_up_size=1)
+    c.update('zerg', {'z_type': 2, 'spine': 1.0})
+    c.update('zerg', {'z_type': 1, 'spine': 3.0})
+    agg = c.stat_info_dict['zerg']['spine']
+    assert agg[2] == 1.0 and agg[1] == 3.0 and agg[0] == 0.0
+
+    u = UnitNumStat(decay=0.5, warm_up_size=1)
+    u.update('zerg', 0, {'unit_num': {'drone': 12, 'zergling': 6}})
+    assert u.stat_info_dict['zerg']['drone'] == 12.0
+    # all three must survive the league resume pickle
+    for obj in (d, c, u):
+        clone = pickle.loads(pickle.dumps(obj))
+        assert clone.stat_info_dict == obj.stat_info_dict
